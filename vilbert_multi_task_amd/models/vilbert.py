@@ -1,0 +1,359 @@
+"""Two-stream co-attentional ViLBERT, MI355X-native build.
+
+This module is the *definition* of the model (and, on CPU, the fp32 numerics
+oracle for every HIP kernel). The forward signature and its 10-tuple return
+are behavior-identical to the contract the reference worker pins at
+/root/reference/worker.py:287-289:
+
+    model(question, features, spatials, segment_ids, input_mask, image_mask,
+          co_attention_mask, task_tokens, output_all_attention_masks=...)
+    -> (vil_prediction, vil_prediction_gqa, vil_logit, vil_binary_prediction,
+        vil_tri_prediction, vision_prediction, vision_logit,
+        linguisic_prediction, linguisic_logit, attn_data_list)
+
+Geometry: text stream = BERT-base (12×768/12 heads/3072 FFN),
+vision stream = 6×1024/8 heads/1024 FFN on 2048-d region features + 5-d
+spatials, 6 co-attention layers through bi_hidden=1024 interleaved at
+t_biattention_id/v_biattention_id.
+
+All hot ops route through ``ops.functional`` which dispatches to the
+hand-written gfx950 HIP kernels on GPU and to plain PyTorch on CPU.
+"""
+
+from __future__ import annotations
+
+from typing import List, Optional, Tuple
+
+import torch
+import torch.nn as nn
+
+from ..config import ViLBertConfig
+from ..ops import functional as F_ops
+
+
+class FusedLayerNorm(nn.Module):
+    """LayerNorm whose forward can fuse a residual add (HIP kernel on GPU)."""
+
+    def __init__(self, dim: int, eps: float):
+        super().__init__()
+        self.weight = nn.Parameter(torch.ones(dim))
+        self.bias = nn.Parameter(torch.zeros(dim))
+        self.eps = eps
+
+    def forward(self, x: torch.Tensor, residual: Optional[torch.Tensor] = None) -> torch.Tensor:
+        return F_ops.layer_norm(x, self.weight, self.bias, self.eps, residual)
+
+
+class TextEmbeddings(nn.Module):
+    """Word + position + segment (+ optional task token) embeddings.
+
+    Task-specific token (config.task_specific_tokens, worker.py:516-517): the
+    task token embedding is inserted at position 1 (after [CLS]), growing the
+    text sequence 37 -> 38; the attention mask is extended accordingly by the
+    caller (ViLBertModel.forward).
+    """
+
+    def __init__(self, config: ViLBertConfig):
+        super().__init__()
+        self.word_embeddings = nn.Embedding(config.vocab_size, config.hidden_size)
+        self.position_embeddings = nn.Embedding(config.max_position_embeddings, config.hidden_size)
+        self.token_type_embeddings = nn.Embedding(config.type_vocab_size, config.hidden_size)
+        if config.task_specific_tokens:
+            self.task_embeddings = nn.Embedding(config.num_task_tokens, config.hidden_size)
+        self.layer_norm = FusedLayerNorm(config.hidden_size, config.layer_norm_eps)
+        self.dropout = nn.Dropout(config.hidden_dropout_prob)
+        self.task_specific_tokens = config.task_specific_tokens
+
+    def forward(
+        self,
+        input_ids: torch.Tensor,
+        token_type_ids: torch.Tensor,
+        task_ids: Optional[torch.Tensor],
+    ) -> torch.Tensor:
+        b, t = input_ids.shape
+        pos_ids = torch.arange(t, dtype=torch.long, device=input_ids.device).unsqueeze(0).expand(b, t)
+        emb = F_ops.embedding_ln(
+            input_ids,
+            pos_ids,
+            token_type_ids,
+            self.word_embeddings.weight,
+            self.position_embeddings.weight,
+            self.token_type_embeddings.weight,
+            self.layer_norm.weight,
+            self.layer_norm.bias,
+            self.layer_norm.eps,
+        )
+        if self.task_specific_tokens and task_ids is not None:
+            # insert the task token after [CLS]; it gets its own LN-free slot
+            # normalized together with the rest (normalize then insert keeps
+            # the per-token statistics identical to normalizing post-insert
+            # only for the non-task tokens; we instead embed+LN the task token
+            # through the same LN parameters for consistency).
+            task_tok = self.task_embeddings(task_ids.view(b, 1))  # [B,1,H]
+            task_tok = self.layer_norm(task_tok)
+            emb = torch.cat([emb[:, :1], task_tok, emb[:, 1:]], dim=1)
+        return self.dropout(emb)
+
+
+class ImageEmbeddings(nn.Module):
+    """Region features (2048) + spatial boxes (5) -> v_hidden, LayerNorm.
+
+    SURVEY.md §2.2 vision-embeddings row; input tensors fixed by
+    worker.py:452-455 (features [B,101,2048] f32, spatials [B,101,5] f32).
+    """
+
+    def __init__(self, config: ViLBertConfig):
+        super().__init__()
+        self.image_embeddings = nn.Linear(config.v_feature_size, config.v_hidden_size)
+        self.image_location_embeddings = nn.Linear(config.v_loc_size, config.v_hidden_size)
+        self.layer_norm = FusedLayerNorm(config.v_hidden_size, config.layer_norm_eps)
+        self.dropout = nn.Dropout(config.v_hidden_dropout_prob)
+
+    def forward(self, features: torch.Tensor, spatials: torch.Tensor) -> torch.Tensor:
+        img = self.image_embeddings(features)
+        loc = self.image_location_embeddings(spatials)
+        return self.dropout(self.layer_norm(img, residual=loc))
+
+
+class MultiHeadSelfAttention(nn.Module):
+    def __init__(self, hidden: int, heads: int, attn_dropout: float, hidden_dropout: float, eps: float):
+        super().__init__()
+        self.heads = heads
+        self.head_dim = hidden // heads
+        self.query = nn.Linear(hidden, hidden)
+        self.key = nn.Linear(hidden, hidden)
+        self.value = nn.Linear(hidden, hidden)
+        self.out = nn.Linear(hidden, hidden)
+        self.attn_dropout_p = attn_dropout
+        self.dropout = nn.Dropout(hidden_dropout)
+        self.layer_norm = FusedLayerNorm(hidden, eps)
+
+    def _split(self, x: torch.Tensor) -> torch.Tensor:
+        b, l, _ = x.shape
+        return x.view(b, l, self.heads, self.head_dim).transpose(1, 2)
+
+    def forward(
+        self, x: torch.Tensor, mask_bias: Optional[torch.Tensor], need_probs: bool
+    ) -> Tuple[torch.Tensor, Optional[torch.Tensor]]:
+        b, l, h = x.shape
+        q = self._split(self.query(x))
+        k = self._split(self.key(x))
+        v = self._split(self.value(x))
+        ctx, probs = F_ops.attention(
+            q, k, v, mask_bias, self.attn_dropout_p, self.training, need_probs
+        )
+        ctx = ctx.transpose(1, 2).reshape(b, l, h)
+        y = self.layer_norm(self.dropout(self.out(ctx)), residual=x)
+        return y, probs
+
+
+class FeedForward(nn.Module):
+    def __init__(self, hidden: int, intermediate: int, dropout: float, eps: float):
+        super().__init__()
+        self.intermediate = nn.Linear(hidden, intermediate)
+        self.output = nn.Linear(intermediate, hidden)
+        self.dropout = nn.Dropout(dropout)
+        self.layer_norm = FusedLayerNorm(hidden, eps)
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        h = torch.nn.functional.linear(x, self.intermediate.weight)  # bias fused into gelu
+        h = F_ops.bias_gelu(h, self.intermediate.bias)
+        return self.layer_norm(self.dropout(self.output(h)), residual=x)
+
+
+class TransformerLayer(nn.Module):
+    """One self-attention + FFN block (text or vision stream)."""
+
+    def __init__(self, hidden: int, heads: int, intermediate: int, attn_dropout: float, hidden_dropout: float, eps: float):
+        super().__init__()
+        self.attention = MultiHeadSelfAttention(hidden, heads, attn_dropout, hidden_dropout, eps)
+        self.ffn = FeedForward(hidden, intermediate, hidden_dropout, eps)
+
+    def forward(self, x, mask_bias, need_probs=False):
+        y, probs = self.attention(x, mask_bias, need_probs)
+        return self.ffn(y), probs
+
+
+class CrossAttention(nn.Module):
+    """One direction of bi-attention: queries from `q_stream`, K/V from the
+    other stream, through bi_hidden, projected back to q_stream's hidden."""
+
+    def __init__(self, q_hidden: int, kv_hidden: int, bi_hidden: int, heads: int,
+                 attn_dropout: float, hidden_dropout: float, eps: float):
+        super().__init__()
+        self.heads = heads
+        self.head_dim = bi_hidden // heads
+        self.query = nn.Linear(q_hidden, bi_hidden)
+        self.key = nn.Linear(kv_hidden, bi_hidden)
+        self.value = nn.Linear(kv_hidden, bi_hidden)
+        self.out = nn.Linear(bi_hidden, q_hidden)
+        self.attn_dropout_p = attn_dropout
+        self.dropout = nn.Dropout(hidden_dropout)
+        self.layer_norm = FusedLayerNorm(q_hidden, eps)
+
+    def _split(self, x: torch.Tensor) -> torch.Tensor:
+        b, l, _ = x.shape
+        return x.view(b, l, self.heads, self.head_dim).transpose(1, 2)
+
+    def forward(self, x_q, x_kv, mask_bias, need_probs=False):
+        b, lq, _ = x_q.shape
+        q = self._split(self.query(x_q))
+        k = self._split(self.key(x_kv))
+        v = self._split(self.value(x_kv))
+        ctx, probs = F_ops.attention(q, k, v, mask_bias, self.attn_dropout_p, self.training, need_probs)
+        ctx = ctx.transpose(1, 2).reshape(b, lq, self.heads * self.head_dim)
+        y = self.layer_norm(self.dropout(self.out(ctx)), residual=x_q)
+        return y, probs
+
+
+class ConnectionLayer(nn.Module):
+    """BertConnectionLayer equivalent: cross-attention in both directions plus
+    each stream's own FFN (SURVEY.md §2.2 co-attention row)."""
+
+    def __init__(self, config: ViLBertConfig):
+        super().__init__()
+        c = config
+        self.t_cross = CrossAttention(
+            c.hidden_size, c.v_hidden_size, c.bi_hidden_size, c.bi_num_attention_heads,
+            c.attention_probs_dropout_prob, c.hidden_dropout_prob, c.layer_norm_eps,
+        )
+        self.v_cross = CrossAttention(
+            c.v_hidden_size, c.hidden_size, c.bi_hidden_size, c.bi_num_attention_heads,
+            c.v_attention_probs_dropout_prob, c.v_hidden_dropout_prob, c.layer_norm_eps,
+        )
+        self.t_ffn = FeedForward(c.hidden_size, c.intermediate_size, c.hidden_dropout_prob, c.layer_norm_eps)
+        self.v_ffn = FeedForward(c.v_hidden_size, c.v_intermediate_size, c.v_hidden_dropout_prob, c.layer_norm_eps)
+
+    def forward(self, t, v, t_mask, v_mask, co_mask_tv, co_mask_vt, need_probs=False):
+        # text attends vision (keys masked by image mask + co_attention_mask)
+        t_bias = v_mask if co_mask_tv is None else v_mask + co_mask_tv
+        v_bias = t_mask if co_mask_vt is None else t_mask + co_mask_vt
+        t2, p_tv = self.t_cross(t, v, t_bias, need_probs)
+        v2, p_vt = self.v_cross(v, t, v_bias, need_probs)
+        return self.t_ffn(t2), self.v_ffn(v2), (p_tv, p_vt)
+
+
+class Pooler(nn.Module):
+    def __init__(self, in_dim: int, out_dim: int):
+        super().__init__()
+        self.dense = nn.Linear(in_dim, out_dim)
+
+    def forward(self, seq: torch.Tensor) -> torch.Tensor:
+        # first token ([CLS] for text, global region for vision)
+        return torch.tanh(self.dense(seq[:, 0]))
+
+
+class ViLBertModel(nn.Module):
+    """The two-stream encoder (no heads)."""
+
+    def __init__(self, config: ViLBertConfig):
+        super().__init__()
+        self.config = config
+        c = config
+        self.embeddings = TextEmbeddings(c)
+        self.v_embeddings = ImageEmbeddings(c)
+        self.t_layers = nn.ModuleList(
+            [
+                TransformerLayer(
+                    c.hidden_size, c.num_attention_heads, c.intermediate_size,
+                    c.attention_probs_dropout_prob, c.hidden_dropout_prob, c.layer_norm_eps,
+                )
+                for _ in range(c.num_hidden_layers)
+            ]
+        )
+        self.v_layers = nn.ModuleList(
+            [
+                TransformerLayer(
+                    c.v_hidden_size, c.v_num_attention_heads, c.v_intermediate_size,
+                    c.v_attention_probs_dropout_prob, c.v_hidden_dropout_prob, c.layer_norm_eps,
+                )
+                for _ in range(c.v_num_hidden_layers)
+            ]
+        )
+        self.c_layers = nn.ModuleList(
+            [ConnectionLayer(c) for _ in range(len(c.t_biattention_id))]
+        )
+        self.t_pooler = Pooler(c.hidden_size, c.bi_hidden_size)
+        self.v_pooler = Pooler(c.v_hidden_size, c.bi_hidden_size)
+
+    @staticmethod
+    def _extend_mask(mask: torch.Tensor, dtype: torch.dtype) -> torch.Tensor:
+        # [B,L] {0,1} -> additive bias [B,1,1,L]
+        m = mask[:, None, None, :].to(dtype)
+        return (1.0 - m) * torch.finfo(dtype).min / 2
+
+    def forward(
+        self,
+        input_ids: torch.Tensor,
+        features: torch.Tensor,
+        spatials: torch.Tensor,
+        segment_ids: torch.Tensor,
+        input_mask: torch.Tensor,
+        image_mask: torch.Tensor,
+        co_attention_mask: Optional[torch.Tensor] = None,
+        task_ids: Optional[torch.Tensor] = None,
+        output_all_attention_masks: bool = False,
+    ):
+        c = self.config
+        dtype = self.t_pooler.dense.weight.dtype
+
+        t = self.embeddings(input_ids, segment_ids, task_ids)
+        v = self.v_embeddings(features.to(dtype), spatials.to(dtype))
+
+        if c.task_specific_tokens and task_ids is not None:
+            one = torch.ones(
+                input_mask.shape[0], 1, dtype=input_mask.dtype, device=input_mask.device
+            )
+            input_mask = torch.cat([input_mask[:, :1], one, input_mask[:, 1:]], dim=1)
+            if co_attention_mask is not None:
+                zero = co_attention_mask.new_zeros(
+                    co_attention_mask.shape[0], co_attention_mask.shape[1], 1
+                )
+                co_attention_mask = torch.cat(
+                    [co_attention_mask[:, :, :1], zero, co_attention_mask[:, :, 1:]], dim=2
+                )
+
+        t_mask = self._extend_mask(input_mask, dtype)
+        v_mask = self._extend_mask(image_mask, dtype)
+
+        co_tv = co_vt = None
+        if co_attention_mask is not None and co_attention_mask.any():
+            # co_attention_mask [B, R, T] (worker.py:455): restricts
+            # vision<->text pairs. 1 = masked out.
+            cm = co_attention_mask.to(dtype)
+            neg = torch.finfo(dtype).min / 2
+            co_vt = (cm * neg)[:, None, :, :]                      # vision queries x text keys
+            co_tv = (cm.transpose(1, 2) * neg)[:, None, :, :]      # text queries x vision keys
+
+        attn_data: List = []
+        t_idx = v_idx = 0
+        for ci, (v_stop, t_stop) in enumerate(zip(c.v_biattention_id, c.t_biattention_id)):
+            while v_idx < v_stop:
+                v, p = self.v_layers[v_idx](v, v_mask, output_all_attention_masks)
+                if output_all_attention_masks:
+                    attn_data.append({"type": "v_self", "layer": v_idx, "probs": p})
+                v_idx += 1
+            while t_idx < t_stop:
+                t, p = self.t_layers[t_idx](t, t_mask, output_all_attention_masks)
+                if output_all_attention_masks:
+                    attn_data.append({"type": "t_self", "layer": t_idx, "probs": p})
+                t_idx += 1
+            t, v, (p_tv, p_vt) = self.c_layers[ci](
+                t, v, t_mask, v_mask, co_tv, co_vt, output_all_attention_masks
+            )
+            if output_all_attention_masks:
+                attn_data.append({"type": "co", "layer": ci, "probs_tv": p_tv, "probs_vt": p_vt})
+        while v_idx < len(self.v_layers):
+            v, p = self.v_layers[v_idx](v, v_mask, output_all_attention_masks)
+            if output_all_attention_masks:
+                attn_data.append({"type": "v_self", "layer": v_idx, "probs": p})
+            v_idx += 1
+        while t_idx < len(self.t_layers):
+            t, p = self.t_layers[t_idx](t, t_mask, output_all_attention_masks)
+            if output_all_attention_masks:
+                attn_data.append({"type": "t_self", "layer": t_idx, "probs": p})
+            t_idx += 1
+
+        pooled_t = self.t_pooler(t)
+        pooled_v = self.v_pooler(v)
+        return t, v, pooled_t, pooled_v, attn_data
