@@ -1,0 +1,232 @@
+"""GPU numerics: every HIP kernel vs the plain-PyTorch fp32 reference
+(SURVEY.md §4 consequence (3): single-GPU numerical parity, bf16 tolerances).
+
+All tests build reference values from the SAME bf16-rounded inputs so the
+measured error is kernel error, not input quantization.
+"""
+
+import math
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+
+@pytest.fixture(scope="module")
+def ext():
+    from vilbert_multi_task_amd.ops import hip_ext
+
+    return hip_ext.load()
+
+
+def _rand_bf16(*shape, seed=0, scale=1.0):
+    g = torch.Generator(device="cpu").manual_seed(seed)
+    return (torch.randn(*shape, generator=g) * scale).to(torch.bfloat16).cuda()
+
+
+# ---------------------------------------------------------------------------
+# MFMA fragment-layout probe: asymmetric operands catch transposes
+# (cdna_hip_programming.md §3 "Always A=I-check with ASYMMETRIC B")
+# ---------------------------------------------------------------------------
+
+def test_mfma_probe_layout(ext):
+    torch.manual_seed(0)
+    a = _rand_bf16(16, 32, seed=1)
+    b = (torch.arange(32 * 16).reshape(32, 16).float() * 0.01 - 2.0).to(
+        torch.bfloat16
+    ).cuda()
+    c = torch.ops.vilbert_amd.mfma_probe(a, b)
+    ref = a.float() @ b.float()
+    assert torch.allclose(c, ref, atol=1e-2, rtol=1e-2), (
+        (c - ref).abs().max().item()
+    )
+
+
+def test_mfma_probe_identity(ext):
+    a = torch.eye(16, 32).to(torch.bfloat16).cuda()
+    b = _rand_bf16(32, 16, seed=3)
+    c = torch.ops.vilbert_amd.mfma_probe(a, b)
+    assert torch.allclose(c, b.float()[:16], atol=1e-2)
+
+
+# ---------------------------------------------------------------------------
+# residual + LayerNorm
+# ---------------------------------------------------------------------------
+
+@pytest.mark.parametrize("dim", [768, 1024, 1536, 3072])
+@pytest.mark.parametrize("with_res", [False, True])
+def test_residual_layer_norm(ext, dim, with_res):
+    rows = 37 * 8
+    x = _rand_bf16(rows, dim, seed=dim)
+    res = _rand_bf16(rows, dim, seed=dim + 1) if with_res else None
+    w = _rand_bf16(dim, seed=dim + 2, scale=0.5)
+    b = _rand_bf16(dim, seed=dim + 3, scale=0.1)
+    y = torch.ops.vilbert_amd.residual_layer_norm(x, res, w, b, 1e-12)
+    xf = x.float() + (res.float() if with_res else 0.0)
+    ref = torch.nn.functional.layer_norm(xf, (dim,), w.float(), b.float(), 1e-12)
+    assert (y.float() - ref).abs().max() < 3e-2
+
+
+def test_layer_norm_f32(ext):
+    x = torch.randn(64, 768, device="cuda")
+    w = torch.randn(768, device="cuda")
+    b = torch.randn(768, device="cuda")
+    y = torch.ops.vilbert_amd.residual_layer_norm(x, None, w, b, 1e-12)
+    ref = torch.nn.functional.layer_norm(x, (768,), w, b, 1e-12)
+    assert (y - ref).abs().max() < 1e-5
+
+
+# ---------------------------------------------------------------------------
+# bias + GELU
+# ---------------------------------------------------------------------------
+
+def test_bias_gelu(ext):
+    x = _rand_bf16(37 * 16, 3072, seed=7, scale=2.0)
+    bias = _rand_bf16(3072, seed=8)
+    y = torch.ops.vilbert_amd.bias_gelu(x, bias)
+    ref = torch.nn.functional.gelu(x.float() + bias.float())
+    assert (y.float() - ref).abs().max() < 2e-2
+
+
+def test_gelu_no_bias_odd_tail(ext):
+    x = _rand_bf16(5, 97, seed=9)  # odd dim exercises scalar path
+    y = torch.ops.vilbert_amd.bias_gelu(x, None)
+    ref = torch.nn.functional.gelu(x.float())
+    assert (y.float() - ref).abs().max() < 2e-2
+
+
+# ---------------------------------------------------------------------------
+# embedding + LayerNorm
+# ---------------------------------------------------------------------------
+
+def test_embedding_ln(ext):
+    V, P, S, H = 30522, 512, 2, 768
+    B, T = 4, 38
+    g = torch.Generator().manual_seed(0)
+    ids = torch.randint(0, V, (B, T), generator=g).cuda()
+    pos = torch.arange(T).unsqueeze(0).expand(B, T).contiguous().cuda()
+    typ = torch.zeros(B, T, dtype=torch.long).cuda()
+    ww = _rand_bf16(V, H, seed=10, scale=0.02)
+    pw = _rand_bf16(P, H, seed=11, scale=0.02)
+    tw = _rand_bf16(2, H, seed=12, scale=0.02)
+    lw = _rand_bf16(H, seed=13, scale=0.5)
+    lb = _rand_bf16(H, seed=14, scale=0.1)
+    y = torch.ops.vilbert_amd.embedding_ln(ids, pos, typ, ww, pw, tw, lw, lb, 1e-12)
+    e = ww.float()[ids] + pw.float()[pos] + tw.float()[typ]
+    ref = torch.nn.functional.layer_norm(e, (H,), lw.float(), lb.float(), 1e-12)
+    assert (y.float() - ref).abs().max() < 3e-2
+
+
+# ---------------------------------------------------------------------------
+# fused attention: every serving shape family
+# ---------------------------------------------------------------------------
+
+def _attn_ref(q, k, v, heads, mask_bias):
+    b, lq, hd = q.shape
+    lk = k.shape[1]
+    d = hd // heads
+    qh = q.float().view(b, lq, heads, d).transpose(1, 2)
+    kh = k.float().view(b, lk, heads, d).transpose(1, 2)
+    vh = v.float().view(b, lk, heads, d).transpose(1, 2)
+    s = qh @ kh.transpose(-1, -2) / math.sqrt(d)
+    if mask_bias is not None:
+        s = s + mask_bias.float()
+    p = torch.softmax(s, dim=-1)
+    return (p @ vh).transpose(1, 2).reshape(b, lq, hd)
+
+
+@pytest.mark.parametrize(
+    "B,H,Lq,Lk,D,mask_mode",
+    [
+        (3, 12, 38, 38, 64, 1),    # text self-attention w/ input_mask
+        (2, 8, 101, 101, 128, 1),  # vision self-attention w/ image_mask
+        (2, 8, 38, 101, 128, 1),   # co-attn: text queries x vision keys
+        (2, 8, 101, 38, 128, 2),   # co-attn: vision queries x text keys, [B,Lq,Lk] bias
+        (1, 12, 37, 37, 64, 0),    # no mask, odd length
+        (5, 8, 20, 33, 128, 1),    # ragged non-multiple-of-16 lengths
+    ],
+)
+def test_attention_vs_reference(ext, B, H, Lq, Lk, D, mask_mode):
+    q = _rand_bf16(B, Lq, H * D, seed=B * 100 + Lq)
+    k = _rand_bf16(B, Lk, H * D, seed=B * 100 + Lk + 1)
+    v = _rand_bf16(B, Lk, H * D, seed=B * 100 + Lk + 2)
+    if mask_mode == 0:
+        mask = None
+        ref_mask = None
+    elif mask_mode == 1:
+        keep = torch.ones(B, Lk)
+        keep[:, Lk - 5 :] = 0  # mask the tail keys
+        mask = ((1 - keep) * -1e9).to(torch.bfloat16).cuda().view(B, 1, 1, Lk)
+        ref_mask = mask.float()
+    else:
+        keep = torch.ones(B, Lq, Lk)
+        keep[:, :, Lk - 7 :] = 0
+        mask = ((1 - keep) * -1e9).to(torch.bfloat16).cuda().view(B, 1, Lq, Lk)
+        ref_mask = mask.float()
+    out = torch.ops.vilbert_amd.attention(q, k, v, H, mask)
+    ref = _attn_ref(q, k, v, H, ref_mask)
+    err = (out.float() - ref).abs().max().item()
+    assert err < 4e-2, f"max err {err}"
+
+
+def test_attention_masked_rows_sum_to_valid(ext):
+    """Fully masked tail keys must contribute exactly zero probability."""
+    B, H, Lq, Lk, D = 2, 8, 101, 101, 128
+    q = _rand_bf16(B, Lq, H * D, seed=50)
+    k = _rand_bf16(B, Lk, H * D, seed=51)
+    v_const = torch.zeros(B, Lk, H * D)
+    v_const[:, :40] = 1.0  # valid keys have value 1, masked keys 0
+    v = v_const.to(torch.bfloat16).cuda()
+    keep = torch.zeros(B, Lk)
+    keep[:, :40] = 1
+    mask = ((1 - keep) * -1e9).to(torch.bfloat16).cuda().view(B, 1, 1, Lk)
+    out = torch.ops.vilbert_amd.attention(q, k, v, H, mask)
+    # weighted average of ones over valid keys = 1 exactly
+    assert (out.float() - 1.0).abs().max() < 2e-2
+
+
+# ---------------------------------------------------------------------------
+# batched multi-class NMS vs greedy CPU reference
+# ---------------------------------------------------------------------------
+
+def _nms_cpu(boxes, scores, iou_thr):
+    order = scores.argsort(descending=True)
+    keep = []
+    suppressed = torch.zeros(len(boxes), dtype=torch.bool)
+    for idx in order.tolist():
+        if suppressed[idx] or scores[idx] <= 0:
+            continue
+        keep.append(idx)
+        b = boxes[idx]
+        ix = (
+            torch.minimum(b[2], boxes[:, 2]) - torch.maximum(b[0], boxes[:, 0])
+        ).clamp(min=0)
+        iy = (
+            torch.minimum(b[3], boxes[:, 3]) - torch.maximum(b[1], boxes[:, 1])
+        ).clamp(min=0)
+        inter = ix * iy
+        area_a = (b[2] - b[0]).clamp(min=0) * (b[3] - b[1]).clamp(min=0)
+        area_b = (boxes[:, 2] - boxes[:, 0]).clamp(min=0) * (
+            boxes[:, 3] - boxes[:, 1]
+        ).clamp(min=0)
+        iou = inter / (area_a + area_b - inter)
+        suppressed |= iou > iou_thr
+        suppressed[idx] = False  # self-iou is 1, keep the anchor
+    return sorted(keep)
+
+
+def test_nms_multiclass_vs_cpu(ext):
+    torch.manual_seed(0)
+    R, C = 300, 25
+    centers = torch.rand(R, 2) * 100
+    wh = torch.rand(R, 2) * 20 + 2
+    boxes = torch.cat([centers - wh / 2, centers + wh / 2], dim=1)
+    scores = torch.rand(R, C)
+    out = torch.ops.vilbert_amd.nms_multiclass(
+        boxes.cuda(), scores.cuda(), 0.5, 0.0
+    ).cpu()
+    for c in range(0, C, 7):
+        keep_ref = _nms_cpu(boxes, scores[:, c], 0.5)
+        keep_gpu = sorted(torch.nonzero(out[:, c] > 0).flatten().tolist())
+        assert keep_gpu == keep_ref, f"class {c}"

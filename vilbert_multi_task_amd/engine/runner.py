@@ -1,0 +1,120 @@
+"""hipGraph-captured forward runner.
+
+Replaces the reference's one-message-at-a-time eager forward
+(/root/reference/worker.py:286-289) with per-(batch-bucket) hipGraph capture:
+the whole two-stream forward (every HIP kernel + hipBLASLt GEMM launch) is
+captured once and replayed per batch, eliminating per-launch host overhead —
+the BASELINE.json north-star serving structure.
+
+On CPU (tests) the runner degrades to eager execution with the same API.
+"""
+
+from __future__ import annotations
+
+from typing import Dict, Optional, Tuple
+
+import torch
+
+from ..data.synthetic import synthetic_batch
+
+
+class GraphRunner:
+    def __init__(
+        self,
+        model: torch.nn.Module,
+        seq_len: int = 37,
+        regions: int = 101,
+        feat_dim: int = 2048,
+        device: str = "cuda",
+        dtype: torch.dtype = torch.bfloat16,
+        use_graphs: bool = True,
+    ):
+        self.model = model.eval()
+        self.seq_len = seq_len
+        self.regions = regions
+        self.feat_dim = feat_dim
+        self.device = device
+        self.dtype = dtype
+        self.use_graphs = use_graphs and device.startswith("cuda")
+        self._graphs: Dict[int, Tuple[torch.cuda.CUDAGraph, dict, tuple]] = {}
+
+    # -- bucket management -------------------------------------------------
+    def bucket_for(self, batch: int) -> int:
+        """Smallest captured power-of-two bucket >= batch."""
+        b = 1
+        while b < batch:
+            b <<= 1
+        return b
+
+    def _static_inputs(self, bucket: int) -> dict:
+        s = synthetic_batch(
+            bucket,
+            seq_len=self.seq_len,
+            regions=self.regions,
+            feat_dim=self.feat_dim,
+            device=self.device,
+        )
+        s["features"] = s["features"].to(self.dtype)
+        s["spatials"] = s["spatials"].to(self.dtype)
+        return s
+
+    def _forward(self, inp: dict):
+        return self.model(
+            inp["question"],
+            inp["features"],
+            inp["spatials"],
+            inp["segment_ids"],
+            inp["input_mask"],
+            inp["image_mask"],
+            None,  # co_attention_mask: zero in the serving path -> zero bias
+            inp["task_tokens"],
+            False,
+        )
+
+    def capture(self, bucket: int) -> None:
+        if not self.use_graphs or bucket in self._graphs:
+            return
+        inp = self._static_inputs(bucket)
+        torch.cuda.synchronize()
+        # warmup on a side stream (required before capture)
+        s = torch.cuda.Stream()
+        s.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(s), torch.no_grad():
+            for _ in range(2):
+                self._forward(inp)
+        torch.cuda.current_stream().wait_stream(s)
+        torch.cuda.synchronize()
+        g = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(g), torch.no_grad():
+            out = self._forward(inp)
+        self._graphs[bucket] = (g, inp, out)
+
+    # -- execution ---------------------------------------------------------
+    @torch.no_grad()
+    def run(self, batch: dict):
+        """batch: dict of serving tensors (synthetic_batch schema). Returns
+        the 10-output tuple truncated/copied to the true batch size."""
+        n = batch["question"].shape[0]
+        if not self.use_graphs:
+            return self._forward({k: v.to(self.device) for k, v in batch.items()})
+        bucket = self.bucket_for(n)
+        if bucket not in self._graphs:
+            self.capture(bucket)
+        g, inp, out = self._graphs[bucket]
+        for key in ("question", "segment_ids", "input_mask", "image_mask", "task_tokens"):
+            dst = inp[key]
+            dst[:n].copy_(batch[key][:n].to(dst.device, non_blocking=True))
+            if n < bucket and key in ("input_mask", "image_mask"):
+                dst[n:].fill_(1)  # pad rows: harmless full-attend
+        inp["features"][:n].copy_(batch["features"][:n].to(self.device, self.dtype, non_blocking=True))
+        inp["spatials"][:n].copy_(batch["spatials"][:n].to(self.device, self.dtype, non_blocking=True))
+        g.replay()
+        res = []
+        for i, o in enumerate(out):
+            if not torch.is_tensor(o):
+                res.append(o)
+            elif i == 3:  # vil_binary_prediction has B/2 pair rows
+                res.append(o[: max(n // 2, 1)])
+            else:
+                res.append(o[:n])
+        return tuple(res)

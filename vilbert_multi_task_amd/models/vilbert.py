@@ -128,21 +128,13 @@ class MultiHeadSelfAttention(nn.Module):
         self.dropout = nn.Dropout(hidden_dropout)
         self.layer_norm = FusedLayerNorm(hidden, eps)
 
-    def _split(self, x: torch.Tensor) -> torch.Tensor:
-        b, l, _ = x.shape
-        return x.view(b, l, self.heads, self.head_dim).transpose(1, 2)
-
     def forward(
         self, x: torch.Tensor, mask_bias: Optional[torch.Tensor], need_probs: bool
     ) -> Tuple[torch.Tensor, Optional[torch.Tensor]]:
-        b, l, h = x.shape
-        q = self._split(self.query(x))
-        k = self._split(self.key(x))
-        v = self._split(self.value(x))
         ctx, probs = F_ops.attention(
-            q, k, v, mask_bias, self.attn_dropout_p, self.training, need_probs
+            self.query(x), self.key(x), self.value(x), self.heads,
+            mask_bias, self.attn_dropout_p, self.training, need_probs,
         )
-        ctx = ctx.transpose(1, 2).reshape(b, l, h)
         y = self.layer_norm(self.dropout(self.out(ctx)), residual=x)
         return y, probs
 
@@ -191,17 +183,11 @@ class CrossAttention(nn.Module):
         self.dropout = nn.Dropout(hidden_dropout)
         self.layer_norm = FusedLayerNorm(q_hidden, eps)
 
-    def _split(self, x: torch.Tensor) -> torch.Tensor:
-        b, l, _ = x.shape
-        return x.view(b, l, self.heads, self.head_dim).transpose(1, 2)
-
     def forward(self, x_q, x_kv, mask_bias, need_probs=False):
-        b, lq, _ = x_q.shape
-        q = self._split(self.query(x_q))
-        k = self._split(self.key(x_kv))
-        v = self._split(self.value(x_kv))
-        ctx, probs = F_ops.attention(q, k, v, mask_bias, self.attn_dropout_p, self.training, need_probs)
-        ctx = ctx.transpose(1, 2).reshape(b, lq, self.heads * self.head_dim)
+        ctx, probs = F_ops.attention(
+            self.query(x_q), self.key(x_kv), self.value(x_kv), self.heads,
+            mask_bias, self.attn_dropout_p, self.training, need_probs,
+        )
         y = self.layer_norm(self.dropout(self.out(ctx)), residual=x_q)
         return y, probs
 
@@ -317,7 +303,10 @@ class ViLBertModel(nn.Module):
         v_mask = self._extend_mask(image_mask, dtype)
 
         co_tv = co_vt = None
-        if co_attention_mask is not None and co_attention_mask.any():
+        # NOTE: no `.any()` here — that is a device->host sync and breaks
+        # hipGraph capture. A zero mask yields a zero bias (identical math);
+        # the serving runner passes None when the mask is known-zero.
+        if co_attention_mask is not None:
             # co_attention_mask [B, R, T] (worker.py:455): restricts
             # vision<->text pairs. 1 = masked out.
             cm = co_attention_mask.to(dtype)

@@ -1,0 +1,191 @@
+// Torch operator registration for the gfx950 kernels.
+// Pure C++/HIP (no Python.h): ops register under torch.ops.vilbert_amd and
+// the .so is loaded in-tree via torch.ops.load_library (ops/hip_ext.py).
+
+#include <ATen/ATen.h>
+#include <c10/hip/HIPStream.h>
+#include <hip/hip_runtime.h>
+#include <torch/library.h>
+
+#include <hip/hip_bf16.h>
+
+using bf16 = __hip_bfloat16;
+
+// launchers (defined in the .hip files)
+template <typename T>
+void launch_residual_ln(const T*, const T*, const T*, const T*, T*, long, int,
+                        float, hipStream_t);
+template <typename T>
+void launch_bias_gelu(const T*, const T*, T*, long, int, hipStream_t);
+template <typename T>
+void launch_embedding_ln(const long*, const long*, const long*, const T*,
+                         const T*, const T*, const T*, const T*, T*, long, int,
+                         float, hipStream_t);
+void launch_attention(const bf16*, const bf16*, const bf16*, const bf16*, bf16*,
+                      int, int, int, int, int, int, hipStream_t);
+void launch_mfma_probe(const bf16*, const bf16*, float*, hipStream_t);
+void launch_nms_multiclass(const float*, const float*, const long*, float*, int,
+                           int, float, float, hipStream_t);
+
+namespace {
+
+hipStream_t cur_stream() { return c10::hip::getCurrentHIPStream().stream(); }
+
+void check_same(const at::Tensor& a, const at::Tensor& b, const char* msg) {
+  TORCH_CHECK(a.scalar_type() == b.scalar_type(), msg, ": dtype mismatch");
+  TORCH_CHECK(a.is_cuda() && b.is_cuda(), msg, ": tensors must be on GPU");
+}
+
+at::Tensor residual_layer_norm(const at::Tensor& x,
+                               const c10::optional<at::Tensor>& res,
+                               const at::Tensor& w, const at::Tensor& b,
+                               double eps) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous(), "residual_layer_norm: x");
+  const int dim = (int)x.size(-1);
+  const long rows = x.numel() / dim;
+  auto y = at::empty_like(x);
+  const bool hr = res.has_value();
+  if (hr) TORCH_CHECK(res->is_contiguous() && res->sizes() == x.sizes());
+  if (x.scalar_type() == at::kBFloat16) {
+    launch_residual_ln<bf16>(
+        (const bf16*)x.data_ptr(), hr ? (const bf16*)res->data_ptr() : nullptr,
+        (const bf16*)w.data_ptr(), (const bf16*)b.data_ptr(),
+        (bf16*)y.data_ptr(), rows, dim, (float)eps, cur_stream());
+  } else if (x.scalar_type() == at::kFloat) {
+    launch_residual_ln<float>(
+        x.data_ptr<float>(), hr ? res->data_ptr<float>() : nullptr,
+        w.data_ptr<float>(), b.data_ptr<float>(),
+        y.data_ptr<float>(), rows, dim, (float)eps, cur_stream());
+  } else {
+    TORCH_CHECK(false, "residual_layer_norm: dtype must be bf16/f32");
+  }
+  return y;
+}
+
+at::Tensor bias_gelu(const at::Tensor& x, const c10::optional<at::Tensor>& bias) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous(), "bias_gelu: x");
+  const int dim = (int)x.size(-1);
+  const long n = x.numel();
+  auto y = at::empty_like(x);
+  const bool hb = bias.has_value();
+  if (x.scalar_type() == at::kBFloat16) {
+    launch_bias_gelu<bf16>((const bf16*)x.data_ptr(),
+                           hb ? (const bf16*)bias->data_ptr() : nullptr,
+                           (bf16*)y.data_ptr(), n, dim, cur_stream());
+  } else if (x.scalar_type() == at::kFloat) {
+    launch_bias_gelu<float>(x.data_ptr<float>(),
+                            hb ? bias->data_ptr<float>() : nullptr,
+                            y.data_ptr<float>(), n, dim, cur_stream());
+  } else {
+    TORCH_CHECK(false, "bias_gelu: dtype must be bf16/f32");
+  }
+  return y;
+}
+
+at::Tensor attention(const at::Tensor& q, const at::Tensor& k,
+                     const at::Tensor& v, int64_t heads,
+                     const c10::optional<at::Tensor>& mask) {
+  TORCH_CHECK(q.is_cuda() && q.is_contiguous() && k.is_contiguous() && v.is_contiguous());
+  TORCH_CHECK(q.scalar_type() == at::kBFloat16, "attention: bf16 only");
+  TORCH_CHECK(q.dim() == 3 && k.dim() == 3 && v.dim() == 3, "attention: [B,L,H*D]");
+  const int B = (int)q.size(0), Lq = (int)q.size(1), HD = (int)q.size(2);
+  const int Lk = (int)k.size(1);
+  const int H = (int)heads;
+  const int D = HD / H;
+  TORCH_CHECK(HD % H == 0 && (D == 64 || D == 128), "attention: head_dim must be 64/128");
+  TORCH_CHECK(Lq <= 128 && Lk <= 128, "attention: serving kernel handles L<=128");
+  int mask_mode = 0;
+  const bf16* mptr = nullptr;
+  if (mask.has_value() && mask->defined()) {
+    auto& m = *mask;
+    TORCH_CHECK(m.scalar_type() == at::kBFloat16, "attention: mask must be bf16");
+    TORCH_CHECK(m.is_contiguous());
+    const long mn = m.numel();
+    if (mn == (long)B * Lk) mask_mode = 1;
+    else if (mn == (long)B * Lq * Lk) mask_mode = 2;
+    else TORCH_CHECK(false, "attention: mask numel must be B*Lk or B*Lq*Lk");
+    mptr = (const bf16*)m.data_ptr();
+  }
+  auto out = at::empty_like(q);
+  launch_attention((const bf16*)q.data_ptr(), (const bf16*)k.data_ptr(),
+                   (const bf16*)v.data_ptr(), mptr,
+                   (bf16*)out.data_ptr(), B, H, Lq, Lk, D, mask_mode,
+                   cur_stream());
+  return out;
+}
+
+at::Tensor embedding_ln(const at::Tensor& ids, const at::Tensor& pos_ids,
+                        const at::Tensor& type_ids, const at::Tensor& word_w,
+                        const at::Tensor& pos_w, const at::Tensor& type_w,
+                        const at::Tensor& ln_w, const at::Tensor& ln_b,
+                        double eps) {
+  TORCH_CHECK(ids.is_cuda() && ids.scalar_type() == at::kLong);
+  const int dim = (int)word_w.size(1);
+  const long tokens = ids.numel();
+  auto y = at::empty({ids.size(0), ids.size(1), (long)dim}, word_w.options());
+  if (word_w.scalar_type() == at::kBFloat16) {
+    launch_embedding_ln<bf16>(
+        ids.data_ptr<long>(), pos_ids.data_ptr<long>(),
+        type_ids.data_ptr<long>(), (const bf16*)word_w.data_ptr(),
+        (const bf16*)pos_w.data_ptr(), (const bf16*)type_w.data_ptr(),
+        (const bf16*)ln_w.data_ptr(), (const bf16*)ln_b.data_ptr(),
+        (bf16*)y.data_ptr(), tokens, dim, (float)eps, cur_stream());
+  } else if (word_w.scalar_type() == at::kFloat) {
+    launch_embedding_ln<float>(
+        ids.data_ptr<long>(), pos_ids.data_ptr<long>(),
+        type_ids.data_ptr<long>(), word_w.data_ptr<float>(),
+        pos_w.data_ptr<float>(), type_w.data_ptr<float>(),
+        ln_w.data_ptr<float>(), ln_b.data_ptr<float>(),
+        y.data_ptr<float>(), tokens, dim, (float)eps, cur_stream());
+  } else {
+    TORCH_CHECK(false, "embedding_ln: dtype must be bf16/f32");
+  }
+  return y;
+}
+
+at::Tensor mfma_probe(const at::Tensor& a, const at::Tensor& b) {
+  TORCH_CHECK(a.is_cuda() && a.scalar_type() == at::kBFloat16);
+  TORCH_CHECK(a.sizes() == at::IntArrayRef({16, 32}) &&
+              b.sizes() == at::IntArrayRef({32, 16}));
+  auto c = at::empty({16, 16}, a.options().dtype(at::kFloat));
+  launch_mfma_probe((const bf16*)a.contiguous().data_ptr(),
+                    (const bf16*)b.contiguous().data_ptr(),
+                    c.data_ptr<float>(), cur_stream());
+  return c;
+}
+
+at::Tensor nms_multiclass(const at::Tensor& boxes, const at::Tensor& scores,
+                          double iou_thr, double score_thr) {
+  TORCH_CHECK(boxes.is_cuda() && boxes.scalar_type() == at::kFloat);
+  TORCH_CHECK(scores.is_cuda() && scores.scalar_type() == at::kFloat);
+  const int R = (int)boxes.size(0);
+  const int C = (int)scores.size(1);
+  TORCH_CHECK(R <= 1024, "nms_multiclass: R <= 1024");
+  auto order = std::get<1>(scores.sort(0, /*descending=*/true)).contiguous();
+  auto out = at::zeros_like(scores);
+  launch_nms_multiclass(boxes.contiguous().data_ptr<float>(),
+                        scores.contiguous().data_ptr<float>(),
+                        order.data_ptr<long>(), out.data_ptr<float>(),
+                        R, C, (float)iou_thr, (float)score_thr, cur_stream());
+  return out;
+}
+
+}  // namespace
+
+TORCH_LIBRARY(vilbert_amd, m) {
+  m.def("residual_layer_norm(Tensor x, Tensor? res, Tensor w, Tensor b, float eps) -> Tensor");
+  m.def("bias_gelu(Tensor x, Tensor? bias) -> Tensor");
+  m.def("attention(Tensor q, Tensor k, Tensor v, int heads, Tensor? mask) -> Tensor");
+  m.def("embedding_ln(Tensor ids, Tensor pos, Tensor type, Tensor word_w, Tensor pos_w, Tensor type_w, Tensor ln_w, Tensor ln_b, float eps) -> Tensor");
+  m.def("mfma_probe(Tensor a, Tensor b) -> Tensor");
+  m.def("nms_multiclass(Tensor boxes, Tensor scores, float iou_thr, float score_thr) -> Tensor");
+}
+
+TORCH_LIBRARY_IMPL(vilbert_amd, CUDA, m) {
+  m.impl("residual_layer_norm", residual_layer_norm);
+  m.impl("bias_gelu", bias_gelu);
+  m.impl("attention", attention);
+  m.impl("embedding_ln", embedding_ln);
+  m.impl("mfma_probe", mfma_probe);
+  m.impl("nms_multiclass", nms_multiclass);
+}

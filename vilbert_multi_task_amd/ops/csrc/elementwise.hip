@@ -1,0 +1,317 @@
+// Fused memory-bound kernels: residual+LayerNorm, bias+GELU, embedding+LN.
+//
+// These replace the implicit CUDA elementwise/normalization kernels of the
+// reference's PyTorch forward (SURVEY.md §2.3 "Implicit CUDA kernels") with
+// hand-written CDNA4 code: one wave per row, bf16 loads vectorized 8-wide
+// (16 B/lane — cdna_hip_programming.md Guideline 13: scalar bf16 loads are
+// ~2-2.5x slower), f32 accumulation, fused residual add so the tensor is
+// read once from HBM instead of twice.
+
+#include "common.h"
+
+// ---------------------------------------------------------------------------
+// residual + LayerNorm.  One wave per row; rows assigned grid-stride.
+// dim % 8 == 0 fast path (uint4 = 8 bf16); scalar fallback otherwise.
+// ---------------------------------------------------------------------------
+
+template <typename T>
+struct VecIO;
+
+template <>
+struct VecIO<bf16> {
+  // 8 elements = 16 bytes
+  DEV static void load8(const bf16* p, float* out) {
+    const uint4 raw = *reinterpret_cast<const uint4*>(p);
+    const unsigned int w[4] = {raw.x, raw.y, raw.z, raw.w};
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      out[2 * i] = us2f((unsigned short)(w[i] & 0xffff));
+      out[2 * i + 1] = us2f((unsigned short)(w[i] >> 16));
+    }
+  }
+  DEV static void store8(bf16* p, const float* in) {
+    uint4 raw;
+    unsigned int w[4];
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      w[i] = (unsigned int)f2us(in[2 * i]) | ((unsigned int)f2us(in[2 * i + 1]) << 16);
+    }
+    raw.x = w[0]; raw.y = w[1]; raw.z = w[2]; raw.w = w[3];
+    *reinterpret_cast<uint4*>(p) = raw;
+  }
+  DEV static float ld(const bf16* p) { return bf2f(*p); }
+  DEV static void st(bf16* p, float v) { *p = f2bf(v); }
+};
+
+template <>
+struct VecIO<float> {
+  DEV static void load8(const float* p, float* out) {
+    const float4 a = *reinterpret_cast<const float4*>(p);
+    const float4 b = *reinterpret_cast<const float4*>(p + 4);
+    out[0] = a.x; out[1] = a.y; out[2] = a.z; out[3] = a.w;
+    out[4] = b.x; out[5] = b.y; out[6] = b.z; out[7] = b.w;
+  }
+  DEV static void store8(float* p, const float* in) {
+    *reinterpret_cast<float4*>(p) = make_float4(in[0], in[1], in[2], in[3]);
+    *reinterpret_cast<float4*>(p + 4) = make_float4(in[4], in[5], in[6], in[7]);
+  }
+  DEV static float ld(const float* p) { return *p; }
+  DEV static void st(float* p, float v) { *p = v; }
+};
+
+// MAX_CHUNKS*8*64 = max dim cached in registers per wave (4 -> 2048; dims
+// beyond that re-read from L2 in the normalize pass).
+#define LN_MAX_CHUNKS 4
+
+template <typename T, bool HAS_RES>
+__global__ void residual_ln_kernel(const T* __restrict__ x,
+                                   const T* __restrict__ res,
+                                   const T* __restrict__ w,
+                                   const T* __restrict__ b,
+                                   T* __restrict__ y,
+                                   long rows, int dim, float eps) {
+  const int lane = lane_id();
+  const int wid = wave_id();
+  const int waves_per_blk = blockDim.x / WAVE;
+  const bool vec8 = (dim % 8) == 0;
+  const int nchunk = vec8 ? ceil_div(dim / 8, WAVE) : 0;
+  const bool cached = vec8 && nchunk <= LN_MAX_CHUNKS;
+  float cache[LN_MAX_CHUNKS * 8];
+
+  for (long row = (long)blockIdx.x * waves_per_blk + wid; row < rows;
+       row += (long)gridDim.x * waves_per_blk) {
+    const T* xr = x + row * dim;
+    const T* rr = HAS_RES ? res + row * dim : nullptr;
+    T* yr = y + row * dim;
+    float s = 0.f, sq = 0.f;
+    if (cached) {
+#pragma unroll
+      for (int c = 0; c < LN_MAX_CHUNKS; ++c) {
+        if (c >= nchunk) break;
+        const int j = (c * WAVE + lane) * 8;
+        if (j < dim) {
+          VecIO<T>::load8(xr + j, &cache[c * 8]);
+          if (HAS_RES) {
+            float r8[8];
+            VecIO<T>::load8(rr + j, r8);
+#pragma unroll
+            for (int i = 0; i < 8; ++i) cache[c * 8 + i] += r8[i];
+          }
+#pragma unroll
+          for (int i = 0; i < 8; ++i) {
+            s += cache[c * 8 + i];
+            sq += cache[c * 8 + i] * cache[c * 8 + i];
+          }
+        }
+      }
+    } else {
+      for (int j = lane; j < dim; j += WAVE) {
+        float v = VecIO<T>::ld(xr + j);
+        if (HAS_RES) v += VecIO<T>::ld(rr + j);
+        s += v;
+        sq += v * v;
+      }
+    }
+    s = wave_sum(s);
+    sq = wave_sum(sq);
+    const float mean = s / dim;
+    const float var = sq / dim - mean * mean;
+    const float rstd = rsqrtf(var + eps);
+    if (cached) {
+#pragma unroll
+      for (int c = 0; c < LN_MAX_CHUNKS; ++c) {
+        if (c >= nchunk) break;
+        const int j = (c * WAVE + lane) * 8;
+        if (j < dim) {
+          float w8[8], b8[8], o8[8];
+          VecIO<T>::load8(w + j, w8);
+          VecIO<T>::load8(b + j, b8);
+#pragma unroll
+          for (int i = 0; i < 8; ++i)
+            o8[i] = (cache[c * 8 + i] - mean) * rstd * w8[i] + b8[i];
+          VecIO<T>::store8(yr + j, o8);
+        }
+      }
+    } else {
+      for (int j = lane; j < dim; j += WAVE) {
+        float v = VecIO<T>::ld(xr + j);
+        if (HAS_RES) v += VecIO<T>::ld(rr + j);
+        VecIO<T>::st(yr + j,
+                     (v - mean) * rstd * VecIO<T>::ld(w + j) + VecIO<T>::ld(b + j));
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// bias + exact GELU:  y = 0.5*(x+b)*(1+erf((x+b)/sqrt(2)))
+// ---------------------------------------------------------------------------
+
+template <typename T, bool HAS_BIAS>
+__global__ void bias_gelu_kernel(const T* __restrict__ x,
+                                 const T* __restrict__ bias,
+                                 T* __restrict__ y,
+                                 long n, int dim) {
+  const long i8 = ((long)blockIdx.x * blockDim.x + threadIdx.x) * 8;
+  const long stride8 = (long)gridDim.x * blockDim.x * 8;
+  constexpr float kInvSqrt2 = 0.70710678118654752440f;
+  for (long i = i8; i < n; i += stride8) {
+    if (i + 8 <= n && (!HAS_BIAS || ((i % dim) + 8 <= dim))) {
+      float v[8];
+      VecIO<T>::load8(x + i, v);
+      if (HAS_BIAS) {
+        float b8[8];
+        VecIO<T>::load8(bias + (i % dim), b8);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) v[j] += b8[j];
+      }
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        v[j] = 0.5f * v[j] * (1.0f + erff(v[j] * kInvSqrt2));
+      VecIO<T>::store8(y + i, v);
+    } else {
+      for (long k = i; k < min(i + 8, n); ++k) {
+        float v = VecIO<T>::ld(x + k);
+        if (HAS_BIAS) v += VecIO<T>::ld(bias + (k % dim));
+        VecIO<T>::st(y + k, 0.5f * v * (1.0f + erff(v * kInvSqrt2)));
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// word + position + segment embedding gather, sum, LayerNorm.
+// One wave per token (replaces three gathers + add + LN round trips).
+// ---------------------------------------------------------------------------
+
+template <typename T>
+__global__ void embedding_ln_kernel(const long* __restrict__ ids,
+                                    const long* __restrict__ pos_ids,
+                                    const long* __restrict__ type_ids,
+                                    const T* __restrict__ word_w,
+                                    const T* __restrict__ pos_w,
+                                    const T* __restrict__ type_w,
+                                    const T* __restrict__ ln_w,
+                                    const T* __restrict__ ln_b,
+                                    T* __restrict__ y,
+                                    long tokens, int dim, float eps) {
+  const int lane = lane_id();
+  const int wid = wave_id();
+  const int waves_per_blk = blockDim.x / WAVE;
+  const bool vec8 = (dim % 8) == 0;
+  const int nchunk = vec8 ? ceil_div(dim / 8, WAVE) : 0;
+  const bool cached = vec8 && nchunk <= LN_MAX_CHUNKS;
+  float cache[LN_MAX_CHUNKS * 8];
+
+  for (long tok = (long)blockIdx.x * waves_per_blk + wid; tok < tokens;
+       tok += (long)gridDim.x * waves_per_blk) {
+    const T* wrow = word_w + (long)ids[tok] * dim;
+    const T* prow = pos_w + (long)pos_ids[tok] * dim;
+    const T* trow = type_w + (long)type_ids[tok] * dim;
+    T* yr = y + tok * dim;
+    float s = 0.f, sq = 0.f;
+    if (cached) {
+#pragma unroll
+      for (int c = 0; c < LN_MAX_CHUNKS; ++c) {
+        if (c >= nchunk) break;
+        const int j = (c * WAVE + lane) * 8;
+        if (j < dim) {
+          float a[8], b8[8], c8[8];
+          VecIO<T>::load8(wrow + j, a);
+          VecIO<T>::load8(prow + j, b8);
+          VecIO<T>::load8(trow + j, c8);
+#pragma unroll
+          for (int i = 0; i < 8; ++i) {
+            cache[c * 8 + i] = a[i] + b8[i] + c8[i];
+            s += cache[c * 8 + i];
+            sq += cache[c * 8 + i] * cache[c * 8 + i];
+          }
+        }
+      }
+    } else {
+      for (int j = lane; j < dim; j += WAVE) {
+        float v = VecIO<T>::ld(wrow + j) + VecIO<T>::ld(prow + j) + VecIO<T>::ld(trow + j);
+        s += v;
+        sq += v * v;
+      }
+    }
+    s = wave_sum(s);
+    sq = wave_sum(sq);
+    const float mean = s / dim;
+    const float rstd = rsqrtf(sq / dim - mean * mean + eps);
+    if (cached) {
+#pragma unroll
+      for (int c = 0; c < LN_MAX_CHUNKS; ++c) {
+        if (c >= nchunk) break;
+        const int j = (c * WAVE + lane) * 8;
+        if (j < dim) {
+          float w8[8], b8[8], o8[8];
+          VecIO<T>::load8(ln_w + j, w8);
+          VecIO<T>::load8(ln_b + j, b8);
+#pragma unroll
+          for (int i = 0; i < 8; ++i)
+            o8[i] = (cache[c * 8 + i] - mean) * rstd * w8[i] + b8[i];
+          VecIO<T>::store8(yr + j, o8);
+        }
+      }
+    } else {
+      for (int j = lane; j < dim; j += WAVE) {
+        float v = VecIO<T>::ld(wrow + j) + VecIO<T>::ld(prow + j) + VecIO<T>::ld(trow + j);
+        VecIO<T>::st(yr + j, (v - mean) * rstd * VecIO<T>::ld(ln_w + j) + VecIO<T>::ld(ln_b + j));
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// host-side launchers (called from bindings.cpp)
+// ---------------------------------------------------------------------------
+
+template <typename T>
+void launch_residual_ln(const T* x, const T* res, const T* w, const T* b, T* y,
+                        long rows, int dim, float eps, hipStream_t stream) {
+  const int block = 256;
+  const int waves = block / WAVE;
+  const int grid = (int)min((rows + waves - 1) / waves, (long)2048);
+  if (res)
+    hipLaunchKernelGGL((residual_ln_kernel<T, true>), dim3(grid), dim3(block), 0,
+                       stream, x, res, w, b, y, rows, dim, eps);
+  else
+    hipLaunchKernelGGL((residual_ln_kernel<T, false>), dim3(grid), dim3(block), 0,
+                       stream, x, res, w, b, y, rows, dim, eps);
+}
+
+template <typename T>
+void launch_bias_gelu(const T* x, const T* bias, T* y, long n, int dim,
+                      hipStream_t stream) {
+  const int block = 256;
+  const long want = (n + 8 * block - 1) / (8 * block);
+  const int grid = (int)min(want, (long)2048);
+  if (bias)
+    hipLaunchKernelGGL((bias_gelu_kernel<T, true>), dim3(grid), dim3(block), 0,
+                       stream, x, bias, y, n, dim);
+  else
+    hipLaunchKernelGGL((bias_gelu_kernel<T, false>), dim3(grid), dim3(block), 0,
+                       stream, x, bias, y, n, dim);
+}
+
+template <typename T>
+void launch_embedding_ln(const long* ids, const long* pos_ids, const long* type_ids,
+                         const T* word_w, const T* pos_w, const T* type_w,
+                         const T* ln_w, const T* ln_b, T* y, long tokens, int dim,
+                         float eps, hipStream_t stream) {
+  const int block = 256;
+  const int waves = block / WAVE;
+  const int grid = (int)min((tokens + waves - 1) / waves, (long)2048);
+  hipLaunchKernelGGL((embedding_ln_kernel<T>), dim3(grid), dim3(block), 0, stream,
+                     ids, pos_ids, type_ids, word_w, pos_w, type_w, ln_w, ln_b, y,
+                     tokens, dim, eps);
+}
+
+// explicit instantiations
+template void launch_residual_ln<float>(const float*, const float*, const float*, const float*, float*, long, int, float, hipStream_t);
+template void launch_residual_ln<bf16>(const bf16*, const bf16*, const bf16*, const bf16*, bf16*, long, int, float, hipStream_t);
+template void launch_bias_gelu<float>(const float*, const float*, float*, long, int, hipStream_t);
+template void launch_bias_gelu<bf16>(const bf16*, const bf16*, bf16*, long, int, hipStream_t);
+template void launch_embedding_ln<float>(const long*, const long*, const long*, const float*, const float*, const float*, const float*, const float*, float*, long, int, float, hipStream_t);
+template void launch_embedding_ln<bf16>(const long*, const long*, const long*, const bf16*, const bf16*, const bf16*, const bf16*, const bf16*, bf16*, long, int, float, hipStream_t);

@@ -2,12 +2,14 @@
 extension.
 
 Rules of the build (BASELINE.json north star):
-- On a GPU box the hand-written gfx950 kernels ARE the compute path; if the
-  extension is missing on a CUDA/HIP device we fail loudly instead of
-  silently falling back to eager PyTorch (the driver checks which .so the
-  GPU tests actually loaded).
+- On a GPU box the hand-written gfx950 kernels ARE the inference compute
+  path; if the extension is missing on a CUDA/HIP device we fail loudly
+  instead of silently falling back to eager PyTorch (the driver checks which
+  .so the GPU tests actually loaded).
 - On CPU the plain fp32 PyTorch implementations below are the numerics
   oracle each HIP kernel is tested against (SURVEY.md §4 consequence (1)).
+- Autograd paths (training) use the PyTorch ops; the HIP kernels are
+  forward/serving kernels (hipGraph-capturable, no sync, no alloc churn).
 
 Set VILBERT_AMD_FORCE_EAGER=1 to force the PyTorch path on GPU (debug only).
 """
@@ -29,7 +31,7 @@ def _load_extension():
     if _EXT is not None or _EXT_ERR is not None:
         return _EXT
     try:
-        from . import hip_ext  # built in-tree by setup/build (vilbert_hip.so)
+        from . import hip_ext  # built in-tree (ops/_C/vilbert_hip.so)
 
         _EXT = hip_ext.load()
     except Exception as e:  # pragma: no cover - exercised on GPU box only
@@ -42,11 +44,15 @@ def extension_available() -> bool:
     return _load_extension() is not None
 
 
-def _want_hip(x: torch.Tensor) -> bool:
+def _want_hip(*tensors: torch.Tensor) -> bool:
+    """True iff the HIP path should run. Raises on GPU without the ext."""
+    x = tensors[0]
     if not x.is_cuda:
         return False
     if os.environ.get("VILBERT_AMD_FORCE_EAGER") == "1":
         return False
+    if torch.is_grad_enabled() and any(t is not None and t.requires_grad for t in tensors):
+        return False  # training path -> autograd-capable torch ops
     ext = _load_extension()
     if ext is None:
         raise RuntimeError(
@@ -70,12 +76,16 @@ def layer_norm(
     residual: Optional[torch.Tensor] = None,
 ) -> torch.Tensor:
     """y = LayerNorm(x + residual) over the last dim."""
-    if _want_hip(x) and not (torch.is_grad_enabled() and x.requires_grad):
+    if _want_hip(x, residual):
         ext = _load_extension()
-        return ext.residual_layer_norm(x, residual, weight, bias, eps)
+        return ext.residual_layer_norm(x.contiguous(), _opt(residual), weight, bias, eps)
     if residual is not None:
         x = x + residual
     return torch.nn.functional.layer_norm(x, (x.shape[-1],), weight, bias, eps)
+
+
+def _opt(t: Optional[torch.Tensor]) -> Optional[torch.Tensor]:
+    return t.contiguous() if t is not None else None
 
 
 # --------------------------------------------------------------------------
@@ -84,50 +94,55 @@ def layer_norm(
 
 def bias_gelu(x: torch.Tensor, bias: Optional[torch.Tensor]) -> torch.Tensor:
     """y = gelu(x + bias) — exact (erf) GELU, matching BERT."""
-    if _want_hip(x) and not (torch.is_grad_enabled() and x.requires_grad):
+    if _want_hip(x, bias):
         ext = _load_extension()
-        return ext.bias_gelu(x, bias)
+        return ext.bias_gelu(x.contiguous(), _opt(bias))
     if bias is not None:
         x = x + bias
     return torch.nn.functional.gelu(x)
 
 
-def bias_tanh(x: torch.Tensor, bias: Optional[torch.Tensor]) -> torch.Tensor:
-    if bias is not None:
-        x = x + bias
-    return torch.tanh(x)
-
-
 # --------------------------------------------------------------------------
-# Scaled-dot-product attention with additive mask
+# Multi-head scaled-dot-product attention with additive mask.
+# Flattened [B, L, H*D] layout so the GPU path needs no transpose copies.
 # --------------------------------------------------------------------------
 
 def attention(
     q: torch.Tensor,
     k: torch.Tensor,
     v: torch.Tensor,
+    num_heads: int,
     mask_bias: Optional[torch.Tensor],
     dropout_p: float = 0.0,
     training: bool = False,
     need_probs: bool = False,
 ) -> Tuple[torch.Tensor, Optional[torch.Tensor]]:
-    """Multi-head SDPA.
-
-    q: [B,H,Lq,D], k/v: [B,H,Lk,D]; mask_bias additive [B,1,1,Lk] or
-    [B,1,Lq,Lk] (0 for keep, -inf-ish for masked). Returns (ctx [B,H,Lq,D],
-    probs or None).
-    """
-    if (
-        _want_hip(q)
+    """q: [B,Lq,H*D], k/v: [B,Lk,H*D]; mask_bias additive [B,1,1,Lk] or
+    [B,1,Lq,Lk] (0 keep, large-negative masked). Returns
+    (ctx [B,Lq,H*D], probs [B,H,Lq,Lk] or None)."""
+    use_hip = (
+        _want_hip(q, k, v)
         and not need_probs
         and not (training and dropout_p > 0.0)
-        and not (torch.is_grad_enabled() and (q.requires_grad or k.requires_grad or v.requires_grad))
-    ):
+    )
+    if use_hip:
         ext = _load_extension()
-        return ext.attention(q, k, v, mask_bias), None
+        mb = mask_bias
+        if mb is not None:
+            mb = mb.contiguous()
+        return (
+            ext.attention(q.contiguous(), k.contiguous(), v.contiguous(), num_heads, mb),
+            None,
+        )
 
-    scale = 1.0 / math.sqrt(q.shape[-1])
-    scores = torch.matmul(q, k.transpose(-1, -2)) * scale
+    b, lq, hd = q.shape
+    lk = k.shape[1]
+    d = hd // num_heads
+    qh = q.view(b, lq, num_heads, d).transpose(1, 2)
+    kh = k.view(b, lk, num_heads, d).transpose(1, 2)
+    vh = v.view(b, lk, num_heads, d).transpose(1, 2)
+    scale = 1.0 / math.sqrt(d)
+    scores = torch.matmul(qh, kh.transpose(-1, -2)) * scale
     if mask_bias is not None:
         scores = scores + mask_bias
     probs = torch.softmax(scores, dim=-1)
@@ -135,7 +150,8 @@ def attention(
         probs_d = torch.nn.functional.dropout(probs, p=dropout_p, training=True)
     else:
         probs_d = probs
-    ctx = torch.matmul(probs_d, v)
+    ctx = torch.matmul(probs_d, vh)
+    ctx = ctx.transpose(1, 2).reshape(b, lq, hd)
     return ctx, (probs if need_probs else None)
 
 
@@ -154,9 +170,12 @@ def embedding_ln(
     ln_b: torch.Tensor,
     eps: float,
 ) -> torch.Tensor:
-    if _want_hip(word_w) and ids.is_cuda and not torch.is_grad_enabled():
+    if word_w.is_cuda and not torch.is_grad_enabled() and _want_hip(word_w):
         ext = _load_extension()
-        return ext.embedding_ln(ids, pos_ids, type_ids, word_w, pos_w, type_w, ln_w, ln_b, eps)
+        return ext.embedding_ln(
+            ids.contiguous(), pos_ids.contiguous(), type_ids.contiguous(),
+            word_w, pos_w, type_w, ln_w, ln_b, eps,
+        )
     e = (
         torch.nn.functional.embedding(ids, word_w)
         + torch.nn.functional.embedding(pos_ids, pos_w)
