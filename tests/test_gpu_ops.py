@@ -86,7 +86,9 @@ def test_bias_gelu(ext):
     bias = _rand_bf16(3072, seed=8)
     y = torch.ops.vilbert_amd.bias_gelu(x, bias)
     ref = torch.nn.functional.gelu(x.float() + bias.float())
-    assert (y.float() - ref).abs().max() < 2e-2
+    # bf16 output ulp is |v|/128; allow rounding of large activations
+    err = (y.float() - ref).abs() - ref.abs() / 128
+    assert err.max() < 2e-2
 
 
 def test_gelu_no_bias_odd_tail(ext):

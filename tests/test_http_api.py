@@ -1,0 +1,80 @@
+"""HTTP/WS API contract tests (urls fixed by demo/urls.py:7-11)."""
+
+import io
+import json
+
+import pytest
+
+fastapi = pytest.importorskip("fastapi")
+from fastapi.testclient import TestClient
+
+from vilbert_multi_task_amd.serve.app import create_app
+
+
+@pytest.fixture
+def client(tmp_path):
+    app = create_app(
+        db_path=str(tmp_path / "db.sqlite3"),
+        queue_path=str(tmp_path / "q.sqlite3"),
+        media_root=str(tmp_path / "media"),
+        hub_port=0,
+    )
+    with TestClient(app) as c:
+        yield c, app
+
+
+def test_index(client):
+    c, _ = client
+    r = c.get("/")
+    assert r.status_code == 200 and "ViLBERT" in r.text
+
+
+def test_submit_enqueues_message(client):
+    c, app = client
+    r = c.post(
+        "/",
+        data={
+            "socket_id": "sockA",
+            "task_id": "1",
+            "question": "What COLOR is the sky?",
+            "image_list[]": ["demo/x.jpg"],
+        },
+    )
+    assert r.status_code == 200
+    assert app.state.broker.depth() == 1
+    d = app.state.broker.get()[0]
+    # message schema fixed by sender.py:19-24
+    assert set(d.body) == {"image_path", "question", "socket_id", "task_id"}
+    assert d.body["question"] == "what color is the sky?"  # lowercased (views.py:28)
+    assert d.body["task_id"] == "1"
+    assert d.body["socket_id"] == "sockA"
+
+
+def test_get_task_details(client):
+    c, _ = client
+    r = c.get("/get_task_details/1/")
+    assert r.status_code == 200
+    body = r.json()
+    assert body["unique_id"] == 1 and body["name"] == "VQA"
+    assert c.get("/get_task_details/99/").status_code == 404
+
+
+def test_upload_image(client):
+    c, _ = client
+    files = {"file": ("cat.jpg", io.BytesIO(b"\xff\xd8fakejpeg"), "image/jpeg")}
+    r = c.post("/upload_image/", files=files)
+    assert r.status_code == 200
+    paths = r.json()["file_paths"]
+    assert len(paths) == 1 and paths[0].endswith(".jpg")
+
+
+def test_websocket_push_roundtrip(client):
+    c, app = client
+    with c.websocket_connect("/chat/") as ws:
+        ws.send_text("sockWS")
+        import time
+
+        time.sleep(0.05)
+        app.state.push.publish("sockWS", {"terminal": "hello"})
+        msg = json.loads(ws.receive_text())
+        assert msg == {"terminal": "hello"}

@@ -1,0 +1,160 @@
+"""Golden-output tests of the serving contract (SURVEY.md §4 consequence (2)):
+queue message in -> websocket messages + DB rows out, per-task result schemas
+from worker.py:574-645."""
+
+import json
+
+import pytest
+import torch
+
+from vilbert_multi_task_amd.config import ViLBertConfig
+from vilbert_multi_task_amd.engine.runner import GraphRunner
+from vilbert_multi_task_amd.models import VILBertForVLTasks
+from vilbert_multi_task_amd.serve.broker import Broker, vilbert_task
+from vilbert_multi_task_amd.serve.db import Database
+from vilbert_multi_task_amd.serve.features import SyntheticFeatureProvider
+from vilbert_multi_task_amd.serve.push import NullPush
+from vilbert_multi_task_amd.serve.worker import ServingWorker
+from vilbert_multi_task_amd.data.tokenizer import BertWordPieceTokenizer
+from vilbert_multi_task_amd.serve.decode import AnswerVocab
+
+
+@pytest.fixture
+def serving(tmp_path, tiny_model, tiny_config):
+    broker = Broker(str(tmp_path / "q.sqlite3"), max_attempts=3)
+    db = Database(str(tmp_path / "db.sqlite3"))
+    db.seed_tasks()
+    push = NullPush()
+    runner = GraphRunner(
+        tiny_model, device="cpu", use_graphs=False,
+        feat_dim=tiny_config.v_feature_size,
+    )
+    worker = ServingWorker(
+        runner, broker, db, push,
+        provider=SyntheticFeatureProvider(feat_dim=tiny_config.v_feature_size),
+        tokenizer=BertWordPieceTokenizer(vocab_size=tiny_config.vocab_size),
+        vqa_vocab=AnswerVocab(tiny_config.num_labels_vqa),
+        gqa_vocab=AnswerVocab(tiny_config.num_labels_gqa),
+    )
+    return broker, db, push, worker
+
+
+def _result_messages(push, socket_id):
+    return [
+        p for s, p in push.messages if s == socket_id and "result" in p
+    ]
+
+
+def test_vqa_roundtrip(serving):
+    broker, db, push, worker = serving
+    vilbert_task(broker, ["/img/cat.jpg"], "What animal is this?", 1, "sock1")
+    assert worker.process_once() == 1
+    assert broker.depth() == 0
+    res = _result_messages(push, "sock1")
+    assert len(res) == 1
+    payload = json.loads(res[0]["result"])
+    assert payload["task_id"] == 1
+    assert len(payload["result"]) == 3
+    for entry in payload["result"]:
+        assert set(entry) == {"answer", "confidence"}
+        assert 0.0 <= entry["confidence"] <= 1.0
+    # DB row saved (worker.py:548-552,579-645 contract)
+    qa = db.get_question(1)
+    assert qa["input_text"] == "what animal is this?"
+    assert json.loads(qa["answer_text"])["task_id"] == 1
+    # terminal messages bracket the result (worker.py:647-649)
+    terms = [p for s, p in push.messages if s == "sock1" and "terminal" in p]
+    assert any("Completed" in str(p["terminal"]) for p in terms)
+
+
+def test_nlvr2_pair(serving):
+    broker, db, push, worker = serving
+    vilbert_task(broker, ["/a.jpg", "/b.jpg"], "both images contain dogs", 12, "s12")
+    assert worker.process_once() == 1
+    payload = json.loads(_result_messages(push, "s12")[0]["result"])
+    answers = {e["answer"] for e in payload["result"]}
+    assert answers == {"True", "False"}
+    assert abs(sum(e["confidence"] for e in payload["result"]) - 1.0) < 1e-5
+
+
+def test_retrieval_orders_all_images(serving):
+    broker, db, push, worker = serving
+    imgs = [f"/imgs/{i}.jpg" for i in range(4)]
+    vilbert_task(broker, imgs, "a dog on a beach", 7, "s7")
+    assert worker.process_once() == 1
+    payload = json.loads(_result_messages(push, "s7")[0]["result"])
+    assert payload["task_id"] == 7
+    assert sorted(payload["image_name_list"]) == sorted(f"{i}.jpg" for i in range(4))
+    confs = payload["confidence_list"]
+    assert confs == sorted(confs, reverse=True)
+    assert abs(sum(confs) - 1.0) < 1e-5
+
+
+def test_grounding_boxes(serving):
+    broker, db, push, worker = serving
+    vilbert_task(broker, ["/img/street.jpg"], "the red car", 11, "s11")
+    assert worker.process_once() == 1
+    payload = json.loads(_result_messages(push, "s11")[0]["result"])
+    assert payload["task_id"] == 11
+    assert len(payload["boxes"]) == 3
+    for box in payload["boxes"]:
+        x1, y1, x2, y2 = box
+        assert x2 >= x1 and y2 >= y1  # pixel-space, denormalized
+
+
+def test_mixed_task_batch(serving):
+    """BASELINE.json config 5: concurrent mixed-task dynamic batching."""
+    broker, db, push, worker = serving
+    vilbert_task(broker, ["/1.jpg"], "what is this?", 1, "m1")
+    vilbert_task(broker, ["/2.jpg", "/3.jpg"], "both have cats", 12, "m2")
+    vilbert_task(broker, ["/4.jpg"], "the blue ball", 11, "m3")
+    vilbert_task(broker, ["/5.jpg"], "is it raining?", 15, "m4")
+    assert worker.process_once() == 4
+    assert broker.depth() == 0
+    for sid, tid in [("m1", 1), ("m2", 12), ("m3", 11), ("m4", 15)]:
+        payload = json.loads(_result_messages(push, sid)[0]["result"])
+        assert payload["task_id"] == tid
+
+
+def test_invalid_image_count_rejected(serving):
+    """worker.py:256-263 arity check, incl. the task-2 dead path."""
+    broker, db, push, worker = serving
+    vilbert_task(broker, ["/a.jpg", "/b.jpg"], "what is this?", 1, "bad1")
+    vilbert_task(broker, ["/a.jpg"], "vg question", 2, "bad2")  # dead path
+    assert worker.process_once() == 0
+    assert broker.depth() == 0  # refused + acked, no redelivery
+    errs = [p for s, p in push.messages if "Error" in str(p.get("terminal", ""))]
+    assert len(errs) == 2
+
+
+def test_broker_redelivery_and_dead_letter(tmp_path):
+    broker = Broker(str(tmp_path / "q.sqlite3"), lease_timeout_s=0.0, max_attempts=2)
+    broker.publish({"x": 1})
+    d1 = broker.get()[0]
+    broker.nack(d1.msg_id)  # crash analogue -> redelivered (worker.py:653-655)
+    d2 = broker.get()[0]
+    assert d2.msg_id == d1.msg_id and d2.attempts == 2
+    broker.nack(d2.msg_id)
+    assert broker.get() == []  # attempt cap -> dead letter, not a poison loop
+    assert broker.dead_count() == 1
+
+
+def test_broker_lease_expiry(tmp_path):
+    broker = Broker(str(tmp_path / "q.sqlite3"), lease_timeout_s=0.0)
+    broker.publish({"x": 1})
+    assert len(broker.get()) == 1
+    # consumer died without ack; lease expired -> ready again
+    assert len(broker.get()) == 1
+
+
+def test_tokenizer_padding_contract():
+    tok = BertWordPieceTokenizer()
+    ids, mask, seg = tok.encode_for_serving("what color is the sky?")
+    assert len(ids) == 37 and len(mask) == 37 and len(seg) == 37
+    assert ids[0] == 101  # [CLS]
+    sep_pos = ids.index(102)
+    # END padding after [SEP] (worker.py:408-414 code, not its comment)
+    assert all(i == 0 for i in ids[sep_pos + 1 :])
+    assert all(m == 0 for m in mask[sep_pos + 1 :])
+    assert all(m == 1 for m in mask[: sep_pos + 1])
+    assert seg == [0] * 37

@@ -110,11 +110,11 @@ class VILBertForVLTasks(nn.Module):
         # NLVR2: batch is interleaved image pairs (worker.py:266-276 replicates
         # the text x2); concatenate consecutive pair representations.
         b = fused.shape[0]
-        if b % 2 == 0 and b > 0:
-            pair = fused.view(b // 2, -1)
+        if b >= 2:
+            pair = fused[: 2 * (b // 2)].reshape(b // 2, -1)
             vil_binary_prediction = self.vil_binary_prediction(pair)
         else:
-            vil_binary_prediction = fused.new_zeros(max(b // 2, 1), 2)
+            vil_binary_prediction = fused.new_zeros(1, 2)
         vil_tri_prediction = self.vil_tri_prediction(fused)
         vision_prediction = self.vision_prediction(self.dropout(v))
         vision_logit = self.vision_logit(self.dropout(v))

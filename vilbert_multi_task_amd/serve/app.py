@@ -1,0 +1,215 @@
+"""HTTP + websocket API, contract-identical to the reference's Django app.
+
+URL surface (from /root/reference/demo/urls.py:7-11 and
+vilbert_multitask/urls.py:21-31):
+  GET  /                        demo page (views.py:39-42)
+  POST /                        submit job: socket_id, task_id, question,
+                                image_list[] (views.py:20-38)
+  GET  /get_task_details/<id>/  JSON of the Tasks row (views.py:45-61)
+  POST /upload_image/           multi-file upload -> media/demo/<uuid>,
+                                {"file_paths": [...]} (views.py:89-106)
+  WS   /chat/                   client sends its socket_id first, then
+                                receives {terminal|result|info} pushes
+                                (routers.py:4-7, consumers.py:4-12)
+
+Implementation: FastAPI + the embedded PushHub (offline image has no
+Django/channels/Redis; the wire contracts are what the frontend consumes).
+"""
+
+from __future__ import annotations
+
+import asyncio
+import json
+import os
+import uuid
+from typing import List, Optional
+
+from fastapi import FastAPI, Request, WebSocket, WebSocketDisconnect
+from fastapi.responses import HTMLResponse, JSONResponse
+
+
+async def _parse_form(request: Request):
+    """stdlib form parsing (offline image lacks python-multipart):
+    returns (fields: {name: [values]}, files: [(name, filename, bytes)])."""
+    import email
+    import email.policy
+    from urllib.parse import parse_qs
+
+    ctype = request.headers.get("content-type", "")
+    body = await request.body()
+    fields: dict = {}
+    files: list = []
+    if ctype.startswith("multipart/form-data"):
+        raw = (
+            b"Content-Type: " + ctype.encode() + b"\r\nMIME-Version: 1.0\r\n\r\n" + body
+        )
+        msg = email.message_from_bytes(raw, policy=email.policy.HTTP)
+        for part in msg.iter_parts():
+            name = part.get_param("name", header="content-disposition")
+            filename = part.get_filename()
+            payload = part.get_payload(decode=True) or b""
+            if filename:
+                files.append((name, filename, payload))
+            else:
+                fields.setdefault(name, []).append(payload.decode(errors="replace"))
+    else:
+        for k, vs in parse_qs(body.decode(errors="replace")).items():
+            fields[k] = vs
+    return fields, files
+
+from .broker import Broker, vilbert_task
+from .db import Database
+from .push import PushClient, PushHub, log_to_terminal
+
+_INDEX_HTML = """<!doctype html>
+<html><head><title>ViLBERT Multi-Task Demo (MI355X)</title></head>
+<body>
+<h2>ViLBERT 12-in-1 — MI355X-native demo</h2>
+<p>POST / with socket_id, task_id, question, image_list[] to submit a job;
+open a websocket to /chat/ and send your socket_id to receive results.</p>
+</body></html>"""
+
+
+def create_app(
+    db_path: str = "vilbert_demo.sqlite3",
+    queue_path: str = "vilbert_queue.sqlite3",
+    media_root: str = "media",
+    hub_port: int = 0,
+) -> FastAPI:
+    app = FastAPI(title="vilbert-multitask-amd")
+    db = Database(db_path)
+    db.seed_tasks()
+    broker = Broker(queue_path)
+    hub = PushHub(port=hub_port) if hub_port else None
+    app.state.db = db
+    app.state.broker = broker
+    app.state.hub = hub
+    app.state.push = _LocalPush(None)  # in-process fanout (single-proc mode/tests)
+    os.makedirs(os.path.join(media_root, "demo"), exist_ok=True)
+
+    if hub is not None:
+
+        @app.on_event("startup")
+        async def _start_hub():
+            await hub.start()
+
+        @app.on_event("shutdown")
+        async def _stop_hub():
+            await hub.stop()
+
+    @app.get("/", response_class=HTMLResponse)
+    async def index() -> str:
+        return _INDEX_HTML
+
+    @app.post("/")
+    async def submit(request: Request):
+        fields, _ = await _parse_form(request)
+        socket_id = fields.get("socket_id", [""])[0]
+        task_id = fields.get("task_id", [""])[0]
+        question = fields.get("question", [""])[0].lower()  # views.py:28
+        image_list: List[str] = fields.get("image_list[]") or fields.get("image_list") or []
+        # absolute-path resolution analogue of views.py:30-32
+        paths = [
+            p if os.path.isabs(p) else os.path.join(media_root, p.lstrip("/"))
+            for p in image_list
+        ]
+        _push_local(app, socket_id, {"info": "Task submitted"})
+        vilbert_task(app.state.broker, paths, question, task_id or "1", socket_id)
+        return HTMLResponse(_INDEX_HTML)
+
+    @app.get("/get_task_details/{task_id}/")
+    async def get_task_details(task_id: int):
+        row = app.state.db.get_task(task_id)
+        if row is None:
+            return JSONResponse({"error": f"task {task_id} not found"}, status_code=404)
+        return JSONResponse(row)
+
+    @app.post("/upload_image/")
+    async def upload_image(request: Request):
+        _, files = await _parse_form(request)
+        file_paths = []
+        for _name, filename, payload in files[:4]:  # <=4 files (demo_images.html)
+            ext = os.path.splitext(filename or "img.jpg")[1] or ".jpg"
+            name = f"{uuid.uuid4().hex}{ext}"
+            dst = os.path.join(media_root, "demo", name)
+            with open(dst, "wb") as out:
+                out.write(payload)
+            file_paths.append(dst)
+        return JSONResponse({"file_paths": file_paths})
+
+    @app.websocket("/chat/")
+    async def chat(ws: WebSocket):
+        await ws.accept()
+        socket_id: Optional[str] = None
+        q: asyncio.Queue = asyncio.Queue()
+
+        def _send(payload: dict):
+            q.put_nowait(payload)
+
+        try:
+            # first client message is the socket id (consumers.py:8-11)
+            socket_id = (await ws.receive_text()).strip()
+            _register(app, socket_id, _send)
+            while True:
+                getter = asyncio.create_task(q.get())
+                recv = asyncio.create_task(ws.receive_text())
+                done, pending = await asyncio.wait(
+                    {getter, recv}, return_when=asyncio.FIRST_COMPLETED
+                )
+                for t in pending:
+                    t.cancel()
+                if getter in done:
+                    await ws.send_text(json.dumps(getter.result()))
+                if recv in done:
+                    recv.result()  # raises on disconnect; content ignored
+        except (WebSocketDisconnect, RuntimeError):
+            pass
+        finally:
+            if socket_id:
+                _unregister(app, socket_id, _send)
+
+    return app
+
+
+class _LocalPush:
+    """In-process push used when hub_port=0 (single-process app+worker, and
+    tests): delivers straight to registered WS queues."""
+
+    def __init__(self, app):
+        self.app = app
+        self.groups = {}
+
+    def publish(self, socket_id: str, payload: dict) -> bool:
+        for send in list(self.groups.get(socket_id, [])):
+            send(payload)
+        return True
+
+
+def _push_local(app: FastAPI, socket_id: str, payload: dict) -> None:
+    p = app.state.push
+    if isinstance(p, _LocalPush):
+        p.publish(socket_id, payload)
+    else:
+        log_to_terminal(p, socket_id, payload)
+    if app.state.hub is not None:
+        try:
+            loop = asyncio.get_event_loop()
+            loop.create_task(app.state.hub.dispatch(socket_id, payload))
+        except RuntimeError:
+            pass
+
+
+def _register(app: FastAPI, socket_id: str, send) -> None:
+    if app.state.hub is not None:
+        app.state.hub.join(socket_id, send)
+    if isinstance(app.state.push, _LocalPush):
+        app.state.push.groups.setdefault(socket_id, []).append(send)
+
+
+def _unregister(app: FastAPI, socket_id: str, send) -> None:
+    if app.state.hub is not None:
+        app.state.hub.leave(socket_id, send)
+    if isinstance(app.state.push, _LocalPush):
+        g = app.state.push.groups.get(socket_id, [])
+        if send in g:
+            g.remove(send)

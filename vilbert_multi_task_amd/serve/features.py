@@ -1,0 +1,107 @@
+"""Region-feature providers + the region tensorization of the reference.
+
+Tensorization is behavior-identical to /root/reference/worker.py:421-457:
+  - mean-pooled global feature PREPENDED -> 101 regions x 2048
+  - 5-d spatial = normalized x1,y1,x2,y2 + fractional area; global box
+    [0,0,1,1,1] (worker.py:436-444)
+  - image_mask all ones, co_attention_mask zeros
+
+Providers:
+  SyntheticFeatureProvider  — deterministic per-path features (demo/bench
+                              without the detector; BASELINE.md synthetic mode)
+  PrecomputedFeatureProvider — .npz files {features, bbox, image_w, image_h}
+  DetectorFeatureProvider    — full Faster R-CNN stack (detector module)
+"""
+
+from __future__ import annotations
+
+import hashlib
+import os
+from typing import Dict, List, Sequence
+
+import torch
+
+from ..tasks import FEATURE_DIM, MAX_SEQ_LENGTH, NUM_REGIONS
+
+
+class SyntheticFeatureProvider:
+    """Deterministic pseudo-features keyed by image path (no detector)."""
+
+    def __init__(self, num_boxes: int = 100, feat_dim: int = FEATURE_DIM):
+        self.num_boxes = num_boxes
+        self.feat_dim = feat_dim
+
+    def extract(self, image_paths: Sequence[str]) -> List[Dict]:
+        out = []
+        for p in image_paths:
+            seed = int.from_bytes(hashlib.sha1(p.encode()).digest()[:4], "little")
+            g = torch.Generator().manual_seed(seed)
+            feats = torch.randn(self.num_boxes, self.feat_dim, generator=g).abs()
+            w, h = 640.0, 480.0
+            c = torch.rand(self.num_boxes, 2, generator=g) * torch.tensor([w, h])
+            wh = torch.rand(self.num_boxes, 2, generator=g) * torch.tensor([w / 3, h / 3]) + 8
+            bbox = torch.cat([c - wh / 2, c + wh / 2], dim=1).clamp(min=0)
+            bbox[:, 2].clamp_(max=w)
+            bbox[:, 3].clamp_(max=h)
+            out.append(
+                {
+                    "features": feats,
+                    "bbox": bbox,
+                    "image_width": w,
+                    "image_height": h,
+                    "num_boxes": self.num_boxes,
+                }
+            )
+        return out
+
+
+class PrecomputedFeatureProvider:
+    def __init__(self, root: str = ""):
+        self.root = root
+
+    def extract(self, image_paths: Sequence[str]) -> List[Dict]:
+        import numpy as np
+
+        out = []
+        for p in image_paths:
+            fp = p if p.endswith(".npz") else os.path.join(self.root, os.path.basename(p) + ".npz")
+            z = np.load(fp)
+            out.append(
+                {
+                    "features": torch.from_numpy(z["features"]).float(),
+                    "bbox": torch.from_numpy(z["bbox"]).float(),
+                    "image_width": float(z["image_w"]),
+                    "image_height": float(z["image_h"]),
+                    "num_boxes": int(z["features"].shape[0]),
+                }
+            )
+        return out
+
+
+def tensorize_regions(infos: Sequence[Dict]) -> Dict[str, torch.Tensor]:
+    """Per-image region tensors (worker.py:421-457): returns features
+    [N,101,2048], spatials [N,101,5], image_mask [N,101]."""
+    feats_l, spat_l = [], []
+    for info in infos:
+        f = info["features"]  # [num_boxes, 2048]
+        num = int(info["num_boxes"])
+        w, h = float(info["image_width"]), float(info["image_height"])
+        g = f.mean(dim=0, keepdim=True)  # global mean-pooled feature
+        feats = torch.cat([g, f], dim=0)  # prepend -> 101 x 2048
+        bbox = info["bbox"].float()
+        x1, y1 = bbox[:, 0] / w, bbox[:, 1] / h
+        x2, y2 = bbox[:, 2] / w, bbox[:, 3] / h
+        area = (x2 - x1) * (y2 - y1)
+        sp = torch.stack([x1, y1, x2, y2, area], dim=1)
+        gsp = torch.tensor([[0.0, 0.0, 1.0, 1.0, 1.0]])  # worker.py:443
+        spat = torch.cat([gsp, sp], dim=0)
+        feats_l.append(feats)
+        spat_l.append(spat)
+    features = torch.stack(feats_l)
+    spatials = torch.stack(spat_l)
+    n, r = features.shape[0], features.shape[1]
+    return {
+        "features": features,
+        "spatials": spatials,
+        "image_mask": torch.ones(n, r, dtype=torch.long),
+    }

@@ -1,0 +1,223 @@
+"""The serving worker: queue consumer with dynamic cross-task batching.
+
+Replaces the reference's one-message-at-a-time blocking consumer
+(/root/reference/worker.py:542-673) with a polling consumer that drains up
+to ``max_batch_rows`` of queued requests per cycle and runs them through ONE
+hipGraph-replayed forward (BASELINE.json config 5: mixed-task dynamic
+batching). Everything user-observable is contract-identical:
+
+  message in : {"image_path": [...], "question": str, "socket_id": str,
+                "task_id": str}                     (sender.py:19-24)
+  push out   : {"terminal": json}, {"result": json}, {"terminal": "Completed..."}
+                                                    (worker.py:647-649)
+  DB rows    : questionanswer insert before inference, answer update after
+                                                    (worker.py:548-552,579-645)
+  ack        : only after success; failures redeliver (worker.py:650,653-655)
+                + dead-letter cap (broker.py) fixing the poison-message loop.
+
+Batch layout rule: NLVR2 (task 12) messages are placed FIRST so their two
+rows land at an even offset — the pair head concatenates rows (2i, 2i+1)
+(models/heads.py), matching the reference's pair batching (worker.py:266-276).
+"""
+
+from __future__ import annotations
+
+import json
+import time
+import traceback
+import uuid
+from dataclasses import dataclass, field
+from typing import Any, Dict, List, Optional, Sequence
+
+import torch
+
+from ..data.tokenizer import BertWordPieceTokenizer
+from ..tasks import MAX_SEQ_LENGTH, get_task, validate_request, DecodeFamily
+from .broker import Broker, Delivery, QUEUE_NAME
+from .db import Database
+from .decode import (
+    AnswerVocab,
+    decode_answer_task,
+    decode_grounding,
+    decode_retrieval,
+)
+from .features import SyntheticFeatureProvider, tensorize_regions
+from .push import PushClient, log_to_terminal
+
+
+@dataclass
+class _Request:
+    delivery: Delivery
+    task_id: int
+    question: str
+    image_paths: List[str]
+    socket_id: str
+    qa_id: int = -1
+    row_start: int = 0
+    num_rows: int = 0
+    infos: List[Dict] = field(default_factory=list)
+
+
+class ServingWorker:
+    def __init__(
+        self,
+        runner,
+        broker: Broker,
+        db: Database,
+        push: PushClient,
+        provider=None,
+        tokenizer: Optional[BertWordPieceTokenizer] = None,
+        vqa_vocab: Optional[AnswerVocab] = None,
+        gqa_vocab: Optional[AnswerVocab] = None,
+        max_batch_rows: int = 64,
+        queue: str = QUEUE_NAME,
+    ):
+        self.runner = runner
+        self.broker = broker
+        self.db = db
+        self.push = push
+        self.provider = provider or SyntheticFeatureProvider()
+        self.tokenizer = tokenizer or BertWordPieceTokenizer()
+        self.vqa_vocab = vqa_vocab or AnswerVocab(3129)
+        self.gqa_vocab = gqa_vocab or AnswerVocab(1533)
+        self.max_batch_rows = max_batch_rows
+        self.queue = queue
+
+    # ------------------------------------------------------------------
+    def _parse(self, d: Delivery) -> Optional[_Request]:
+        body = d.body
+        try:
+            task_id = int(body["task_id"])  # int(), not eval() (worker.py:562)
+            req = _Request(
+                delivery=d,
+                task_id=task_id,
+                question=str(body["question"]).lower(),
+                image_paths=list(body["image_path"]),
+                socket_id=str(body["socket_id"]),
+            )
+        except (KeyError, ValueError, TypeError):
+            self.broker.ack(d.msg_id)  # malformed: drop
+            return None
+        err = validate_request(req.task_id, len(req.image_paths))
+        if err:
+            log_to_terminal(self.push, req.socket_id, {"terminal": f"Error: {err}"})
+            self.broker.ack(d.msg_id)
+            return None
+        return req
+
+    # ------------------------------------------------------------------
+    def gather_batch(self, max_wait_s: float = 0.0) -> List[_Request]:
+        deadline = time.time() + max_wait_s
+        reqs: List[_Request] = []
+        rows = 0
+        while rows < self.max_batch_rows:
+            deliveries = self.broker.get(self.queue, max_n=self.max_batch_rows - rows)
+            for d in deliveries:
+                r = self._parse(d)
+                if r is None:
+                    continue
+                r.num_rows = len(r.image_paths)
+                reqs.append(r)
+                rows += r.num_rows
+            if deliveries or time.time() >= deadline:
+                break
+            time.sleep(0.005)
+        # NLVR2 first for even pair alignment
+        reqs.sort(key=lambda r: 0 if r.task_id == 12 else 1)
+        off = 0
+        for r in reqs:
+            r.row_start = off
+            off += r.num_rows
+        return reqs
+
+    # ------------------------------------------------------------------
+    def build_batch(self, reqs: Sequence[_Request]) -> Dict[str, torch.Tensor]:
+        q_rows, mask_rows, seg_rows, task_rows = [], [], [], []
+        infos_all: List[Dict] = []
+        for r in reqs:
+            ids, mask, seg = self.tokenizer.encode_for_serving(r.question, MAX_SEQ_LENGTH)
+            r.infos = self.provider.extract(r.image_paths)
+            infos_all.extend(r.infos)
+            for _ in range(r.num_rows):  # text replicated per image row
+                q_rows.append(ids)
+                mask_rows.append(mask)
+                seg_rows.append(seg)
+                task_rows.append([r.task_id])
+        reg = tensorize_regions(infos_all)
+        n = len(q_rows)
+        return {
+            "question": torch.tensor(q_rows, dtype=torch.long),
+            "input_mask": torch.tensor(mask_rows, dtype=torch.long),
+            "segment_ids": torch.tensor(seg_rows, dtype=torch.long),
+            "task_tokens": torch.tensor(task_rows, dtype=torch.long),
+            "features": reg["features"],
+            "spatials": reg["spatials"],
+            "image_mask": reg["image_mask"],
+            "co_attention_mask": torch.zeros(n, reg["features"].shape[1], MAX_SEQ_LENGTH),
+        }
+
+    # ------------------------------------------------------------------
+    def decode_request(self, r: _Request, outputs, batch) -> Dict[str, Any]:
+        spec = get_task(r.task_id)
+        row = r.row_start
+        if spec.decode == DecodeFamily.RETRIEVAL:
+            names = [p.split("/")[-1] for p in r.image_paths]
+            return decode_retrieval(
+                r.task_id, outputs, range(row, row + r.num_rows), names
+            )
+        if spec.decode == DecodeFamily.GROUNDING:
+            info = r.infos[0]
+            return decode_grounding(
+                r.task_id, outputs, row, batch["spatials"],
+                info["image_width"], info["image_height"],
+            )
+        if spec.decode == DecodeFamily.BINARY:
+            return decode_answer_task(
+                r.task_id, outputs, row // 2, self.vqa_vocab, self.gqa_vocab
+            )
+        return decode_answer_task(r.task_id, outputs, row, self.vqa_vocab, self.gqa_vocab)
+
+    # ------------------------------------------------------------------
+    def process_once(self, max_wait_s: float = 0.0) -> int:
+        """One drain-batch-infer-respond cycle; returns #requests served."""
+        reqs = self.gather_batch(max_wait_s)
+        if not reqs:
+            return 0
+        t0 = time.time()
+        for r in reqs:
+            r.qa_id = self.db.create_question(
+                r.task_id, r.question, r.image_paths, r.socket_id
+            )
+            log_to_terminal(
+                self.push, r.socket_id, {"terminal": "Processing request..."}
+            )
+        try:
+            batch = self.build_batch(reqs)
+            outputs = self.runner.run(batch)
+        except Exception:
+            traceback.print_exc()
+            for r in reqs:
+                self.broker.nack(r.delivery.msg_id)
+            return 0
+        served = 0
+        for r in reqs:
+            try:
+                result = self.decode_request(r, outputs, batch)
+                self.db.save_answer(r.qa_id, json.dumps(result))
+                log_to_terminal(self.push, r.socket_id, {"terminal": json.dumps(result)})
+                log_to_terminal(self.push, r.socket_id, {"result": json.dumps(result)})
+                log_to_terminal(
+                    self.push, r.socket_id, {"terminal": "Completed VilBERT task"}
+                )
+                self.broker.ack(r.delivery.msg_id)
+                served += 1
+            except Exception:
+                traceback.print_exc()
+                self.broker.nack(r.delivery.msg_id)
+        _ = time.time() - t0  # per-batch wall time (worker.py:657-658 analogue)
+        return served
+
+    def run_forever(self, poll_s: float = 0.02) -> None:
+        while True:
+            if self.process_once(max_wait_s=poll_s) == 0:
+                time.sleep(poll_s)
