@@ -1,0 +1,167 @@
+"""Multi-task training engine tests: round-robin sampler, per-task losses,
+bucketed DP all-reduce on gloo world_size=2 (SURVEY.md §4 consequence (4):
+gradient checks vs single-GPU accumulation)."""
+
+import os
+
+import pytest
+import torch
+import torch.distributed as dist
+import torch.multiprocessing as mp
+
+from vilbert_multi_task_amd.config import ViLBertConfig
+from vilbert_multi_task_amd.parallel.sampler import RoundRobinTaskSampler
+from vilbert_multi_task_amd.parallel.trainer import (
+    MultiTaskTrainer,
+    make_training_batch,
+    task_loss,
+)
+from vilbert_multi_task_amd.data.synthetic import forward_args
+from vilbert_multi_task_amd.models import VILBertForVLTasks
+from vilbert_multi_task_amd.tasks import TRAINING_DATASETS
+
+
+def _tiny_cfg_no_dropout():
+    cfg = ViLBertConfig.tiny()
+    cfg.hidden_dropout_prob = 0.0
+    cfg.attention_probs_dropout_prob = 0.0
+    cfg.v_hidden_dropout_prob = 0.0
+    cfg.v_attention_probs_dropout_prob = 0.0
+    return cfg
+
+
+def _tiny_trainer(device="cpu", batch=4, **kw):
+    cfg = _tiny_cfg_no_dropout()
+    torch.manual_seed(0)
+    model = VILBertForVLTasks(cfg)
+    return MultiTaskTrainer(
+        model, cfg, batch_size=batch, device=device, seq_len=20, regions=36, **kw
+    )
+
+
+def test_sampler_round_robin_and_state():
+    s = RoundRobinTaskSampler()
+    seq = [s.next_task() for _ in range(24)]
+    assert seq[:12] == list(TRAINING_DATASETS)
+    assert seq[12:] == list(TRAINING_DATASETS)
+    sd = s.state_dict()
+    s2 = RoundRobinTaskSampler()
+    s2.load_state_dict(sd)
+    assert s2.next_task() == s.next_task()
+
+
+def test_sampler_rank_agreement_disjoint_shards():
+    a = RoundRobinTaskSampler(rank=0, world_size=2)
+    b = RoundRobinTaskSampler(rank=1, world_size=2)
+    for _ in range(5):
+        ta, tb = a.next_task(), b.next_task()
+        assert ta == tb  # gradient layouts must agree across ranks
+        assert a.shard_seed(ta) != b.shard_seed(tb)  # data shards differ
+
+
+@pytest.mark.parametrize("dataset", list(TRAINING_DATASETS))
+def test_every_task_loss_backward(dataset, tiny_config):
+    torch.manual_seed(0)
+    model = VILBertForVLTasks(tiny_config)
+    model.train()
+    batch, targets = make_training_batch(
+        dataset, 4, tiny_config, seed=1, seq_len=20, regions=36
+    )
+    out = model(*forward_args(batch))
+    loss = task_loss(dataset, out, targets)
+    assert torch.isfinite(loss)
+    loss.backward()
+    # the shared trunk always receives gradients
+    assert model.bert.t_layers[0].attention.query.weight.grad is not None
+
+
+def test_trainer_steps_all_twelve_tasks():
+    tr = _tiny_trainer()
+    seen = set()
+    params_before = [p.detach().clone() for p in tr.model.parameters()][:3]
+    for _ in range(12):
+        ds, loss = tr.train_step()
+        seen.add(ds)
+        assert loss == loss  # finite
+    assert seen == set(TRAINING_DATASETS)
+    changed = any(
+        not torch.equal(a, b)
+        for a, b in zip(params_before, list(tr.model.parameters())[:3])
+    )
+    assert changed
+
+
+def test_checkpoint_roundtrip(tmp_path):
+    tr = _tiny_trainer()
+    for _ in range(3):
+        tr.train_step()
+    path = str(tmp_path / "ckpt.bin")
+    tr.save_checkpoint(path)
+    tr2 = _tiny_trainer()
+    tr2.load_checkpoint(path)
+    for (k1, v1), (k2, v2) in zip(
+        tr.model.state_dict().items(), tr2.model.state_dict().items()
+    ):
+        assert k1 == k2 and torch.equal(v1, v2), k1
+    assert tr2.sampler.state.step == 3
+    # resumed trainer draws the same next task
+    assert tr2.sampler.next_task() == tr.sampler.next_task()
+
+
+# ---------------------------------------------------------------------------
+# gloo world_size=2: DP grads == average of both shards computed serially
+# ---------------------------------------------------------------------------
+
+def _ddp_worker(rank, world, tmpdir):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = "29571"
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        torch.manual_seed(0)
+        tr = _tiny_trainer(rank=rank, world_size=world, bucket_bytes=1 << 18, grad_clip=0.0)
+        ds, _ = tr.train_step()
+        g = tr.model.bert.t_layers[0].attention.query.weight.grad.clone()
+        head_g = tr.model.vil_prediction.decoder.weight.grad
+        head_g = None if head_g is None else head_g.clone()
+        torch.save({"ds": ds, "g": g, "h": head_g}, os.path.join(tmpdir, f"r{rank}.pt"))
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(240)
+def test_ddp_grads_match_serial_average(tmp_path):
+    ctx = mp.get_context("spawn")
+    procs = [
+        ctx.Process(target=_ddp_worker, args=(r, 2, str(tmp_path)))
+        for r in range(2)
+    ]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(180)
+        assert p.exitcode == 0
+    r0 = torch.load(tmp_path / "r0.pt", weights_only=False)
+    r1 = torch.load(tmp_path / "r1.pt", weights_only=False)
+    ds0, g0, h0 = r0["ds"], r0["g"], r0["h"]
+    ds1, g1, h1 = r1["ds"], r1["g"], r1["h"]
+    assert ds0 == ds1 == "vqa_v2"
+    # both ranks end with identical (all-reduced) gradients
+    assert torch.allclose(g0, g1, atol=1e-6)
+    assert h0 is not None and torch.allclose(h0, h1, atol=1e-6)
+
+    # serial reference: average of per-shard grads, same shard seeds
+    cfg = _tiny_cfg_no_dropout()
+    torch.manual_seed(0)
+    model = VILBertForVLTasks(cfg)
+    model.train()
+    grads = []
+    for rank in range(2):
+        s = RoundRobinTaskSampler(rank=rank, world_size=2)
+        dsr = s.next_task()
+        batch, targets = make_training_batch(dsr, 4, cfg, s.shard_seed(dsr), seq_len=20, regions=36)
+        model.zero_grad()
+        out = model(*forward_args(batch))
+        task_loss(dsr, out, targets).backward()
+        grads.append(model.bert.t_layers[0].attention.query.weight.grad.clone())
+    ref = (grads[0] + grads[1]) / 2
+    assert torch.allclose(g0, ref, atol=1e-5), (g0 - ref).abs().max()

@@ -1,0 +1,182 @@
+"""12-task round-robin multi-task trainer with overlapped DP all-reduce.
+
+The engine the reference links to but does not vendor (SURVEY.md §0
+obligation 2: the 12-in-1 round-robin training engine), built MI355X-first:
+one process per GPU over RCCL (`nccl` backend on ROCm), bucketed all-reduce
+overlapped with backward (parallel/ddp.py), bf16 params with fp32 AdamW
+master state, checkpoints that round-trip to the upstream .bin layout
+(models/checkpoint.py).
+
+Per-dataset losses (12-in-1 recipe, pinned by the head dims the serving
+worker decodes — worker.py:295-386):
+  vqa_v2 / visual_genome_qa -> BCE on vil_prediction (3129 soft labels)
+  gqa                       -> BCE on vil_prediction_gqa (1533)
+  refcoco/refcoco+/refcocog/visual7w/guesswhat
+                            -> BCE on per-region vision_logit (grounding)
+  coco/flickr30k retrieval  -> CE over G=4 candidate images per caption
+  snli_ve                   -> CE 3-way on vil_tri_prediction
+  nlvr2                     -> CE 2-way on vil_binary_prediction (pairs)
+"""
+
+from __future__ import annotations
+
+import os
+from typing import Dict, Optional, Tuple
+
+import torch
+import torch.nn.functional as F
+
+from ..config import ViLBertConfig
+from ..data.synthetic import synthetic_batch, forward_args
+from .ddp import BucketedDataParallel
+from .sampler import RoundRobinTaskSampler
+
+RETRIEVAL_GROUP = 4  # candidates per caption (1 positive + 3 distractors)
+
+_TASK_ID_FOR_DATASET = {
+    "vqa_v2": 1,
+    "visual_genome_qa": 2,
+    "gqa": 15,
+    "refcoco": 11,
+    "refcoco_plus": 11,
+    "refcocog": 11,
+    "visual7w": 4,
+    "guesswhat": 16,
+    "coco_retrieval": 7,
+    "flickr30k_retrieval": 7,
+    "snli_ve": 13,
+    "nlvr2": 12,
+}
+
+
+def make_training_batch(
+    dataset: str, batch: int, cfg: ViLBertConfig, seed: int, device: str = "cpu",
+    seq_len: int = 37, regions: int = 101,
+) -> Tuple[dict, torch.Tensor]:
+    """Synthetic supervised batch for `dataset` (no network for real data —
+    BASELINE.md). Returns (forward batch, targets)."""
+    if dataset == "nlvr2" and batch % 2:
+        batch += 1
+    b = synthetic_batch(
+        batch, seq_len=seq_len, regions=regions,
+        feat_dim=cfg.v_feature_size, vocab_size=cfg.vocab_size,
+        task_id=_TASK_ID_FOR_DATASET[dataset], seed=seed, device=device,
+    )
+    g = torch.Generator().manual_seed(seed ^ 0x5EED)
+    if dataset in ("vqa_v2", "visual_genome_qa"):
+        t = torch.zeros(batch, cfg.num_labels_vqa)
+        idx = torch.randint(0, cfg.num_labels_vqa, (batch,), generator=g)
+        t[torch.arange(batch), idx] = 1.0
+    elif dataset == "gqa":
+        t = torch.zeros(batch, cfg.num_labels_gqa)
+        idx = torch.randint(0, cfg.num_labels_gqa, (batch,), generator=g)
+        t[torch.arange(batch), idx] = 1.0
+    elif dataset in ("refcoco", "refcoco_plus", "refcocog", "visual7w", "guesswhat"):
+        t = torch.zeros(batch, regions)
+        idx = torch.randint(1, regions, (batch,), generator=g)  # not the global box
+        t[torch.arange(batch), idx] = 1.0
+    elif dataset in ("coco_retrieval", "flickr30k_retrieval"):
+        assert batch % RETRIEVAL_GROUP == 0
+        t = torch.randint(0, RETRIEVAL_GROUP, (batch // RETRIEVAL_GROUP,), generator=g)
+    elif dataset == "snli_ve":
+        t = torch.randint(0, 3, (batch,), generator=g)
+    elif dataset == "nlvr2":
+        t = torch.randint(0, 2, (batch // 2,), generator=g)
+    else:
+        raise KeyError(dataset)
+    return b, t.to(device)
+
+
+def task_loss(dataset: str, outputs, targets: torch.Tensor) -> torch.Tensor:
+    if dataset in ("vqa_v2", "visual_genome_qa"):
+        return F.binary_cross_entropy_with_logits(outputs[0].float(), targets)
+    if dataset == "gqa":
+        return F.binary_cross_entropy_with_logits(outputs[1].float(), targets)
+    if dataset in ("refcoco", "refcoco_plus", "refcocog", "visual7w", "guesswhat"):
+        return F.binary_cross_entropy_with_logits(
+            outputs[6].squeeze(-1).float(), targets
+        )
+    if dataset in ("coco_retrieval", "flickr30k_retrieval"):
+        logits = outputs[2].float().view(-1, RETRIEVAL_GROUP)
+        return F.cross_entropy(logits, targets)
+    if dataset == "snli_ve":
+        return F.cross_entropy(outputs[4].float(), targets)
+    if dataset == "nlvr2":
+        return F.cross_entropy(outputs[3].float(), targets)
+    raise KeyError(dataset)
+
+
+class MultiTaskTrainer:
+    def __init__(
+        self,
+        model: torch.nn.Module,
+        cfg: ViLBertConfig,
+        lr: float = 4e-5,
+        batch_size: int = 8,
+        device: str = "cpu",
+        rank: int = 0,
+        world_size: int = 1,
+        bucket_bytes: int = 50 * 1024 * 1024,
+        seq_len: int = 37,
+        regions: int = 101,
+        grad_clip: float = 1.0,
+    ):
+        self.cfg = cfg
+        self.device = device
+        self.batch_size = batch_size
+        self.seq_len = seq_len
+        self.regions = regions
+        self.grad_clip = grad_clip
+        self.model = model.to(device)
+        self.ddp = BucketedDataParallel(self.model, bucket_bytes=bucket_bytes)
+        self.opt = torch.optim.AdamW(self.model.parameters(), lr=lr, weight_decay=0.01)
+        self.sampler = RoundRobinTaskSampler(rank=rank, world_size=world_size)
+
+    def train_step(self) -> Tuple[str, float]:
+        self.model.train()
+        dataset = self.sampler.next_task()
+        seed = self.sampler.shard_seed(dataset)
+        batch, targets = make_training_batch(
+            dataset, self.batch_size, self.cfg, seed, self.device,
+            self.seq_len, self.regions,
+        )
+        self.ddp.zero_grad()
+        out = self.ddp(*forward_args(batch))
+        loss = task_loss(dataset, out, targets)
+        loss.backward()
+        self.ddp.finalize_backward()
+        if self.grad_clip:
+            torch.nn.utils.clip_grad_norm_(self.model.parameters(), self.grad_clip)
+        self.opt.step()
+        return dataset, float(loss.detach())
+
+    # -- checkpoint (model + optimizer + sampler: SURVEY.md §5) -----------
+    def save_checkpoint(self, path: str, upstream_layout: bool = True) -> None:
+        from ..models.checkpoint import export_upstream_state_dict
+
+        state = {
+            "model": (
+                export_upstream_state_dict(self.model)
+                if upstream_layout
+                else self.model.state_dict()
+            ),
+            "model_layout": "upstream" if upstream_layout else "native",
+            "optimizer": self.opt.state_dict(),
+            "sampler": self.sampler.state_dict(),
+            "config": self.cfg.to_dict(),
+        }
+        tmp = path + ".tmp"
+        torch.save(state, tmp)
+        os.replace(tmp, path)
+
+    def load_checkpoint(self, path: str) -> None:
+        from ..models.checkpoint import load_upstream_state_dict
+
+        state = torch.load(path, map_location="cpu", weights_only=False)
+        if state.get("model_layout") == "upstream":
+            load_upstream_state_dict(self.model, state["model"])
+        else:
+            self.model.load_state_dict(state["model"])
+        self.model.to(self.device)
+        self.opt.load_state_dict(state["optimizer"])
+        self.sampler.load_state_dict(state["sampler"])
