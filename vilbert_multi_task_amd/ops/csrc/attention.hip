@@ -1,16 +1,24 @@
-// Fused multi-head attention for the ViLBERT serving shapes, gfx950 MFMA.
+// Fused multi-head attention for the ViLBERT serving shapes, gfx950 MFMA. v2
 //
 // Replaces the reference's unfused QK^T-matmul / masked-softmax / PV-matmul
 // chain (implicit cuBLAS+CUDA kernels) with ONE kernel per
 // forward: S = QK^T (mfma_f32_16x16x32_bf16, K staged in LDS), row softmax
 // entirely in registers (the whole key axis fits: Lk <= 128 for 101 regions /
 // 38 tokens — SURVEY.md §5: the scale axis is batch, not sequence), then
-// O = P·V with V staged TRANSPOSED in LDS so the B-fragment reads are
-// contiguous ds_read_b128.
+// O = P·V with V staged as [dblock][key-subtile][4][16] tiles consumed by the
+// gfx950 hardware transpose read ds_read_b64_tr_b16 (guide T10) — no
+// transpose pass, no padded rows.
 //
-// Layout contract (chosen so serving needs zero transpose copies): q,k,v and
-// out are the flattened projection outputs [B, L, H*D]; one workgroup (4
-// waves) owns one (b,h) pair; each wave owns 16-query-row stripes.
+// v2 over v1 (profiles/r01_serving_b256_kernels.md: v1 = 143.6 us/call, 19.5%
+// of the serving forward):
+//   - V transpose-staging (8 scalar ds_write_b16 per chunk) -> b128 subtile
+//     writes + tr_b16 fragment reads
+//   - +8-element row pads -> XOR swizzle (byte ^= (row&7)<<4) on K and P
+//     (guide §6 G4), shrinking LDS 87 KB -> 80 KB = 2 workgroups/CU
+//
+// Layout contract: q,k,v and out are the flattened projection outputs
+// [B, L, H*D]; one workgroup (4 waves) owns one (b,h) pair; each wave owns
+// 16-query-row stripes.
 //
 // MFMA fragment maps (gfx950 v_mfma_f32_16x16x32_bf16, verified on-device by
 // the mfma_probe op + tests/test_gpu_ops.py):
@@ -21,28 +29,23 @@
 #include "common.h"
 
 #define ATTN_MAX_L 128  // max Lq/Lk this kernel serves (serving shapes <=101+pad)
-#define ROWPAD 8        // +8 bf16 = +16B row padding: breaks the 256B-stride
-                        // 16-way ds_read_b128 bank conflict (guide §6 G4)
+
+typedef __attribute__((ext_vector_type(4))) short s4_vec;
+typedef __attribute__((address_space(3))) s4_vec* lds_v4s;
 
 DEV bf16x8 load_bf16x8(const bf16* p) {
-  union {
-    uint4 u;
-    bf16x8 v;
-  } c;
+  union { uint4 u; bf16x8 v; } c;
   c.u = *reinterpret_cast<const uint4*>(p);
   return c.v;
 }
 
-// 8 bf16 -> float[8] global load (16 B)
-DEV void VecIO_attn_load(const bf16* p, float* out) {
-  const uint4 raw = *reinterpret_cast<const uint4*>(p);
-  const unsigned int w[4] = {raw.x, raw.y, raw.z, raw.w};
-#pragma unroll
-  for (int i = 0; i < 4; ++i) {
-    out[2 * i] = us2f((unsigned short)(w[i] & 0xffff));
-    out[2 * i + 1] = us2f((unsigned short)(w[i] >> 16));
-  }
+DEV bf16x8 lds_b128(const char* p) {
+  union { uint4 u; bf16x8 v; } c;
+  c.u = *reinterpret_cast<const uint4*>(p);
+  return c.v;
 }
+
+DEV void lds_store_b128(char* p, uint4 v) { *reinterpret_cast<uint4*>(p) = v; }
 
 template <int D>
 __global__ __launch_bounds__(256) void attn_kernel(
@@ -50,7 +53,7 @@ __global__ __launch_bounds__(256) void attn_kernel(
     const bf16* __restrict__ vg, const bf16* __restrict__ mask,
     bf16* __restrict__ out, int B, int H, int Lq, int Lk, int mask_mode,
     float scale) {
-  constexpr int KCH = D / 8;    // 16B chunks per row
+  constexpr int KCH = D / 8;  // 16B chunks per row
   const int HD = H * D;
   const int bh = blockIdx.x;
   const int b = bh / H;
@@ -60,36 +63,29 @@ __global__ __launch_bounds__(256) void attn_kernel(
   const int wid = wave_id();
 
   const int LK_PAD = (Lk + 31) & ~31;
-  const int NT = LK_PAD / 16;          // <= 8 score tiles per stripe
-  const int KSTR = D + ROWPAD;         // K_lds row stride (bf16 elems)
-  const int VSTR = LK_PAD + ROWPAD;    // VT_lds / P_lds row stride
+  const int NT = LK_PAD / 16;  // <= 8 score tiles per stripe
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  bf16* K_lds = reinterpret_cast<bf16*>(smem);
-  bf16* VT_lds = K_lds + LK_PAD * KSTR;
-  bf16* P_lds = VT_lds + D * VSTR + wid * 16 * VSTR;  // per-wave P stripe
+  char* K_lds = smem;                          // [LK_PAD][D] bf16, XOR-swizzled
+  char* V_lds = K_lds + LK_PAD * D * 2;        // [D/16][LK_PAD/4][4][16] bf16
+  char* P_lds = V_lds + LK_PAD * D * 2 + wid * 16 * LK_PAD * 2;  // per-wave
 
-  // ---- stage K (row-major, padded) and V (transposed) into LDS ----------
+  // ---- stage K (swizzled row-major) and V (tr-readable subtiles) ---------
   {
     const int rows_per_pass = blockDim.x / KCH;  // 16 (D=128) or 32 (D=64)
     const int r0 = tid / KCH;
     const int c = tid % KCH;
     const long base = ((long)b * Lk) * HD + (long)h * D;
     for (int r = r0; r < LK_PAD; r += rows_per_pass) {
-      float kv[8];
-      float vv[8];
+      uint4 kraw = {0, 0, 0, 0}, vraw = {0, 0, 0, 0};
       if (r < Lk) {
-        VecIO_attn_load(kg + base + (long)r * HD + c * 8, kv);
-        VecIO_attn_load(vg + base + (long)r * HD + c * 8, vv);
-      } else {
-#pragma unroll
-        for (int j = 0; j < 8; ++j) kv[j] = vv[j] = 0.f;
+        kraw = *reinterpret_cast<const uint4*>(kg + base + (long)r * HD + c * 8);
+        vraw = *reinterpret_cast<const uint4*>(vg + base + (long)r * HD + c * 8);
       }
-      bf16* krow = K_lds + r * KSTR + c * 8;
-#pragma unroll
-      for (int j = 0; j < 8; ++j) krow[j] = f2bf(kv[j]);
-#pragma unroll
-      for (int j = 0; j < 8; ++j) VT_lds[(c * 8 + j) * VSTR + r] = f2bf(vv[j]);
+      lds_store_b128(K_lds + r * (D * 2) + ((c * 16) ^ ((r & 7) << 4)), kraw);
+      // V subtile: db = c/2 (16-col block), ks = r/4, jrow = r%4, half = c%2
+      const int voff = (c / 2) * (LK_PAD * 16) + (r / 4) * 64 + (r % 4) * 16 + (c % 2) * 8;
+      lds_store_b128(V_lds + voff * 2, vraw);
     }
   }
   __syncthreads();
@@ -114,10 +110,12 @@ __global__ __launch_bounds__(256) void attn_kernel(
 #pragma unroll
     for (int nt = 0; nt < 8; ++nt) {
       if (nt >= NT) break;
-      const bf16* kbase = K_lds + (nt * 16 + (lane & 15)) * KSTR + (lane >> 4) * 8;
+      const int key = nt * 16 + (lane & 15);
+      const char* kbase = K_lds + key * (D * 2);
+      const int ksw = (key & 7) << 4;
 #pragma unroll
       for (int kk = 0; kk < D / 32; ++kk) {
-        const bf16x8 bk = load_bf16x8(kbase + kk * 32);
+        const bf16x8 bk = lds_b128(kbase + (((kk * 64) + ((lane >> 4) * 16)) ^ ksw));
         acc_s[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
             (bf16x8)aq[kk], bk, acc_s[nt], 0, 0, 0);
       }
@@ -126,7 +124,6 @@ __global__ __launch_bounds__(256) void attn_kernel(
     // ---- mask + softmax (rows live across the 16-lane group) -------------
     const int col0 = lane & 15;
     float inv_l[4];
-    float mrow[4], lrow[4];
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
       const int row = qrow0 + (lane >> 4) * 4 + r;
@@ -147,46 +144,56 @@ __global__ __launch_bounds__(256) void attn_kernel(
           acc_s[nt][r] = -3.0e38f;
         }
       }
-      mrow[r] = group16_max(mx);
+      const float mrow = group16_max(mx);
       float sum = 0.f;
 #pragma unroll
       for (int nt = 0; nt < 8; ++nt) {
         if (nt >= NT) break;
         const int col = nt * 16 + col0;
-        const float p = (col < Lk) ? __expf(acc_s[nt][r] - mrow[r]) : 0.f;
+        const float p = (col < Lk) ? __expf(acc_s[nt][r] - mrow) : 0.f;
         acc_s[nt][r] = p;
         sum += p;
       }
-      lrow[r] = group16_sum(sum);
-      inv_l[r] = 1.0f / lrow[r];
+      inv_l[r] = 1.0f / group16_sum(sum);
     }
 
-    // ---- P -> LDS (bf16, A-readable row-major) ----------------------------
+    // ---- P -> LDS (bf16, swizzled row-major; A-frag readable) -------------
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
       const int prow = (lane >> 4) * 4 + r;
+      char* prow_base = P_lds + prow * (LK_PAD * 2);
+      const int psw = (prow & 7) << 4;
 #pragma unroll
       for (int nt = 0; nt < 8; ++nt) {
         if (nt >= NT) break;
-        P_lds[prow * VSTR + nt * 16 + col0] = f2bf(acc_s[nt][r]);
+        const int col = nt * 16 + col0;
+        *reinterpret_cast<short*>(prow_base + ((col * 2) ^ psw)) =
+            (short)f2us(acc_s[nt][r]);
       }
     }
-    // per-wave P buffer: same wave writes then reads — lgkm waits inserted
-    // by the compiler via the address dependence; no cross-wave sharing.
+    // per-wave P buffer: same wave writes then reads (compiler inserts the
+    // lgkm waits through the address dependence); no cross-wave sharing.
 
-    // ---- O = P V ----------------------------------------------------------
+    // ---- O = P V (V fragments via hardware transpose reads) ---------------
     f32x4 acc_o[D / 16];
 #pragma unroll
     for (int nt = 0; nt < D / 16; ++nt) acc_o[nt] = {0.f, 0.f, 0.f, 0.f};
+    const int parow = lane & 15;
+    const char* pa_base = P_lds + parow * (LK_PAD * 2);
+    const int pasw = (parow & 7) << 4;
 #pragma unroll
     for (int kk = 0; kk < 4; ++kk) {  // LK_PAD/32 <= 4
       if (kk * 32 >= LK_PAD) break;
-      const bf16x8 ap = load_bf16x8(P_lds + (lane & 15) * VSTR + kk * 32 + (lane >> 4) * 8);
+      const bf16x8 ap =
+          lds_b128(pa_base + (((kk * 64) + ((lane >> 4) * 16)) ^ pasw));
+      const int ks0 = kk * 8 + (lane >> 4) * 2;  // 4-key subtile index
 #pragma unroll
       for (int nt = 0; nt < D / 16; ++nt) {
-        const bf16x8 bv = load_bf16x8(
-            VT_lds + (nt * 16 + (lane & 15)) * VSTR + kk * 32 + (lane >> 4) * 8);
-        acc_o[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ap, bv, acc_o[nt], 0, 0, 0);
+        char* vbase = V_lds + (nt * (LK_PAD * 16) + ks0 * 64 + (lane & 15)) * 2;
+        union { struct { short4v lo, hi; } p; bf16x8 v; } bv;
+        bv.p.lo = __builtin_amdgcn_ds_read_tr16_b64_v4i16((lds_v4s)vbase);
+        bv.p.hi = __builtin_amdgcn_ds_read_tr16_b64_v4i16((lds_v4s)(vbase + 128));
+        acc_o[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ap, bv.v, acc_o[nt], 0, 0, 0);
       }
     }
 
@@ -234,9 +241,8 @@ void launch_attention(const bf16* q, const bf16* k, const bf16* v,
                       int D, int mask_mode, hipStream_t stream) {
   const float scale = 1.0f / sqrtf((float)D);
   const int LK_PAD = (Lk + 31) & ~31;
-  const int KSTR = D + ROWPAD;
-  const int VSTR = LK_PAD + ROWPAD;
-  const size_t lds = sizeof(bf16) * (LK_PAD * KSTR + D * VSTR + 4 * 16 * VSTR);
+  // K + V + 4x per-wave P (all bf16): (2*LK_PAD*D + 4*16*LK_PAD) elems
+  const size_t lds = sizeof(bf16) * (size_t)(2 * LK_PAD * D + 4 * 16 * LK_PAD);
   const dim3 grid(B * H);
   if (D == 64)
     hipLaunchKernelGGL((attn_kernel<64>), grid, dim3(256), lds, stream, q, k, v,
