@@ -5,16 +5,19 @@
 // forward: S = QK^T (mfma_f32_16x16x32_bf16, K staged in LDS), row softmax
 // entirely in registers (the whole key axis fits: Lk <= 128 for 101 regions /
 // 38 tokens — SURVEY.md §5: the scale axis is batch, not sequence), then
-// O = P·V with V staged as [dblock][key-subtile][4][16] tiles consumed by the
-// gfx950 hardware transpose read ds_read_b64_tr_b16 (guide T10) — no
-// transpose pass, no padded rows.
+// O = P·V with V staged TRANSPOSED [D][LK_PAD] so B-fragment reads are
+// contiguous ds_read_b128 (transpose writes are conflict-free: contiguous
+// bytes per VT row).
 //
 // v2 over v1 (profiles/r01_serving_b256_kernels.md: v1 = 143.6 us/call, 19.5%
-// of the serving forward):
-//   - V transpose-staging (8 scalar ds_write_b16 per chunk) -> b128 subtile
-//     writes + tr_b16 fragment reads
-//   - +8-element row pads -> XOR swizzle (byte ^= (row&7)<<4) on K and P
-//     (guide §6 G4), shrinking LDS 87 KB -> 80 KB = 2 workgroups/CU
+// of the serving forward): +8-element row pads -> XOR swizzle
+// (byte ^= (row&7)<<4) on K, VT and P (guide §6 G4), shrinking LDS
+// 87 KB -> 80 KB = 2 workgroups/CU.
+// NOTE on ds_read_b64_tr_b16: measured semantics (tr16_probe op, MI355X):
+// lane l elem j = mem[addr_of_lane(16*(l>>4) + 4j) + (l&3)*2] — per 16-lane
+// group only lanes 0/4/8/12's addresses are honored (16 distinct values per
+// read), so it CANNOT feed a 16x16x32 B-fragment (128 distinct values per
+// group); the transposed-image staging below is the right structure here.
 //
 // Layout contract: q,k,v and out are the flattened projection outputs
 // [B, L, H*D]; one workgroup (4 waves) owns one (b,h) pair; each wave owns
@@ -83,9 +86,23 @@ __global__ __launch_bounds__(256) void attn_kernel(
         vraw = *reinterpret_cast<const uint4*>(vg + base + (long)r * HD + c * 8);
       }
       lds_store_b128(K_lds + r * (D * 2) + ((c * 16) ^ ((r & 7) << 4)), kraw);
-      // V subtile: db = c/2 (16-col block), ks = r/4, jrow = r%4, half = c%2
-      const int voff = (c / 2) * (LK_PAD * 16) + (r / 4) * 64 + (r % 4) * 16 + (c % 2) * 8;
-      lds_store_b128(V_lds + voff * 2, vraw);
+      // V transposed image [D][LK_PAD] (XOR-swizzled rows): the PV MFMA
+      // B-fragment wants per-lane contiguous keys at fixed d, so transpose
+      // at staging. Writes are conflict-free: for fixed j, consecutive
+      // threads (consecutive r) write contiguous bytes of one VT row.
+      // (ds_read_b64_tr_b16 was measured unusable here: per 16-lane group
+      // it honors only lanes 0/4/8/12's addresses — 16 distinct values per
+      // read vs the 128 a 16x16x32 B-fragment needs; see tr16_probe.)
+      {
+        union { uint4 u; short s[8]; } vv;
+        vv.u = vraw;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          const int d = c * 8 + j;
+          *reinterpret_cast<short*>(
+              V_lds + d * (LK_PAD * 2) + ((r * 2) ^ ((d & 7) << 4))) = vv.s[j];
+        }
+      }
     }
   }
   __syncthreads();
@@ -186,14 +203,13 @@ __global__ __launch_bounds__(256) void attn_kernel(
       if (kk * 32 >= LK_PAD) break;
       const bf16x8 ap =
           lds_b128(pa_base + (((kk * 64) + ((lane >> 4) * 16)) ^ pasw));
-      const int ks0 = kk * 8 + (lane >> 4) * 2;  // 4-key subtile index
+      const int keyoff = (kk * 64) + ((lane >> 4) * 16);  // byte offset of keys
 #pragma unroll
       for (int nt = 0; nt < D / 16; ++nt) {
-        char* vbase = V_lds + (nt * (LK_PAD * 16) + ks0 * 64 + (lane & 15)) * 2;
-        union { struct { short4v lo, hi; } p; bf16x8 v; } bv;
-        bv.p.lo = __builtin_amdgcn_ds_read_tr16_b64_v4i16((lds_v4s)vbase);
-        bv.p.hi = __builtin_amdgcn_ds_read_tr16_b64_v4i16((lds_v4s)(vbase + 128));
-        acc_o[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ap, bv.v, acc_o[nt], 0, 0, 0);
+        const int d = nt * 16 + (lane & 15);
+        const bf16x8 bv =
+            lds_b128(V_lds + d * (LK_PAD * 2) + (keyoff ^ ((d & 7) << 4)));
+        acc_o[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ap, bv, acc_o[nt], 0, 0, 0);
       }
     }
 
@@ -233,6 +249,24 @@ __global__ void mfma_probe_kernel(const bf16* __restrict__ a,
 }
 
 // ---------------------------------------------------------------------------
+// tr16 semantics probe: lds[i] = i for i in [0,1024); each lane does one
+// ds_read_b64_tr_b16. mode 0: per-lane canonical base (l&15)*2+(l>>4)*128 B;
+// mode 1: uniform base 0. Output [64][4] int16 of the values each lane got.
+// ---------------------------------------------------------------------------
+__global__ void tr16_probe_kernel(short* __restrict__ outv, int mode) {
+  __shared__ short lds[1024];
+  for (int i = threadIdx.x; i < 1024; i += blockDim.x) lds[i] = (short)i;
+  __syncthreads();
+  const int l = threadIdx.x;
+  if (l < 64) {
+    int elem_off = (mode == 0) ? ((l & 15) + (l >> 4) * 64) : 0;
+    s4_vec v = __builtin_amdgcn_ds_read_tr16_b64_v4i16((lds_v4s)&lds[elem_off]);
+#pragma unroll
+    for (int j = 0; j < 4; ++j) outv[l * 4 + j] = v[j];
+  }
+}
+
+// ---------------------------------------------------------------------------
 // launchers
 // ---------------------------------------------------------------------------
 
@@ -254,4 +288,8 @@ void launch_attention(const bf16* q, const bf16* k, const bf16* v,
 
 void launch_mfma_probe(const bf16* a, const bf16* b, float* c, hipStream_t stream) {
   hipLaunchKernelGGL(mfma_probe_kernel, dim3(1), dim3(64), 0, stream, a, b, c);
+}
+
+void launch_tr16_probe(short* out, int mode, hipStream_t stream) {
+  hipLaunchKernelGGL(tr16_probe_kernel, dim3(1), dim3(64), 0, stream, out, mode);
 }

@@ -26,6 +26,7 @@ void launch_attention(const bf16*, const bf16*, const bf16*, const bf16*, bf16*,
 void launch_mfma_probe(const bf16*, const bf16*, float*, hipStream_t);
 void launch_nms_multiclass(const float*, const float*, const long*, float*, int,
                            int, float, float, hipStream_t);
+void launch_tr16_probe(short*, int, hipStream_t);
 
 namespace {
 
@@ -154,6 +155,12 @@ at::Tensor mfma_probe(const at::Tensor& a, const at::Tensor& b) {
   return c;
 }
 
+at::Tensor tr16_probe(const at::Tensor& dummy, int64_t mode) {
+  auto out = at::empty({64, 4}, dummy.options().dtype(at::kShort));
+  launch_tr16_probe((short*)out.data_ptr(), (int)mode, cur_stream());
+  return out;
+}
+
 at::Tensor nms_multiclass(const at::Tensor& boxes, const at::Tensor& scores,
                           double iou_thr, double score_thr) {
   TORCH_CHECK(boxes.is_cuda() && boxes.scalar_type() == at::kFloat);
@@ -178,6 +185,7 @@ TORCH_LIBRARY(vilbert_amd, m) {
   m.def("attention(Tensor q, Tensor k, Tensor v, int heads, Tensor? mask) -> Tensor");
   m.def("embedding_ln(Tensor ids, Tensor pos, Tensor type, Tensor word_w, Tensor pos_w, Tensor type_w, Tensor ln_w, Tensor ln_b, float eps) -> Tensor");
   m.def("mfma_probe(Tensor a, Tensor b) -> Tensor");
+  m.def("tr16_probe(Tensor dummy, int mode) -> Tensor");
   m.def("nms_multiclass(Tensor boxes, Tensor scores, float iou_thr, float score_thr) -> Tensor");
 }
 
@@ -187,5 +195,6 @@ TORCH_LIBRARY_IMPL(vilbert_amd, CUDA, m) {
   m.impl("attention", attention);
   m.impl("embedding_ln", embedding_ln);
   m.impl("mfma_probe", mfma_probe);
+  m.impl("tr16_probe", tr16_probe);
   m.impl("nms_multiclass", nms_multiclass);
 }
