@@ -141,3 +141,14 @@ def test_base_model_param_count():
     m = VILBertForVLTasks(cfg)
     n = sum(p.numel() for p in m.parameters())
     assert 200e6 < n < 340e6, f"param count {n/1e6:.1f}M out of expected range"
+
+
+def test_prepare_for_serving_matches_eager(tiny_model, tiny_config):
+    batch = _tiny_batch(tiny_config)
+    with torch.no_grad():
+        base = tiny_model(*forward_args(batch))
+    tiny_model.prepare_for_serving()
+    with torch.no_grad():
+        fused = tiny_model(*forward_args(batch))
+    for i in range(9):
+        assert torch.allclose(base[i], fused[i], atol=1e-5), i

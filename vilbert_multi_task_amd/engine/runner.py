@@ -30,6 +30,8 @@ class GraphRunner:
         use_graphs: bool = True,
     ):
         self.model = model.eval()
+        if hasattr(model, "prepare_for_serving"):
+            model.prepare_for_serving()
         self.seq_len = seq_len
         self.regions = regions
         self.feat_dim = feat_dim

@@ -134,6 +134,13 @@ class VILBertForVLTasks(nn.Module):
             attn_data_list,
         )
 
+    def prepare_for_serving(self) -> None:
+        """Fuse projection GEMMs for the inference path (idempotent).
+        Call after weights are final (post-load, post-.to(device/dtype))."""
+        for m in self.modules():
+            if hasattr(m, "prepare_serving"):
+                m.prepare_serving()
+
     # ---- factory matching the reference loader contract -----------------
     @classmethod
     def from_pretrained(

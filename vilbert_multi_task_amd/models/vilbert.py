@@ -127,12 +127,32 @@ class MultiHeadSelfAttention(nn.Module):
         self.attn_dropout_p = attn_dropout
         self.dropout = nn.Dropout(hidden_dropout)
         self.layer_norm = FusedLayerNorm(hidden, eps)
+        self._wqkv = None  # serving-time fused projection (prepare_serving)
+        self._bqkv = None
+
+    def prepare_serving(self) -> None:
+        """Fuse Q/K/V projections into one GEMM for the inference path
+        (3 hipBLASLt launches -> 1; the attention kernel reads the q/k/v
+        slices of the fused output by stride, no copies)."""
+        with torch.no_grad():
+            self._wqkv = torch.cat(
+                [self.query.weight, self.key.weight, self.value.weight], dim=0
+            ).contiguous()
+            self._bqkv = torch.cat(
+                [self.query.bias, self.key.bias, self.value.bias], dim=0
+            ).contiguous()
 
     def forward(
         self, x: torch.Tensor, mask_bias: Optional[torch.Tensor], need_probs: bool
     ) -> Tuple[torch.Tensor, Optional[torch.Tensor]]:
+        if self._wqkv is not None and not self.training and not torch.is_grad_enabled():
+            h = self.heads * self.head_dim
+            qkv = torch.nn.functional.linear(x, self._wqkv, self._bqkv)
+            q, k, v = qkv[..., :h], qkv[..., h : 2 * h], qkv[..., 2 * h :]
+        else:
+            q, k, v = self.query(x), self.key(x), self.value(x)
         ctx, probs = F_ops.attention(
-            self.query(x), self.key(x), self.value(x), self.heads,
+            q, k, v, self.heads,
             mask_bias, self.attn_dropout_p, self.training, need_probs,
         )
         y = self.layer_norm(self.dropout(self.out(ctx)), residual=x)
@@ -182,10 +202,23 @@ class CrossAttention(nn.Module):
         self.attn_dropout_p = attn_dropout
         self.dropout = nn.Dropout(hidden_dropout)
         self.layer_norm = FusedLayerNorm(q_hidden, eps)
+        self._wkv = None  # serving-time fused K/V projection
+        self._bkv = None
+
+    def prepare_serving(self) -> None:
+        with torch.no_grad():
+            self._wkv = torch.cat([self.key.weight, self.value.weight], dim=0).contiguous()
+            self._bkv = torch.cat([self.key.bias, self.value.bias], dim=0).contiguous()
 
     def forward(self, x_q, x_kv, mask_bias, need_probs=False):
+        if self._wkv is not None and not self.training and not torch.is_grad_enabled():
+            h = self.heads * self.head_dim
+            kv = torch.nn.functional.linear(x_kv, self._wkv, self._bkv)
+            k, v = kv[..., :h], kv[..., h:]
+        else:
+            k, v = self.key(x_kv), self.value(x_kv)
         ctx, probs = F_ops.attention(
-            self.query(x_q), self.key(x_kv), self.value(x_kv), self.heads,
+            self.query(x_q), k, v, self.heads,
             mask_bias, self.attn_dropout_p, self.training, need_probs,
         )
         y = self.layer_norm(self.dropout(self.out(ctx)), residual=x_q)

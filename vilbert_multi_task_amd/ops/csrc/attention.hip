@@ -55,7 +55,9 @@ __global__ __launch_bounds__(256) void attn_kernel(
     const bf16* __restrict__ q, const bf16* __restrict__ kg,
     const bf16* __restrict__ vg, const bf16* __restrict__ mask,
     bf16* __restrict__ out, int B, int H, int Lq, int Lk, int mask_mode,
-    float scale) {
+    float scale, int qs, int ks, int vs) {
+  // qs/ks/vs: row strides (elems) of q/k/v — [B,L,H*D] views into a fused
+  // QKV (or KV) projection pass without any copy (stride 3*HD / 2*HD).
   constexpr int KCH = D / 8;  // 16B chunks per row
   const int HD = H * D;
   const int bh = blockIdx.x;
@@ -78,12 +80,13 @@ __global__ __launch_bounds__(256) void attn_kernel(
     const int rows_per_pass = blockDim.x / KCH;  // 16 (D=128) or 32 (D=64)
     const int r0 = tid / KCH;
     const int c = tid % KCH;
-    const long base = ((long)b * Lk) * HD + (long)h * D;
+    const long kbase0 = (long)b * Lk * ks + (long)h * D;
+    const long vbase0 = (long)b * Lk * vs + (long)h * D;
     for (int r = r0; r < LK_PAD; r += rows_per_pass) {
       uint4 kraw = {0, 0, 0, 0}, vraw = {0, 0, 0, 0};
       if (r < Lk) {
-        kraw = *reinterpret_cast<const uint4*>(kg + base + (long)r * HD + c * 8);
-        vraw = *reinterpret_cast<const uint4*>(vg + base + (long)r * HD + c * 8);
+        kraw = *reinterpret_cast<const uint4*>(kg + kbase0 + (long)r * ks + c * 8);
+        vraw = *reinterpret_cast<const uint4*>(vg + vbase0 + (long)r * vs + c * 8);
       }
       lds_store_b128(K_lds + r * (D * 2) + ((c * 16) ^ ((r & 7) << 4)), kraw);
       // V transposed image [D][LK_PAD] (XOR-swizzled rows): the PV MFMA
@@ -115,7 +118,7 @@ __global__ __launch_bounds__(256) void attn_kernel(
     bf16x8 aq[D / 32];
     {
       const int row = min(qrow0 + (lane & 15), Lq - 1);
-      const long qoff = ((long)b * Lq + row) * HD + (long)h * D + (lane >> 4) * 8;
+      const long qoff = ((long)b * Lq + row) * qs + (long)h * D + (lane >> 4) * 8;
 #pragma unroll
       for (int kk = 0; kk < D / 32; ++kk) aq[kk] = load_bf16x8(q + qoff + kk * 32);
     }
@@ -272,7 +275,8 @@ __global__ void tr16_probe_kernel(short* __restrict__ outv, int mode) {
 
 void launch_attention(const bf16* q, const bf16* k, const bf16* v,
                       const bf16* mask, bf16* out, int B, int H, int Lq, int Lk,
-                      int D, int mask_mode, hipStream_t stream) {
+                      int D, int mask_mode, int qs, int ks, int vs,
+                      hipStream_t stream) {
   const float scale = 1.0f / sqrtf((float)D);
   const int LK_PAD = (Lk + 31) & ~31;
   // K + V + 4x per-wave P (all bf16): (2*LK_PAD*D + 4*16*LK_PAD) elems
@@ -280,10 +284,10 @@ void launch_attention(const bf16* q, const bf16* k, const bf16* v,
   const dim3 grid(B * H);
   if (D == 64)
     hipLaunchKernelGGL((attn_kernel<64>), grid, dim3(256), lds, stream, q, k, v,
-                       mask, out, B, H, Lq, Lk, mask_mode, scale);
+                       mask, out, B, H, Lq, Lk, mask_mode, scale, qs, ks, vs);
   else if (D == 128)
     hipLaunchKernelGGL((attn_kernel<128>), grid, dim3(256), lds, stream, q, k, v,
-                       mask, out, B, H, Lq, Lk, mask_mode, scale);
+                       mask, out, B, H, Lq, Lk, mask_mode, scale, qs, ks, vs);
 }
 
 void launch_mfma_probe(const bf16* a, const bf16* b, float* c, hipStream_t stream) {

@@ -22,7 +22,7 @@ void launch_embedding_ln(const long*, const long*, const long*, const T*,
                          const T*, const T*, const T*, const T*, T*, long, int,
                          float, hipStream_t);
 void launch_attention(const bf16*, const bf16*, const bf16*, const bf16*, bf16*,
-                      int, int, int, int, int, int, hipStream_t);
+                      int, int, int, int, int, int, int, int, int, hipStream_t);
 void launch_mfma_probe(const bf16*, const bf16*, float*, hipStream_t);
 void launch_nms_multiclass(const float*, const float*, const long*, float*, int,
                            int, float, float, hipStream_t);
@@ -83,14 +83,26 @@ at::Tensor bias_gelu(const at::Tensor& x, const c10::optional<at::Tensor>& bias)
   return y;
 }
 
+// accepts strided views along dim 1 (fused QKV projections): requires
+// innermost stride 1 and batch stride == L * row-stride (16B-aligned rows).
+static int row_stride_of(const at::Tensor& t, const char* name) {
+  TORCH_CHECK(t.stride(2) == 1, name, ": innermost dim must be dense");
+  const long rs = t.stride(1);
+  TORCH_CHECK(t.stride(0) == t.size(1) * rs, name, ": batch stride mismatch");
+  TORCH_CHECK(rs % 8 == 0, name, ": row stride must be 16B-aligned");
+  return (int)rs;
+}
+
 at::Tensor attention(const at::Tensor& q, const at::Tensor& k,
                      const at::Tensor& v, int64_t heads,
                      const c10::optional<at::Tensor>& mask) {
-  TORCH_CHECK(q.is_cuda() && q.is_contiguous() && k.is_contiguous() && v.is_contiguous());
+  TORCH_CHECK(q.is_cuda());
   TORCH_CHECK(q.scalar_type() == at::kBFloat16, "attention: bf16 only");
   TORCH_CHECK(q.dim() == 3 && k.dim() == 3 && v.dim() == 3, "attention: [B,L,H*D]");
   const int B = (int)q.size(0), Lq = (int)q.size(1), HD = (int)q.size(2);
   const int Lk = (int)k.size(1);
+  const int qs = row_stride_of(q, "q"), kss = row_stride_of(k, "k"),
+            vss = row_stride_of(v, "v");
   const int H = (int)heads;
   const int D = HD / H;
   TORCH_CHECK(HD % H == 0 && (D == 64 || D == 128), "attention: head_dim must be 64/128");
@@ -107,11 +119,11 @@ at::Tensor attention(const at::Tensor& q, const at::Tensor& k,
     else TORCH_CHECK(false, "attention: mask numel must be B*Lk or B*Lq*Lk");
     mptr = (const bf16*)m.data_ptr();
   }
-  auto out = at::empty_like(q);
+  auto out = at::empty({q.size(0), q.size(1), q.size(2)}, q.options());
   launch_attention((const bf16*)q.data_ptr(), (const bf16*)k.data_ptr(),
                    (const bf16*)v.data_ptr(), mptr,
                    (bf16*)out.data_ptr(), B, H, Lq, Lk, D, mask_mode,
-                   cur_stream());
+                   qs, kss, vss, cur_stream());
   return out;
 }
 

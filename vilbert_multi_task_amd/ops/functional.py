@@ -130,17 +130,16 @@ def attention(
         mb = mask_bias
         if mb is not None:
             mb = mb.contiguous()
-        return (
-            ext.attention(q.contiguous(), k.contiguous(), v.contiguous(), num_heads, mb),
-            None,
-        )
+        # q/k/v may be strided views into a fused QKV projection (dim-1
+        # stride 3*HD); the kernel reads strides directly — no copies.
+        return ext.attention(q, k, v, num_heads, mb), None
 
     b, lq, hd = q.shape
     lk = k.shape[1]
     d = hd // num_heads
-    qh = q.view(b, lq, num_heads, d).transpose(1, 2)
-    kh = k.view(b, lk, num_heads, d).transpose(1, 2)
-    vh = v.view(b, lk, num_heads, d).transpose(1, 2)
+    qh = q.reshape(b, lq, num_heads, d).transpose(1, 2)
+    kh = k.reshape(b, lk, num_heads, d).transpose(1, 2)
+    vh = v.reshape(b, lk, num_heads, d).transpose(1, 2)
     scale = 1.0 / math.sqrt(d)
     scores = torch.matmul(qh, kh.transpose(-1, -2)) * scale
     if mask_bias is not None:
