@@ -50,8 +50,12 @@ def main() -> None:
     local_rank = int(os.environ.get("LOCAL_RANK", "0"))
     distributed = world > 1
     if distributed:
-        torch.cuda.set_device(local_rank)
-        torch.distributed.init_process_group(backend="nccl")
+        if torch.cuda.is_available():
+            torch.cuda.set_device(local_rank)
+            backend = "nccl"  # RCCL on ROCm
+        else:
+            backend = "gloo"  # CPU smoke of the distributed path
+        torch.distributed.init_process_group(backend=backend)
     n_gpus = world if distributed else args.gpus
 
     device = f"cuda:{local_rank}" if torch.cuda.is_available() else "cpu"
