@@ -27,6 +27,9 @@ void launch_mfma_probe(const bf16*, const bf16*, float*, hipStream_t);
 void launch_nms_multiclass(const float*, const float*, const long*, float*, int,
                            int, float, float, hipStream_t);
 void launch_tr16_probe(short*, int, hipStream_t);
+template <typename T>
+void launch_roi_align(const T*, const float*, T*, int, int, int, int, int, int,
+                      int, float, int, hipStream_t);
 
 namespace {
 
@@ -173,6 +176,34 @@ at::Tensor tr16_probe(const at::Tensor& dummy, int64_t mode) {
   return out;
 }
 
+at::Tensor roi_align(const at::Tensor& input, const at::Tensor& rois,
+                     int64_t ph, int64_t pw, double spatial_scale,
+                     int64_t sampling_ratio) {
+  TORCH_CHECK(input.is_cuda() && input.dim() == 4 && input.is_contiguous());
+  TORCH_CHECK(rois.is_cuda() && rois.dim() == 2 && rois.size(1) == 5);
+  TORCH_CHECK(rois.scalar_type() == at::kFloat);
+  const int N = (int)input.size(0), C = (int)input.size(1),
+            H = (int)input.size(2), W = (int)input.size(3);
+  const int R = (int)rois.size(0);
+  auto out = at::empty({R, (long)C, ph, pw}, input.options());
+  if (input.scalar_type() == at::kFloat) {
+    launch_roi_align<float>(input.data_ptr<float>(),
+                            rois.contiguous().data_ptr<float>(),
+                            out.data_ptr<float>(), N, C, H, W, R, (int)ph,
+                            (int)pw, (float)spatial_scale, (int)sampling_ratio,
+                            cur_stream());
+  } else if (input.scalar_type() == at::kBFloat16) {
+    launch_roi_align<bf16>((const bf16*)input.data_ptr(),
+                           rois.contiguous().data_ptr<float>(),
+                           (bf16*)out.data_ptr(), N, C, H, W, R, (int)ph,
+                           (int)pw, (float)spatial_scale, (int)sampling_ratio,
+                           cur_stream());
+  } else {
+    TORCH_CHECK(false, "roi_align: dtype must be f32/bf16");
+  }
+  return out;
+}
+
 at::Tensor nms_multiclass(const at::Tensor& boxes, const at::Tensor& scores,
                           double iou_thr, double score_thr) {
   TORCH_CHECK(boxes.is_cuda() && boxes.scalar_type() == at::kFloat);
@@ -199,6 +230,7 @@ TORCH_LIBRARY(vilbert_amd, m) {
   m.def("mfma_probe(Tensor a, Tensor b) -> Tensor");
   m.def("tr16_probe(Tensor dummy, int mode) -> Tensor");
   m.def("nms_multiclass(Tensor boxes, Tensor scores, float iou_thr, float score_thr) -> Tensor");
+  m.def("roi_align(Tensor input, Tensor rois, int ph, int pw, float spatial_scale, int sampling_ratio) -> Tensor");
 }
 
 TORCH_LIBRARY_IMPL(vilbert_amd, CUDA, m) {
@@ -209,4 +241,5 @@ TORCH_LIBRARY_IMPL(vilbert_amd, CUDA, m) {
   m.impl("mfma_probe", mfma_probe);
   m.impl("tr16_probe", tr16_probe);
   m.impl("nms_multiclass", nms_multiclass);
+  m.impl("roi_align", roi_align);
 }

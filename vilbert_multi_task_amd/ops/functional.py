@@ -181,3 +181,76 @@ def embedding_ln(
         + torch.nn.functional.embedding(type_ids, type_w)
     )
     return torch.nn.functional.layer_norm(e, (e.shape[-1],), ln_w, ln_b, eps)
+
+
+# --------------------------------------------------------------------------
+# RoIAlign (legacy / non-aligned semantics, matching the detector call site)
+# --------------------------------------------------------------------------
+
+def roi_align(
+    x: torch.Tensor,
+    rois: torch.Tensor,
+    out_size: int,
+    spatial_scale: float,
+    sampling_ratio: int = 2,
+) -> torch.Tensor:
+    """x [N,C,H,W]; rois [R,5] (batch_idx,x1,y1,x2,y2) image coords.
+    Returns [R,C,out,out]."""
+    if x.is_cuda and _want_hip(x):
+        ext = _load_extension()
+        return ext.roi_align(
+            x.contiguous(), rois.float().contiguous(), out_size, out_size,
+            spatial_scale, sampling_ratio,
+        )
+    return _roi_align_ref(x, rois, out_size, spatial_scale, sampling_ratio)
+
+
+def _roi_align_ref(x, rois, out_size, spatial_scale, sampling_ratio):
+    """Plain-PyTorch oracle (loops over ROIs; test/CPU scale only)."""
+    N, C, H, W = x.shape
+    R = rois.shape[0]
+    out = x.new_zeros(R, C, out_size, out_size)
+    for r in range(R):
+        n = int(rois[r, 0])
+        x1, y1, x2, y2 = (rois[r, 1:] * spatial_scale).tolist()
+        rw = max(x2 - x1, 1.0)
+        rh = max(y2 - y1, 1.0)
+        bw = rw / out_size
+        bh = rh / out_size
+        gw = sampling_ratio if sampling_ratio > 0 else max(1, math.ceil(bw))
+        gh = sampling_ratio if sampling_ratio > 0 else max(1, math.ceil(bh))
+        ys, xs = [], []
+        for ph in range(out_size):
+            for iy in range(gh):
+                ys.append(y1 + ph * bh + (iy + 0.5) * bh / gh)
+        for pw in range(out_size):
+            for ix in range(gw):
+                xs.append(x1 + pw * bw + (ix + 0.5) * bw / gw)
+        yt = torch.tensor(ys, dtype=x.dtype)
+        xt = torch.tensor(xs, dtype=x.dtype)
+        valid_y = (yt >= -1.0) & (yt <= H)
+        valid_x = (xt >= -1.0) & (xt <= W)
+        yc = yt.clamp(0, H - 1)
+        xc = xt.clamp(0, W - 1)
+        y0 = yc.floor().long().clamp(0, H - 1)
+        x0 = xc.floor().long().clamp(0, W - 1)
+        y1i = (y0 + 1).clamp(max=H - 1)
+        x1i = (x0 + 1).clamp(max=W - 1)
+        ly = (yc - y0.to(x.dtype)).clamp(0, 1)
+        lx = (xc - x0.to(x.dtype)).clamp(0, 1)
+        img = x[n]  # [C,H,W]
+        # gather taps [C, len(ys), len(xs)]
+        v00 = img[:, y0][:, :, x0]
+        v01 = img[:, y0][:, :, x1i]
+        v10 = img[:, y1i][:, :, x0]
+        v11 = img[:, y1i][:, :, x1i]
+        ly_ = ly.view(1, -1, 1)
+        lx_ = lx.view(1, 1, -1)
+        vals = (
+            (1 - ly_) * ((1 - lx_) * v00 + lx_ * v01)
+            + ly_ * ((1 - lx_) * v10 + lx_ * v11)
+        )
+        vals = vals * (valid_y.view(1, -1, 1) & valid_x.view(1, 1, -1)).to(x.dtype)
+        vals = vals.view(C, out_size, gh, out_size, gw)
+        out[r] = vals.mean(dim=(2, 4))
+    return out

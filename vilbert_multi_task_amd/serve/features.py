@@ -78,30 +78,29 @@ class PrecomputedFeatureProvider:
         return out
 
 
-def tensorize_regions(infos: Sequence[Dict]) -> Dict[str, torch.Tensor]:
+def tensorize_regions(
+    infos: Sequence[Dict], num_regions: int = NUM_REGIONS
+) -> Dict[str, torch.Tensor]:
     """Per-image region tensors (worker.py:421-457): returns features
-    [N,101,2048], spatials [N,101,5], image_mask [N,101]."""
-    feats_l, spat_l = [], []
-    for info in infos:
-        f = info["features"]  # [num_boxes, 2048]
-        num = int(info["num_boxes"])
+    [N,101,2048], spatials [N,101,5], image_mask [N,101]. Images with fewer
+    detector boxes than num_regions-1 are zero-padded and masked (the
+    reference always has exactly 100 boxes -> all-ones mask, same behavior)."""
+    feat_dim = infos[0]["features"].shape[1]
+    n = len(infos)
+    features = torch.zeros(n, num_regions, feat_dim)
+    spatials = torch.zeros(n, num_regions, 5)
+    image_mask = torch.zeros(n, num_regions, dtype=torch.long)
+    for i, info in enumerate(infos):
+        f = info["features"].float()
+        num = min(int(info["num_boxes"]), num_regions - 1)
         w, h = float(info["image_width"]), float(info["image_height"])
-        g = f.mean(dim=0, keepdim=True)  # global mean-pooled feature
-        feats = torch.cat([g, f], dim=0)  # prepend -> 101 x 2048
-        bbox = info["bbox"].float()
+        features[i, 0] = f[:num].mean(dim=0)  # global mean-pooled feature
+        features[i, 1 : num + 1] = f[:num]
+        bbox = info["bbox"].float()[:num]
         x1, y1 = bbox[:, 0] / w, bbox[:, 1] / h
         x2, y2 = bbox[:, 2] / w, bbox[:, 3] / h
         area = (x2 - x1) * (y2 - y1)
-        sp = torch.stack([x1, y1, x2, y2, area], dim=1)
-        gsp = torch.tensor([[0.0, 0.0, 1.0, 1.0, 1.0]])  # worker.py:443
-        spat = torch.cat([gsp, sp], dim=0)
-        feats_l.append(feats)
-        spat_l.append(spat)
-    features = torch.stack(feats_l)
-    spatials = torch.stack(spat_l)
-    n, r = features.shape[0], features.shape[1]
-    return {
-        "features": features,
-        "spatials": spatials,
-        "image_mask": torch.ones(n, r, dtype=torch.long),
-    }
+        spatials[i, 0] = torch.tensor([0.0, 0.0, 1.0, 1.0, 1.0])  # worker.py:443
+        spatials[i, 1 : num + 1] = torch.stack([x1, y1, x2, y2, area], dim=1)
+        image_mask[i, : num + 1] = 1
+    return {"features": features, "spatials": spatials, "image_mask": image_mask}
