@@ -97,9 +97,23 @@ def create_app(
         async def _stop_hub():
             await hub.stop()
 
+    static_index = os.path.join(os.path.dirname(__file__), "static", "index.html")
+
     @app.get("/", response_class=HTMLResponse)
     async def index() -> str:
+        if os.path.exists(static_index):
+            with open(static_index, encoding="utf-8") as f:
+                return f.read()
         return _INDEX_HTML
+
+    @app.get("/media/{path:path}")
+    async def media(path: str):
+        from fastapi.responses import FileResponse
+
+        full = os.path.normpath(os.path.join(media_root, path))
+        if not full.startswith(os.path.normpath(media_root)) or not os.path.isfile(full):
+            return JSONResponse({"error": "not found"}, status_code=404)
+        return FileResponse(full)
 
     @app.post("/")
     async def submit(request: Request):

@@ -167,10 +167,14 @@ class ServingWorker:
             )
         if spec.decode == DecodeFamily.GROUNDING:
             info = r.infos[0]
-            return decode_grounding(
+            result = decode_grounding(
                 r.task_id, outputs, row, batch["spatials"],
                 info["image_width"], info["image_height"],
             )
+            imgs = self.render_grounding_images(r.image_paths[0], result["boxes"])
+            if imgs:
+                result["image_name_list"] = imgs
+            return result
         if spec.decode == DecodeFamily.BINARY:
             return decode_answer_task(
                 r.task_id, outputs, row // 2, self.vqa_vocab, self.gqa_vocab
@@ -180,10 +184,13 @@ class ServingWorker:
     # ------------------------------------------------------------------
     def process_once(self, max_wait_s: float = 0.0) -> int:
         """One drain-batch-infer-respond cycle; returns #requests served."""
+        from ..utils.trace import RequestTrace, get_metrics
+
         reqs = self.gather_batch(max_wait_s)
         if not reqs:
             return 0
-        t0 = time.time()
+        m = get_metrics()
+        trace = RequestTrace()
         for r in reqs:
             r.qa_id = self.db.create_question(
                 r.task_id, r.question, r.image_paths, r.socket_id
@@ -192,30 +199,71 @@ class ServingWorker:
                 self.push, r.socket_id, {"terminal": "Processing request..."}
             )
         try:
-            batch = self.build_batch(reqs)
-            outputs = self.runner.run(batch)
+            with trace.stage("build_batch"):
+                batch = self.build_batch(reqs)
+            with trace.stage("forward"):
+                outputs = self.runner.run(batch)
         except Exception:
             traceback.print_exc()
             for r in reqs:
                 self.broker.nack(r.delivery.msg_id)
+                m.requests_total.labels(str(r.task_id), "error").inc()
             return 0
         served = 0
-        for r in reqs:
-            try:
-                result = self.decode_request(r, outputs, batch)
-                self.db.save_answer(r.qa_id, json.dumps(result))
-                log_to_terminal(self.push, r.socket_id, {"terminal": json.dumps(result)})
-                log_to_terminal(self.push, r.socket_id, {"result": json.dumps(result)})
-                log_to_terminal(
-                    self.push, r.socket_id, {"terminal": "Completed VilBERT task"}
-                )
-                self.broker.ack(r.delivery.msg_id)
-                served += 1
-            except Exception:
-                traceback.print_exc()
-                self.broker.nack(r.delivery.msg_id)
-        _ = time.time() - t0  # per-batch wall time (worker.py:657-658 analogue)
+        with trace.stage("decode"):
+            for r in reqs:
+                try:
+                    result = self.decode_request(r, outputs, batch)
+                    self.db.save_answer(r.qa_id, json.dumps(result))
+                    log_to_terminal(self.push, r.socket_id, {"terminal": json.dumps(result)})
+                    log_to_terminal(self.push, r.socket_id, {"result": json.dumps(result)})
+                    log_to_terminal(
+                        self.push, r.socket_id, {"terminal": "Completed VilBERT task"}
+                    )
+                    self.broker.ack(r.delivery.msg_id)
+                    m.requests_total.labels(str(r.task_id), "ok").inc()
+                    served += 1
+                except Exception:
+                    traceback.print_exc()
+                    self.broker.nack(r.delivery.msg_id)
+                    m.requests_total.labels(str(r.task_id), "error").inc()
+        rows = sum(r.num_rows for r in reqs)
+        m.batch_rows.observe(rows)
+        m.request_latency.observe(trace.total_ms() / 1e3)
+        m.queue_depth.set(self.broker.depth(self.queue))
+        trace.report(
+            "serving_batch", requests=len(reqs), rows=rows, served=served,
+            tasks=[r.task_id for r in reqs],
+        )
         return served
+
+    def render_grounding_images(self, image_path: str, boxes) -> List[str]:
+        """Draw top-k grounding boxes into result JPEGs
+        (worker.py:591-600 contract: one image per box under
+        media/refer_expressions_task/, PIL instead of cv2)."""
+        import os
+
+        if not os.path.exists(image_path):
+            return []
+        try:
+            from PIL import Image, ImageDraw
+
+            out_dir = os.path.join("media", "refer_expressions_task")
+            os.makedirs(out_dir, exist_ok=True)
+            names = []
+            base = Image.open(image_path).convert("RGB")
+            for box in boxes:
+                img = base.copy()
+                ImageDraw.Draw(img).rectangle(
+                    [box[0], box[1], box[2], box[3]], outline=(255, 0, 0), width=3
+                )
+                name = f"{uuid.uuid4().hex}.jpg"
+                img.save(os.path.join(out_dir, name), "JPEG")
+                names.append(os.path.join(out_dir, name))
+            return names
+        except Exception:
+            traceback.print_exc()
+            return []
 
     def run_forever(self, poll_s: float = 0.02) -> None:
         while True:
