@@ -30,6 +30,7 @@
 //   C: lane l, reg r hold C[row = (l>>4)*4 + r][col = l&15]
 
 #include "common.h"
+#include <cstdlib>
 
 #define ATTN_MAX_L 128  // max Lq/Lk this kernel serves (serving shapes <=101+pad)
 
@@ -50,17 +51,22 @@ DEV bf16x8 lds_b128(const char* p) {
 
 DEV void lds_store_b128(char* p, uint4 v) { *reinterpret_cast<uint4*>(p) = v; }
 
-template <int D>
+template <int D, bool KGLOBAL>
 __global__ __launch_bounds__(256) void attn_kernel(
     const bf16* __restrict__ q, const bf16* __restrict__ kg,
     const bf16* __restrict__ vg, const bf16* __restrict__ mask,
     bf16* __restrict__ out, int B, int H, int Lq, int Lk, int mask_mode,
-    float scale, int qs, int ks, int vs) {
+    float scale, int qs, int ks, int vs, int nsplit) {
   // qs/ks/vs: row strides (elems) of q/k/v — [B,L,H*D] views into a fused
   // QKV (or KV) projection pass without any copy (stride 3*HD / 2*HD).
+  // nsplit: query-stripe splits per (b,h) — each workgroup stages K/V and
+  // owns 4 stripes (one per wave), so Lq=101 runs as 2 workgroups instead
+  // of serializing 7 stripes over 4 waves (K/V re-staged per split; 64 KB
+  // re-read stays L2-hot).
   constexpr int KCH = D / 8;  // 16B chunks per row
   const int HD = H * D;
-  const int bh = blockIdx.x;
+  const int bh = blockIdx.x / nsplit;
+  const int split = blockIdx.x % nsplit;
   const int b = bh / H;
   const int h = bh % H;
   const int tid = threadIdx.x;
@@ -71,8 +77,10 @@ __global__ __launch_bounds__(256) void attn_kernel(
   const int NT = LK_PAD / 16;  // <= 8 score tiles per stripe
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
+  // KGLOBAL: K fragments read straight from global (L2-resident slice) —
+  // no K_lds, LDS drops to V+P = 48 KB for D=128 -> 3 workgroups/CU.
   char* K_lds = smem;                          // [LK_PAD][D] bf16, XOR-swizzled
-  char* V_lds = K_lds + LK_PAD * D * 2;        // [D/16][LK_PAD/4][4][16] bf16
+  char* V_lds = smem + (KGLOBAL ? 0 : LK_PAD * D * 2);  // VT [D][LK_PAD]
   char* P_lds = V_lds + LK_PAD * D * 2 + wid * 16 * LK_PAD * 2;  // per-wave
 
   // ---- stage K (swizzled row-major) and V (tr-readable subtiles) ---------
@@ -85,10 +93,12 @@ __global__ __launch_bounds__(256) void attn_kernel(
     for (int r = r0; r < LK_PAD; r += rows_per_pass) {
       uint4 kraw = {0, 0, 0, 0}, vraw = {0, 0, 0, 0};
       if (r < Lk) {
-        kraw = *reinterpret_cast<const uint4*>(kg + kbase0 + (long)r * ks + c * 8);
+        if (!KGLOBAL)
+          kraw = *reinterpret_cast<const uint4*>(kg + kbase0 + (long)r * ks + c * 8);
         vraw = *reinterpret_cast<const uint4*>(vg + vbase0 + (long)r * vs + c * 8);
       }
-      lds_store_b128(K_lds + r * (D * 2) + ((c * 16) ^ ((r & 7) << 4)), kraw);
+      if (!KGLOBAL)
+        lds_store_b128(K_lds + r * (D * 2) + ((c * 16) ^ ((r & 7) << 4)), kraw);
       // V transposed image [D][LK_PAD] (XOR-swizzled rows): the PV MFMA
       // B-fragment wants per-lane contiguous keys at fixed d, so transpose
       // at staging. Writes are conflict-free: for fixed j, consecutive
@@ -112,7 +122,8 @@ __global__ __launch_bounds__(256) void attn_kernel(
 
   // ---- per-wave stripes of 16 query rows ---------------------------------
   const int nstripes = (Lq + 15) / 16;
-  for (int s = wid; s < nstripes; s += blockDim.x / WAVE) {
+  for (int s = split * (blockDim.x / WAVE) + wid; s < nstripes;
+       s += nsplit * (blockDim.x / WAVE)) {
     const int qrow0 = s * 16;
     // Q A-fragments straight from global
     bf16x8 aq[D / 32];
@@ -131,13 +142,24 @@ __global__ __launch_bounds__(256) void attn_kernel(
     for (int nt = 0; nt < 8; ++nt) {
       if (nt >= NT) break;
       const int key = nt * 16 + (lane & 15);
-      const char* kbase = K_lds + key * (D * 2);
-      const int ksw = (key & 7) << 4;
+      if (KGLOBAL) {
+        const long kb = (long)b * Lk * ks + (long)h * D +
+                        (long)min(key, Lk - 1) * ks + (lane >> 4) * 8;
 #pragma unroll
-      for (int kk = 0; kk < D / 32; ++kk) {
-        const bf16x8 bk = lds_b128(kbase + (((kk * 64) + ((lane >> 4) * 16)) ^ ksw));
-        acc_s[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-            (bf16x8)aq[kk], bk, acc_s[nt], 0, 0, 0);
+        for (int kk = 0; kk < D / 32; ++kk) {
+          const bf16x8 bk = load_bf16x8(kg + kb + kk * 32);
+          acc_s[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              (bf16x8)aq[kk], bk, acc_s[nt], 0, 0, 0);
+        }
+      } else {
+        const char* kbase = K_lds + key * (D * 2);
+        const int ksw = (key & 7) << 4;
+#pragma unroll
+        for (int kk = 0; kk < D / 32; ++kk) {
+          const bf16x8 bk = lds_b128(kbase + (((kk * 64) + ((lane >> 4) * 16)) ^ ksw));
+          acc_s[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              (bf16x8)aq[kk], bk, acc_s[nt], 0, 0, 0);
+        }
       }
     }
 
@@ -279,15 +301,35 @@ void launch_attention(const bf16* q, const bf16* k, const bf16* v,
                       hipStream_t stream) {
   const float scale = 1.0f / sqrtf((float)D);
   const int LK_PAD = (Lk + 31) & ~31;
-  // K + V + 4x per-wave P (all bf16): (2*LK_PAD*D + 4*16*LK_PAD) elems
-  const size_t lds = sizeof(bf16) * (size_t)(2 * LK_PAD * D + 4 * 16 * LK_PAD);
-  const dim3 grid(B * H);
-  if (D == 64)
-    hipLaunchKernelGGL((attn_kernel<64>), grid, dim3(256), lds, stream, q, k, v,
-                       mask, out, B, H, Lq, Lk, mask_mode, scale, qs, ks, vs);
-  else if (D == 128)
-    hipLaunchKernelGGL((attn_kernel<128>), grid, dim3(256), lds, stream, q, k, v,
-                       mask, out, B, H, Lq, Lk, mask_mode, scale, qs, ks, vs);
+  static const int kglobal_env = [] {
+    const char* e = getenv("VILBERT_ATTN_KGLOBAL");
+    return e ? atoi(e) : 0;  // measured: staged K beats L2-global K
+  }();
+  static const int nsplit_env = [] {
+    const char* e = getenv("VILBERT_ATTN_NSPLIT");
+    return e ? atoi(e) : 0;  // 0 = auto
+  }();
+  // measured on MI355X (B=256): nsplit>1 duplicates staging and loses
+  int nsplit = nsplit_env > 0 ? nsplit_env : 1;
+  // (K if staged) + V + 4x per-wave P (all bf16)
+  const size_t lds = sizeof(bf16) *
+      (size_t)((kglobal_env ? 1 : 2) * LK_PAD * D + 4 * 16 * LK_PAD);
+  const dim3 grid(B * H * nsplit);
+  if (D == 64) {
+    if (kglobal_env)
+      hipLaunchKernelGGL((attn_kernel<64, true>), grid, dim3(256), lds, stream, q, k, v,
+                         mask, out, B, H, Lq, Lk, mask_mode, scale, qs, ks, vs, nsplit);
+    else
+      hipLaunchKernelGGL((attn_kernel<64, false>), grid, dim3(256), lds, stream, q, k, v,
+                         mask, out, B, H, Lq, Lk, mask_mode, scale, qs, ks, vs, nsplit);
+  } else if (D == 128) {
+    if (kglobal_env)
+      hipLaunchKernelGGL((attn_kernel<128, true>), grid, dim3(256), lds, stream, q, k, v,
+                         mask, out, B, H, Lq, Lk, mask_mode, scale, qs, ks, vs, nsplit);
+    else
+      hipLaunchKernelGGL((attn_kernel<128, false>), grid, dim3(256), lds, stream, q, k, v,
+                         mask, out, B, H, Lq, Lk, mask_mode, scale, qs, ks, vs, nsplit);
+  }
 }
 
 void launch_mfma_probe(const bf16* a, const bf16* b, float* c, hipStream_t stream) {
