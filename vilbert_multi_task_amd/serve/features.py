@@ -27,13 +27,19 @@ from ..tasks import FEATURE_DIM, MAX_SEQ_LENGTH, NUM_REGIONS
 class SyntheticFeatureProvider:
     """Deterministic pseudo-features keyed by image path (no detector)."""
 
-    def __init__(self, num_boxes: int = 100, feat_dim: int = FEATURE_DIM):
+    def __init__(self, num_boxes: int = 100, feat_dim: int = FEATURE_DIM, cache_size: int = 1024):
         self.num_boxes = num_boxes
         self.feat_dim = feat_dim
+        self._cache: Dict[str, Dict] = {}
+        self._cache_size = cache_size
 
     def extract(self, image_paths: Sequence[str]) -> List[Dict]:
         out = []
         for p in image_paths:
+            hit = self._cache.get(p)
+            if hit is not None:
+                out.append(hit)
+                continue
             seed = int.from_bytes(hashlib.sha1(p.encode()).digest()[:4], "little")
             g = torch.Generator().manual_seed(seed)
             feats = torch.randn(self.num_boxes, self.feat_dim, generator=g).abs()
@@ -43,15 +49,16 @@ class SyntheticFeatureProvider:
             bbox = torch.cat([c - wh / 2, c + wh / 2], dim=1).clamp(min=0)
             bbox[:, 2].clamp_(max=w)
             bbox[:, 3].clamp_(max=h)
-            out.append(
-                {
-                    "features": feats,
-                    "bbox": bbox,
-                    "image_width": w,
-                    "image_height": h,
-                    "num_boxes": self.num_boxes,
-                }
-            )
+            info = {
+                "features": feats,
+                "bbox": bbox,
+                "image_width": w,
+                "image_height": h,
+                "num_boxes": self.num_boxes,
+            }
+            if len(self._cache) < self._cache_size:
+                self._cache[p] = info
+            out.append(info)
         return out
 
 
