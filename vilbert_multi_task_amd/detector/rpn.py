@@ -31,8 +31,8 @@ def generate_anchors(
 
 def shift_anchors(cell: torch.Tensor, stride: int, h: int, w: int) -> torch.Tensor:
     """Tile [A,4] cell anchors over an h x w grid -> [h*w*A, 4]."""
-    xs = torch.arange(w, dtype=torch.float32) * stride
-    ys = torch.arange(h, dtype=torch.float32) * stride
+    xs = torch.arange(w, dtype=torch.float32, device=cell.device) * stride
+    ys = torch.arange(h, dtype=torch.float32, device=cell.device) * stride
     yy, xx = torch.meshgrid(ys, xs, indexing="ij")
     shifts = torch.stack([xx, yy, xx, yy], dim=-1).reshape(-1, 1, 4)
     return (shifts + cell.view(1, -1, 4)).reshape(-1, 4)

@@ -44,6 +44,9 @@ def load():
         attention=ns.attention,
         embedding_ln=ns.embedding_ln,
         nms_multiclass=ns.nms_multiclass,
+        roi_align=ns.roi_align,
+        mfma_probe=ns.mfma_probe,
+        tr16_probe=ns.tr16_probe,
         lib_path=path,
     )
     return _LOADED
