@@ -232,3 +232,15 @@ def test_nms_multiclass_vs_cpu(ext):
         keep_ref = _nms_cpu(boxes, scores[:, c], 0.5)
         keep_gpu = sorted(torch.nonzero(out[:, c] > 0).flatten().tolist())
         assert keep_gpu == keep_ref, f"class {c}"
+
+
+def test_linear_bias_gelu_fused(ext):
+    """hipBLASLt epilogue-fused GEMM+bias+GELU vs the erf oracle (tanh-form
+    GELU in the epilogue: ~3e-3 divergence, below bf16 resolution here)."""
+    x = _rand_bf16(37 * 8, 768, seed=70)
+    w = _rand_bf16(3072, 768, seed=71, scale=0.03)
+    b = _rand_bf16(3072, seed=72, scale=0.1)
+    y = torch.ops.vilbert_amd.linear_bias_gelu(x, w, b)
+    ref = torch.nn.functional.gelu(x.float() @ w.float().T + b.float())
+    err = (y.float() - ref).abs() - ref.abs() / 64
+    assert err.max() < 3e-2, err.max().item()

@@ -37,8 +37,7 @@ class SimpleClassifier(nn.Module):
         self.decoder = nn.Linear(hid, out_dim)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        h = torch.nn.functional.linear(x, self.dense.weight)
-        h = F_ops.bias_gelu(h, self.dense.bias)
+        h = F_ops.linear_bias_gelu(x, self.dense.weight, self.dense.bias)
         h = self.layer_norm(h)
         return self.decoder(h)
 
@@ -54,8 +53,7 @@ class LMHead(nn.Module):
         self.decoder_bias = nn.Parameter(torch.zeros(config.vocab_size))
 
     def forward(self, t: torch.Tensor) -> torch.Tensor:
-        h = torch.nn.functional.linear(t, self.transform.weight)
-        h = F_ops.bias_gelu(h, self.transform.bias)
+        h = F_ops.linear_bias_gelu(t, self.transform.weight, self.transform.bias)
         h = self.layer_norm(h)
         return torch.nn.functional.linear(h, self.decoder_weight, self.decoder_bias)
 

@@ -168,8 +168,7 @@ class FeedForward(nn.Module):
         self.layer_norm = FusedLayerNorm(hidden, eps)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        h = torch.nn.functional.linear(x, self.intermediate.weight)  # bias fused into gelu
-        h = F_ops.bias_gelu(h, self.intermediate.bias)
+        h = F_ops.linear_bias_gelu(x, self.intermediate.weight, self.intermediate.bias)
         return self.layer_norm(self.dropout(self.output(h)), residual=x)
 
 

@@ -27,6 +27,8 @@ void launch_mfma_probe(const bf16*, const bf16*, float*, hipStream_t);
 void launch_nms_multiclass(const float*, const float*, const long*, float*, int,
                            int, float, float, hipStream_t);
 void launch_tr16_probe(short*, int, hipStream_t);
+int hipblaslt_linear_gelu(const void*, const void*, const void*, void*, long,
+                          long, long, void*, size_t, hipStream_t);
 template <typename T>
 void launch_roi_align(const T*, const float*, T*, int, int, int, int, int, int,
                       int, float, int, hipStream_t);
@@ -176,6 +178,25 @@ at::Tensor tr16_probe(const at::Tensor& dummy, int64_t mode) {
   return out;
 }
 
+at::Tensor linear_bias_gelu(const at::Tensor& x, const at::Tensor& w,
+                            const at::Tensor& bias) {
+  TORCH_CHECK(x.is_cuda() && x.scalar_type() == at::kBFloat16, "linear_bias_gelu: bf16");
+  TORCH_CHECK(w.dim() == 2 && bias.dim() == 1 && w.size(0) == bias.size(0));
+  auto xc = x.contiguous();
+  const long K = x.size(-1), N = w.size(0);
+  const long M = x.numel() / K;
+  TORCH_CHECK(w.size(1) == K);
+  auto sizes = x.sizes().vec();
+  sizes.back() = N;
+  auto y = at::empty(sizes, x.options());
+  auto ws = at::empty({32L * 1024 * 1024}, x.options().dtype(at::kByte));
+  int rc = hipblaslt_linear_gelu(
+      xc.data_ptr(), w.contiguous().data_ptr(), bias.contiguous().data_ptr(),
+      y.data_ptr(), M, N, K, ws.data_ptr(), 32L * 1024 * 1024, cur_stream());
+  TORCH_CHECK(rc == 0, "hipblaslt_linear_gelu failed (no algo)");
+  return y;
+}
+
 at::Tensor roi_align(const at::Tensor& input, const at::Tensor& rois,
                      int64_t ph, int64_t pw, double spatial_scale,
                      int64_t sampling_ratio) {
@@ -231,6 +252,7 @@ TORCH_LIBRARY(vilbert_amd, m) {
   m.def("tr16_probe(Tensor dummy, int mode) -> Tensor");
   m.def("nms_multiclass(Tensor boxes, Tensor scores, float iou_thr, float score_thr) -> Tensor");
   m.def("roi_align(Tensor input, Tensor rois, int ph, int pw, float spatial_scale, int sampling_ratio) -> Tensor");
+  m.def("linear_bias_gelu(Tensor x, Tensor w, Tensor bias) -> Tensor");
 }
 
 TORCH_LIBRARY_IMPL(vilbert_amd, CUDA, m) {
@@ -242,4 +264,5 @@ TORCH_LIBRARY_IMPL(vilbert_amd, CUDA, m) {
   m.impl("tr16_probe", tr16_probe);
   m.impl("nms_multiclass", nms_multiclass);
   m.impl("roi_align", roi_align);
+  m.impl("linear_bias_gelu", linear_bias_gelu);
 }

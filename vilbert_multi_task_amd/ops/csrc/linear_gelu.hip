@@ -1,0 +1,119 @@
+// hipBLASLt GEMM with fused bias+GELU epilogue.
+//
+// The FFN intermediate (768->3072 / 1024->1024, SURVEY.md §2.2 FFN row) is
+// GEMM -> bias -> GELU; unfused that costs a full extra HBM round trip of
+// the intermediate tensor plus a kernel launch (bias_gelu was 6.7% of the
+// serving step, profiles/r01_serving_b256_kernels.md). Here the epilogue
+// runs inside the hipBLASLt kernel (HIPBLASLT_EPILOGUE_GELU_BIAS).
+//
+// Note: hipBLASLt's GELU is the tanh approximation; BERT's reference is the
+// erf form. Max divergence ~3e-3 absolute — below bf16 resolution of these
+// activations; the fp32 oracle tests compare with that tolerance.
+//
+// y[M,N] = gelu(x[M,K] @ w[N,K]^T + b[N]) computed as column-major
+// C[N,M] = op(A=w, T)[N,K] x op(B=x, N)[K,M].
+
+#include <hip/hip_runtime.h>
+#include <hipblaslt/hipblaslt.h>
+
+#include <map>
+#include <mutex>
+#include <stdexcept>
+#include <tuple>
+
+#define HIPBLASLT_CHECK(x)                                          \
+  do {                                                              \
+    hipblasStatus_t s_ = (x);                                       \
+    if (s_ != HIPBLAS_STATUS_SUCCESS)                               \
+      throw std::runtime_error("hipblaslt error " + std::to_string(s_) + \
+                               " at " #x);                          \
+  } while (0)
+
+namespace {
+
+struct Plan {
+  hipblasLtMatmulDesc_t op{};
+  hipblasLtMatrixLayout_t a{}, b{}, c{};
+  hipblasLtMatmulAlgo_t algo{};
+  bool has_algo = false;
+};
+
+hipblasLtHandle_t handle_once() {
+  static hipblasLtHandle_t h = [] {
+    hipblasLtHandle_t t;
+    HIPBLASLT_CHECK(hipblasLtCreate(&t));
+    return t;
+  }();
+  return h;
+}
+
+std::map<std::tuple<long, long, long>, Plan>& plan_cache() {
+  static std::map<std::tuple<long, long, long>, Plan> m;
+  return m;
+}
+std::mutex& plan_mu() {
+  static std::mutex m;
+  return m;
+}
+
+Plan& get_plan(long M, long N, long K, void* workspace, size_t ws_bytes) {
+  auto key = std::make_tuple(M, N, K);
+  auto& cache = plan_cache();
+  auto it = cache.find(key);
+  if (it != cache.end()) return it->second;
+
+  Plan p;
+  HIPBLASLT_CHECK(hipblasLtMatmulDescCreate(&p.op, HIPBLAS_COMPUTE_32F, HIP_R_32F));
+  hipblasOperation_t ta = HIPBLAS_OP_T, tb = HIPBLAS_OP_N;
+  HIPBLASLT_CHECK(hipblasLtMatmulDescSetAttribute(
+      p.op, HIPBLASLT_MATMUL_DESC_TRANSA, &ta, sizeof(ta)));
+  HIPBLASLT_CHECK(hipblasLtMatmulDescSetAttribute(
+      p.op, HIPBLASLT_MATMUL_DESC_TRANSB, &tb, sizeof(tb)));
+  hipblasLtEpilogue_t epi = HIPBLASLT_EPILOGUE_GELU_BIAS;
+  HIPBLASLT_CHECK(hipblasLtMatmulDescSetAttribute(
+      p.op, HIPBLASLT_MATMUL_DESC_EPILOGUE, &epi, sizeof(epi)));
+  // A = w [K,N] col-major view of row-major [N,K], opA = T -> [N,K]
+  HIPBLASLT_CHECK(hipblasLtMatrixLayoutCreate(&p.a, HIP_R_16BF, K, N, K));
+  // B = x [K,M] col-major view of row-major [M,K]
+  HIPBLASLT_CHECK(hipblasLtMatrixLayoutCreate(&p.b, HIP_R_16BF, K, M, K));
+  // C = y [N,M] col-major view of row-major [M,N]
+  HIPBLASLT_CHECK(hipblasLtMatrixLayoutCreate(&p.c, HIP_R_16BF, N, M, N));
+
+  hipblasLtMatmulPreference_t pref;
+  HIPBLASLT_CHECK(hipblasLtMatmulPreferenceCreate(&pref));
+  HIPBLASLT_CHECK(hipblasLtMatmulPreferenceSetAttribute(
+      pref, HIPBLASLT_MATMUL_PREF_MAX_WORKSPACE_BYTES, &ws_bytes, sizeof(ws_bytes)));
+  hipblasLtMatmulHeuristicResult_t results[4];
+  int found = 0;
+  HIPBLASLT_CHECK(hipblasLtMatmulAlgoGetHeuristic(
+      handle_once(), p.op, p.a, p.b, p.c, p.c, pref, 4, results, &found));
+  hipblasLtMatmulPreferenceDestroy(pref);
+  if (found > 0) {
+    p.algo = results[0].algo;
+    p.has_algo = true;
+  }
+  auto r = cache.emplace(key, p);
+  return r.first->second;
+}
+
+}  // namespace
+
+// Returns 0 on success, nonzero on failure (caller falls back to unfused).
+int hipblaslt_linear_gelu(const void* x, const void* w, const void* bias,
+                          void* y, long M, long N, long K, void* workspace,
+                          size_t ws_bytes, hipStream_t stream) {
+  try {
+    std::lock_guard<std::mutex> lock(plan_mu());
+    Plan& p = get_plan(M, N, K, workspace, ws_bytes);
+    if (!p.has_algo) return 1;
+    HIPBLASLT_CHECK(hipblasLtMatmulDescSetAttribute(
+        p.op, HIPBLASLT_MATMUL_DESC_BIAS_POINTER, &bias, sizeof(bias)));
+    float alpha = 1.0f, beta = 0.0f;
+    HIPBLASLT_CHECK(hipblasLtMatmul(
+        handle_once(), p.op, &alpha, w, p.a, x, p.b, &beta, y, p.c, y, p.c,
+        &p.algo, workspace, ws_bytes, stream));
+    return 0;
+  } catch (const std::exception&) {
+    return 1;
+  }
+}

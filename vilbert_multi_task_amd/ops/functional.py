@@ -102,6 +102,23 @@ def bias_gelu(x: torch.Tensor, bias: Optional[torch.Tensor]) -> torch.Tensor:
     return torch.nn.functional.gelu(x)
 
 
+_LINEAR_GELU_OK = True
+
+
+def linear_bias_gelu(x: torch.Tensor, w: torch.Tensor, bias: torch.Tensor) -> torch.Tensor:
+    """y = gelu(x @ w.T + bias) — hipBLASLt fused epilogue on GPU (tanh-form
+    GELU, ~3e-3 from erf, below bf16 resolution); unfused erf path otherwise."""
+    global _LINEAR_GELU_OK
+    if _LINEAR_GELU_OK and _want_hip(x, w, bias):
+        ext = _load_extension()
+        try:
+            return ext.linear_bias_gelu(x, w, bias)
+        except RuntimeError:
+            _LINEAR_GELU_OK = False  # no algo for this arch/shape: fall back
+    h = torch.nn.functional.linear(x, w)
+    return bias_gelu(h, bias)
+
+
 # --------------------------------------------------------------------------
 # Multi-head scaled-dot-product attention with additive mask.
 # Flattened [B, L, H*D] layout so the GPU path needs no transpose copies.
