@@ -37,16 +37,14 @@ def load():
             f"{path} not built; run `python -m vilbert_multi_task_amd.ops.build`"
         )
     torch.ops.load_library(path)
-    ns = torch.ops.vilbert_amd
-    _LOADED = types.SimpleNamespace(
-        residual_layer_norm=ns.residual_layer_norm,
-        bias_gelu=ns.bias_gelu,
-        attention=ns.attention,
-        embedding_ln=ns.embedding_ln,
-        nms_multiclass=ns.nms_multiclass,
-        roi_align=ns.roi_align,
-        mfma_probe=ns.mfma_probe,
-        tr16_probe=ns.tr16_probe,
-        lib_path=path,
-    )
+
+    class _Ext:
+        """Proxies every registered vilbert_amd op (no manual listing)."""
+
+        lib_path = path
+
+        def __getattr__(self, name):
+            return getattr(torch.ops.vilbert_amd, name)
+
+    _LOADED = _Ext()
     return _LOADED
