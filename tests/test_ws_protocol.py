@@ -1,0 +1,97 @@
+"""MinimalWebSocketProtocol test: raw RFC6455 client against uvicorn running
+the demo app (the offline image has no websockets/wsproto, so this protocol
+class IS the realtime channel — serve/ws_protocol.py)."""
+
+import base64
+import json
+import os
+import socket
+import struct
+import threading
+import time
+
+import pytest
+
+uvicorn = pytest.importorskip("uvicorn")
+
+from vilbert_multi_task_amd.serve.app import create_app
+
+
+def _ws_send(sock, msg: str) -> None:
+    p = msg.encode()
+    mask = os.urandom(4)
+    sock.sendall(bytes([0x81, 0x80 | len(p)]) + mask +
+                 bytes(b ^ mask[i % 4] for i, b in enumerate(p)))
+
+
+def _ws_recv(sock) -> str:
+    h = sock.recv(2)
+    ln = h[1] & 0x7F
+    if ln == 126:
+        ln = struct.unpack(">H", sock.recv(2))[0]
+    d = b""
+    while len(d) < ln:
+        d += sock.recv(ln - len(d))
+    return d.decode()
+
+
+@pytest.fixture
+def server(tmp_path):
+    app = create_app(
+        db_path=str(tmp_path / "db.sqlite3"),
+        queue_path=str(tmp_path / "q.sqlite3"),
+        media_root=str(tmp_path / "media"),
+        hub_port=0,
+    )
+    config = uvicorn.Config(
+        app, host="127.0.0.1", port=0, log_level="error",
+        ws="vilbert_multi_task_amd.serve.ws_protocol:MinimalWebSocketProtocol",
+    )
+    srv = uvicorn.Server(config)
+    th = threading.Thread(target=srv.run, daemon=True)
+    th.start()
+    for _ in range(100):
+        if srv.started:
+            break
+        time.sleep(0.05)
+    port = srv.servers[0].sockets[0].getsockname()[1]
+    yield app, port
+    srv.should_exit = True
+    th.join(5)
+
+
+def test_handshake_and_push(server):
+    app, port = server
+    s = socket.create_connection(("127.0.0.1", port), timeout=10)
+    key = base64.b64encode(os.urandom(16)).decode()
+    s.sendall(
+        (f"GET /chat/ HTTP/1.1\r\nHost: x\r\nUpgrade: websocket\r\n"
+         f"Connection: Upgrade\r\nSec-WebSocket-Key: {key}\r\n"
+         f"Sec-WebSocket-Version: 13\r\n\r\n").encode()
+    )
+    resp = s.recv(4096)
+    assert resp.startswith(b"HTTP/1.1 101"), resp[:80]
+    # accept-key per RFC6455
+    import hashlib
+
+    expect = base64.b64encode(
+        hashlib.sha1((key + "258EAFA5-E914-47DA-95CA-C5AB0DC85B11").encode()).digest()
+    ).decode()
+    assert expect.encode() in resp
+    _ws_send(s, "sockP")
+    time.sleep(0.3)
+    # push from the app side -> client receives a frame
+    app.state.push.publish("sockP", {"terminal": "hello-ws"})
+    s.settimeout(10)
+    msg = json.loads(_ws_recv(s))
+    assert msg == {"terminal": "hello-ws"}
+    s.close()
+
+
+def test_bad_handshake_rejected(server):
+    _, port = server
+    s = socket.create_connection(("127.0.0.1", port), timeout=10)
+    s.sendall(b"GET /chat/ HTTP/1.1\r\nHost: x\r\nUpgrade: websocket\r\nConnection: Upgrade\r\n\r\n")
+    resp = s.recv(4096)
+    assert b"400" in resp.split(b"\r\n")[0]
+    s.close()

@@ -28,7 +28,10 @@ def run_app(args) -> None:
         db_path=args.db, queue_path=args.queue, media_root=args.media,
         hub_port=args.hub_port,
     )
-    uvicorn.run(app, host=args.host, port=args.port, log_level="info")
+    uvicorn.run(
+        app, host=args.host, port=args.port, log_level="info",
+        ws="vilbert_multi_task_amd.serve.ws_protocol:MinimalWebSocketProtocol",
+    )
 
 
 def build_worker(args):
