@@ -78,3 +78,10 @@ def test_websocket_push_roundtrip(client):
         app.state.push.publish("sockWS", {"terminal": "hello"})
         msg = json.loads(ws.receive_text())
         assert msg == {"terminal": "hello"}
+
+
+def test_demo_images_endpoint(client, tmp_path):
+    c, _ = client
+    r = c.get("/demo_images/")
+    assert r.status_code == 200
+    assert "images" in r.json()

@@ -106,6 +106,21 @@ def create_app(
                 return f.read()
         return _INDEX_HTML
 
+    @app.get("/demo_images/")
+    async def demo_images():
+        """Random demo gallery (views.py:64-81: 6 random COCO test2014 images,
+        falling back to static samples when the directory is absent)."""
+        import random
+
+        coco_dir = os.path.join(media_root, "test2014")
+        if os.path.isdir(coco_dir):
+            pool = [f for f in os.listdir(coco_dir) if f.lower().endswith((".jpg", ".png"))]
+            picks = random.sample(pool, min(6, len(pool)))
+            return JSONResponse({"images": [f"test2014/{p}" for p in picks]})
+        demo_dir = os.path.join(media_root, "demo")
+        pool = sorted(os.listdir(demo_dir)) if os.path.isdir(demo_dir) else []
+        return JSONResponse({"images": [f"demo/{p}" for p in pool[:6]]})
+
     @app.get("/media/{path:path}")
     async def media(path: str):
         from fastapi.responses import FileResponse
