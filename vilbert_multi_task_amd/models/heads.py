@@ -114,10 +114,20 @@ class VILBertForVLTasks(nn.Module):
         else:
             vil_binary_prediction = fused.new_zeros(1, 2)
         vil_tri_prediction = self.vil_tri_prediction(fused)
-        vision_prediction = self.vision_prediction(self.dropout(v))
         vision_logit = self.vision_logit(self.dropout(v))
-        linguisic_prediction = self.linguistic_prediction(self.dropout(t))
-        linguisic_logit = self.linguistic_logit(self.dropout(t))
+        if getattr(self, "skip_unused_heads", False) and not self.training:
+            # Serving fast path (worker opt-in, NOT the benchmark default):
+            # the demo decode (worker.py:295-386) never reads the masked-LM,
+            # masked-region or per-token-relevance heads — skip their GEMMs
+            # (the LM head alone is rows x 30522 x 768 and a >1 GB logits
+            # write at large batch). Placeholders keep the 10-tuple arity.
+            vision_prediction = v.new_zeros(b, v.shape[1], 0)
+            linguisic_prediction = t.new_zeros(b, t.shape[1], 0)
+            linguisic_logit = t.new_zeros(b, t.shape[1], 0)
+        else:
+            vision_prediction = self.vision_prediction(self.dropout(v))
+            linguisic_prediction = self.linguistic_prediction(self.dropout(t))
+            linguisic_logit = self.linguistic_logit(self.dropout(t))
 
         return (
             vil_prediction,
