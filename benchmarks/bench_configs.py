@@ -104,7 +104,8 @@ def bench_mixed_serving(device: str, requests: int = 512, max_batch: int = 64):
     from vilbert_multi_task_amd.serve.worker import ServingWorker
 
     m, cfg = _model(device)
-    runner = GraphRunner(m, device=device, use_graphs=device.startswith("cuda"))
+    runner = GraphRunner(m, device=device, use_graphs=device.startswith("cuda"),
+                         serving_fast=True)
     with tempfile.TemporaryDirectory() as td:
         broker = Broker(os.path.join(td, "q.sqlite3"))
         db = Database(os.path.join(td, "db.sqlite3"))

@@ -158,3 +158,25 @@ def test_tokenizer_padding_contract():
     assert all(m == 0 for m in mask[sep_pos + 1 :])
     assert all(m == 1 for m in mask[: sep_pos + 1])
     assert seg == [0] * 37
+
+
+def test_serving_fast_mode_same_answers(tmp_path, tiny_model, tiny_config):
+    """skip_unused_heads must not change any decodable output."""
+    import torch
+    from vilbert_multi_task_amd.data.synthetic import forward_args, synthetic_batch
+
+    batch = synthetic_batch(
+        2, seq_len=20, regions=36, feat_dim=tiny_config.v_feature_size,
+        vocab_size=tiny_config.vocab_size,
+    )
+    with torch.no_grad():
+        full = tiny_model(*forward_args(batch))
+    tiny_model.skip_unused_heads = True
+    try:
+        with torch.no_grad():
+            fast = tiny_model(*forward_args(batch))
+    finally:
+        tiny_model.skip_unused_heads = False
+    for i in (0, 1, 2, 3, 4, 6):  # every head the decode reads
+        assert torch.equal(full[i], fast[i]), i
+    assert fast[7].shape[-1] == 0  # placeholders

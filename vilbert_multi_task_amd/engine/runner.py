@@ -28,10 +28,14 @@ class GraphRunner:
         device: str = "cuda",
         dtype: torch.dtype = torch.bfloat16,
         use_graphs: bool = True,
+        serving_fast: bool = False,
     ):
         self.model = model.eval()
         if hasattr(model, "prepare_for_serving"):
             model.prepare_for_serving()
+        if serving_fast:
+            # skip heads the demo decode never reads (models/heads.py)
+            model.skip_unused_heads = True
         self.seq_len = seq_len
         self.regions = regions
         self.feat_dim = feat_dim

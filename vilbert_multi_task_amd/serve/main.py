@@ -58,7 +58,7 @@ def build_worker(args):
     device = "cuda" if torch.cuda.is_available() else "cpu"
     if device == "cuda":
         model = model.to(device=device, dtype=torch.bfloat16)
-    runner = GraphRunner(model, device=device, use_graphs=device == "cuda")
+    runner = GraphRunner(model, device=device, use_graphs=device == "cuda", serving_fast=True)
 
     provider = None
     if args.detector:
