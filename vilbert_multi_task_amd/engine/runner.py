@@ -36,6 +36,8 @@ class GraphRunner:
         if serving_fast:
             # skip heads the demo decode never reads (models/heads.py)
             model.skip_unused_heads = True
+        if use_graphs and device.startswith("cuda") and hasattr(model, "bert"):
+            model.bert.overlap_streams = True  # dual-stream text/vision segments
         self.seq_len = seq_len
         self.regions = regions
         self.feat_dim = feat_dim

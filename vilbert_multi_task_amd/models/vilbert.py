@@ -293,6 +293,10 @@ class ViLBertModel(nn.Module):
         )
         self.t_pooler = Pooler(c.hidden_size, c.bi_hidden_size)
         self.v_pooler = Pooler(c.v_hidden_size, c.bi_hidden_size)
+        # serving: run independent text/vision layer segments on two HIP
+        # streams (captured into the hipGraph) — engine/runner.py opts in
+        self.overlap_streams = False
+        self._side_stream = None
 
     @staticmethod
     def _extend_mask(mask: torch.Tensor, dtype: torch.dtype) -> torch.Tensor:
@@ -348,7 +352,33 @@ class ViLBertModel(nn.Module):
 
         attn_data: List = []
         t_idx = v_idx = 0
-        for ci, (v_stop, t_stop) in enumerate(zip(c.v_biattention_id, c.t_biattention_id)):
+        use_overlap = (
+            self.overlap_streams
+            and t.is_cuda
+            and not self.training
+            and not output_all_attention_masks
+        )
+        if use_overlap and self._side_stream is None:
+            self._side_stream = torch.cuda.Stream()
+
+        def _run_segment(v, t, v_stop, t_stop, v_idx, t_idx):
+            """One inter-connect segment: the pending vision and text layers
+            are data-independent — run vision on the side stream."""
+            if use_overlap and v_idx < v_stop:
+                cur = torch.cuda.current_stream()
+                side = self._side_stream
+                side.wait_stream(cur)
+                with torch.cuda.stream(side):
+                    while v_idx < v_stop:
+                        v = self.v_layers[v_idx](v, v_mask, False)[0]
+                        v_idx += 1
+                while t_idx < t_stop:
+                    t = self.t_layers[t_idx](t, t_mask, False)[0]
+                    t_idx += 1
+                cur.wait_stream(side)
+                if not torch.cuda.is_current_stream_capturing():
+                    v.record_stream(cur)
+                return v, t, v_idx, t_idx
             while v_idx < v_stop:
                 v, p = self.v_layers[v_idx](v, v_mask, output_all_attention_masks)
                 if output_all_attention_masks:
@@ -359,21 +389,18 @@ class ViLBertModel(nn.Module):
                 if output_all_attention_masks:
                     attn_data.append({"type": "t_self", "layer": t_idx, "probs": p})
                 t_idx += 1
+            return v, t, v_idx, t_idx
+
+        for ci, (v_stop, t_stop) in enumerate(zip(c.v_biattention_id, c.t_biattention_id)):
+            v, t, v_idx, t_idx = _run_segment(v, t, v_stop, t_stop, v_idx, t_idx)
             t, v, (p_tv, p_vt) = self.c_layers[ci](
                 t, v, t_mask, v_mask, co_tv, co_vt, output_all_attention_masks
             )
             if output_all_attention_masks:
                 attn_data.append({"type": "co", "layer": ci, "probs_tv": p_tv, "probs_vt": p_vt})
-        while v_idx < len(self.v_layers):
-            v, p = self.v_layers[v_idx](v, v_mask, output_all_attention_masks)
-            if output_all_attention_masks:
-                attn_data.append({"type": "v_self", "layer": v_idx, "probs": p})
-            v_idx += 1
-        while t_idx < len(self.t_layers):
-            t, p = self.t_layers[t_idx](t, t_mask, output_all_attention_masks)
-            if output_all_attention_masks:
-                attn_data.append({"type": "t_self", "layer": t_idx, "probs": p})
-            t_idx += 1
+        v, t, v_idx, t_idx = _run_segment(
+            v, t, len(self.v_layers), len(self.t_layers), v_idx, t_idx
+        )
 
         pooled_t = self.t_pooler(t)
         pooled_v = self.v_pooler(v)
