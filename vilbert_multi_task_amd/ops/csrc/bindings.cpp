@@ -7,6 +7,9 @@
 #include <hip/hip_runtime.h>
 #include <torch/library.h>
 
+#include <map>
+#include <mutex>
+
 #include <hip/hip_bf16.h>
 
 using bf16 = __hip_bfloat16;
@@ -178,6 +181,24 @@ at::Tensor tr16_probe(const at::Tensor& dummy, int64_t mode) {
   return out;
 }
 
+// hipBLASLt workspaces: ONE persistent buffer per stream, never freed.
+// A per-call at::empty workspace is freed while the async GEMM still uses
+// it; with dual-stream hipGraph capture the allocator can hand the same
+// block to the other stream's GEMM -> replay-time race (measured as a
+// stable 2e-2 drift between replays). Keyed by stream so concurrent
+// streams never share.
+static void* ws_for_stream(hipStream_t stream, const at::TensorOptions& opts,
+                           size_t bytes) {
+  static std::mutex mu;
+  static std::map<hipStream_t, at::Tensor> cache;
+  std::lock_guard<std::mutex> lock(mu);
+  auto it = cache.find(stream);
+  if (it == cache.end()) {
+    it = cache.emplace(stream, at::empty({(long)bytes}, opts.dtype(at::kByte))).first;
+  }
+  return it->second.data_ptr();
+}
+
 at::Tensor linear_bias_gelu(const at::Tensor& x, const at::Tensor& w,
                             const at::Tensor& bias) {
   TORCH_CHECK(x.is_cuda() && x.scalar_type() == at::kBFloat16, "linear_bias_gelu: bf16");
@@ -189,10 +210,12 @@ at::Tensor linear_bias_gelu(const at::Tensor& x, const at::Tensor& w,
   auto sizes = x.sizes().vec();
   sizes.back() = N;
   auto y = at::empty(sizes, x.options());
-  auto ws = at::empty({32L * 1024 * 1024}, x.options().dtype(at::kByte));
+  constexpr size_t kWs = 32L * 1024 * 1024;
+  hipStream_t stream = cur_stream();
+  void* ws = ws_for_stream(stream, x.options(), kWs);
   int rc = hipblaslt_linear_gelu(
       xc.data_ptr(), w.contiguous().data_ptr(), bias.contiguous().data_ptr(),
-      y.data_ptr(), M, N, K, ws.data_ptr(), 32L * 1024 * 1024, cur_stream());
+      y.data_ptr(), M, N, K, ws, kWs, stream);
   TORCH_CHECK(rc == 0, "hipblaslt_linear_gelu failed (no algo)");
   return y;
 }
