@@ -67,32 +67,19 @@ def test_graph_capture_replay_matches_eager(base_model_pair):
         assert torch.allclose(a, b, atol=3e-2, rtol=3e-2), f"output {i}"
 
 
-def test_graph_replay_stable(base_model_pair):
-    """Replay-to-replay outputs agree to single-ulp bf16 noise. Bitwise
-    equality does not hold with dual-stream capture: hipBLASLt split-K
-    kernels accumulate via atomics, and cross-stream timing perturbs the
-    fp32 reduction order (measured: max drift == 1 bf16 ulp of the largest
-    logits; with overlap_streams=False replay IS bitwise deterministic)."""
+def test_graph_replay_deterministic(base_model_pair):
+    """Default (single-stream) capture replays bitwise-deterministically.
+    (VILBERT_STREAM_OVERLAP=1 trades this for +1-5% throughput: dual-stream
+    timing perturbs hipBLASLt split-K atomic accumulation order -> bf16-ulp
+    drift; measured and deliberately not the default.)"""
     _, m_gpu = base_model_pair
     from vilbert_multi_task_amd.engine.runner import GraphRunner
 
     runner = GraphRunner(m_gpu, use_graphs=True)
     batch = synthetic_batch(4, seed=13)
-    a = runner.run(batch)[0].float().clone()
-    b = runner.run(batch)[0].float().clone()
-    assert torch.allclose(a, b, atol=6e-2, rtol=3e-2), (a - b).abs().max()
-    assert (a - b).abs().mean() < 3e-3
-
-    # single-stream capture: bitwise-deterministic replay
-    m_gpu.bert.overlap_streams = False
-    try:
-        r2 = GraphRunner(m_gpu, use_graphs=True)
-        m_gpu.bert.overlap_streams = False
-        x = r2.run(batch)[0].clone()
-        y = r2.run(batch)[0].clone()
-        assert torch.equal(x, y)
-    finally:
-        m_gpu.bert.overlap_streams = True
+    a = runner.run(batch)[0].clone()
+    b = runner.run(batch)[0].clone()
+    assert torch.equal(a, b)
 
 
 def test_training_step_bf16(base_model_pair):
