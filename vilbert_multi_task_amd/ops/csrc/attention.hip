@@ -33,6 +33,14 @@
 #include <cstdlib>
 
 #define ATTN_MAX_L 128  // max Lq/Lk this kernel serves (serving shapes <=101+pad)
+// LDS XOR swizzle: include the row's HIGH bits so sibling rows 8 apart land
+// on different 16B slots. PMC on MI355X showed the VT transpose writes were
+// a 16-way conflict (19.6M SQ_LDS_BANK_CONFLICT per dispatch): for a fixed
+// element index j, the 16 staging threads write d = c*8+j — d&7 == j for
+// all of them, so a (row&7)-only swizzle degenerates. (row>>3)&7 varies
+// with c and spreads them; it also kills the residual 2-way read conflicts
+// on K/P (16 consecutive rows now map to 16 distinct slots).
+#define SWZ(row) (((((row) & 7) ^ (((row) >> 3) & 7))) << 4)
 
 typedef __attribute__((ext_vector_type(4))) short s4_vec;
 typedef __attribute__((address_space(3))) s4_vec* lds_v4s;
@@ -98,7 +106,7 @@ __global__ __launch_bounds__(256) void attn_kernel(
         vraw = *reinterpret_cast<const uint4*>(vg + vbase0 + (long)r * vs + c * 8);
       }
       if (!KGLOBAL)
-        lds_store_b128(K_lds + r * (D * 2) + ((c * 16) ^ ((r & 7) << 4)), kraw);
+        lds_store_b128(K_lds + r * (D * 2) + ((c * 16) ^ SWZ(r)), kraw);
       // V transposed image [D][LK_PAD] (XOR-swizzled rows): the PV MFMA
       // B-fragment wants per-lane contiguous keys at fixed d, so transpose
       // at staging. Writes are conflict-free: for fixed j, consecutive
@@ -113,7 +121,7 @@ __global__ __launch_bounds__(256) void attn_kernel(
         for (int j = 0; j < 8; ++j) {
           const int d = c * 8 + j;
           *reinterpret_cast<short*>(
-              V_lds + d * (LK_PAD * 2) + ((r * 2) ^ ((d & 7) << 4))) = vv.s[j];
+              V_lds + d * (LK_PAD * 2) + ((r * 2) ^ SWZ(d))) = vv.s[j];
         }
       }
     }
@@ -153,7 +161,7 @@ __global__ __launch_bounds__(256) void attn_kernel(
         }
       } else {
         const char* kbase = K_lds + key * (D * 2);
-        const int ksw = (key & 7) << 4;
+        const int ksw = SWZ(key);
 #pragma unroll
         for (int kk = 0; kk < D / 32; ++kk) {
           const bf16x8 bk = lds_b128(kbase + (((kk * 64) + ((lane >> 4) * 16)) ^ ksw));
@@ -204,7 +212,7 @@ __global__ __launch_bounds__(256) void attn_kernel(
     for (int r = 0; r < 4; ++r) {
       const int prow = (lane >> 4) * 4 + r;
       char* prow_base = P_lds + prow * (LK_PAD * 2);
-      const int psw = (prow & 7) << 4;
+      const int psw = SWZ(prow);
 #pragma unroll
       for (int nt = 0; nt < 8; ++nt) {
         if (nt >= NT) break;
@@ -222,7 +230,7 @@ __global__ __launch_bounds__(256) void attn_kernel(
     for (int nt = 0; nt < D / 16; ++nt) acc_o[nt] = {0.f, 0.f, 0.f, 0.f};
     const int parow = lane & 15;
     const char* pa_base = P_lds + parow * (LK_PAD * 2);
-    const int pasw = (parow & 7) << 4;
+    const int pasw = SWZ(parow);
 #pragma unroll
     for (int kk = 0; kk < 4; ++kk) {  // LK_PAD/32 <= 4
       if (kk * 32 >= LK_PAD) break;
@@ -233,7 +241,7 @@ __global__ __launch_bounds__(256) void attn_kernel(
       for (int nt = 0; nt < D / 16; ++nt) {
         const int d = nt * 16 + (lane & 15);
         const bf16x8 bv =
-            lds_b128(V_lds + d * (LK_PAD * 2) + (keyoff ^ ((d & 7) << 4)));
+            lds_b128(V_lds + d * (LK_PAD * 2) + (keyoff ^ SWZ(d)));
         acc_o[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ap, bv, acc_o[nt], 0, 0, 0);
       }
     }
