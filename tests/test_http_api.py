@@ -43,8 +43,9 @@ def test_submit_enqueues_message(client):
     assert r.status_code == 200
     assert app.state.broker.depth() == 1
     d = app.state.broker.get()[0]
-    # message schema fixed by sender.py:19-24
-    assert set(d.body) == {"image_path", "question", "socket_id", "task_id"}
+    # message schema fixed by sender.py:19-24 (+ our additive trace_id)
+    assert {"image_path", "question", "socket_id", "task_id"} <= set(d.body)
+    assert set(d.body) - {"image_path", "question", "socket_id", "task_id", "trace_id"} == set()
     assert d.body["question"] == "what color is the sky?"  # lowercased (views.py:28)
     assert d.body["task_id"] == "1"
     assert d.body["socket_id"] == "sockA"

@@ -180,3 +180,38 @@ def test_serving_fast_mode_same_answers(tmp_path, tiny_model, tiny_config):
     for i in (0, 1, 2, 3, 4, 6):  # every head the decode reads
         assert torch.equal(full[i], fast[i]), i
     assert fast[7].shape[-1] == 0  # placeholders
+
+
+def test_multi_worker_competing_consumers(tmp_path, tiny_model, tiny_config):
+    """Scale-out model: N workers share one durable queue (SURVEY.md §2.4);
+    every request is served exactly once."""
+    broker_a = Broker(str(tmp_path / "shared.sqlite3"))
+    broker_b = Broker(str(tmp_path / "shared.sqlite3"))
+    db = Database(str(tmp_path / "db2.sqlite3"))
+    push = NullPush()
+
+    def make_worker(br):
+        runner = GraphRunner(
+            tiny_model, device="cpu", use_graphs=False,
+            feat_dim=tiny_config.v_feature_size,
+        )
+        return ServingWorker(
+            runner, br, db, push,
+            provider=SyntheticFeatureProvider(feat_dim=tiny_config.v_feature_size),
+            tokenizer=BertWordPieceTokenizer(vocab_size=tiny_config.vocab_size),
+            vqa_vocab=AnswerVocab(tiny_config.num_labels_vqa),
+            gqa_vocab=AnswerVocab(tiny_config.num_labels_gqa),
+            max_batch_rows=3,
+        )
+
+    wa, wb = make_worker(broker_a), make_worker(broker_b)
+    for i in range(8):
+        vilbert_task(broker_a, [f"/i{i}.jpg"], f"q {i}", 1, f"mw{i}")
+    served = 0
+    for _ in range(10):
+        served += wa.process_once() + wb.process_once()
+        if served >= 8:
+            break
+    assert served == 8
+    assert broker_a.depth() == 0
+    assert len([p for s, p in push.messages if "result" in p]) == 8

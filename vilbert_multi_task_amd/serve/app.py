@@ -142,8 +142,15 @@ def create_app(
             p if os.path.isabs(p) else os.path.join(media_root, p.lstrip("/"))
             for p in image_list
         ]
+        from ..utils.trace import log_json, new_trace_id
+
+        trace_id = new_trace_id()
         _push_local(app, socket_id, {"info": "Task submitted"})
-        vilbert_task(app.state.broker, paths, question, task_id or "1", socket_id)
+        vilbert_task(
+            app.state.broker, paths, question, task_id or "1", socket_id,
+            trace_id=trace_id,
+        )
+        log_json("submit", trace_id=trace_id, task_id=task_id, socket_id=socket_id)
         return HTMLResponse(_INDEX_HTML)
 
     @app.get("/get_task_details/{task_id}/")

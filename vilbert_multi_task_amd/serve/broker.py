@@ -145,13 +145,16 @@ def vilbert_task(
     question: str,
     task_id: int,
     socket_id: str,
+    trace_id: Optional[str] = None,
 ) -> int:
-    """Producer helper mirroring demo/sender.py:10-31's message schema."""
-    return broker.publish(
-        {
-            "image_path": image_paths,
-            "question": question,
-            "socket_id": socket_id,
-            "task_id": str(task_id),
-        }
-    )
+    """Producer helper mirroring demo/sender.py:10-31's message schema
+    (+ an optional trace_id the reference lacks — utils/trace.py)."""
+    body: Dict[str, Any] = {
+        "image_path": image_paths,
+        "question": question,
+        "socket_id": socket_id,
+        "task_id": str(task_id),
+    }
+    if trace_id:
+        body["trace_id"] = trace_id
+    return broker.publish(body)

@@ -234,6 +234,7 @@ class ServingWorker:
         trace.report(
             "serving_batch", requests=len(reqs), rows=rows, served=served,
             tasks=[r.task_id for r in reqs],
+            request_traces=[r.delivery.body.get("trace_id") for r in reqs],
         )
         return served
 
@@ -266,6 +267,20 @@ class ServingWorker:
             return []
 
     def run_forever(self, poll_s: float = 0.02) -> None:
-        while True:
+        """Blocking consume loop; SIGTERM/SIGINT finish the in-flight batch
+        then exit (unacked messages redeliver — at-least-once)."""
+        import signal
+
+        stop = {"flag": False}
+
+        def _sig(_s, _f):
+            stop["flag"] = True
+
+        try:
+            signal.signal(signal.SIGTERM, _sig)
+            signal.signal(signal.SIGINT, _sig)
+        except ValueError:
+            pass  # not the main thread
+        while not stop["flag"]:
             if self.process_once(max_wait_s=poll_s) == 0:
                 time.sleep(poll_s)
