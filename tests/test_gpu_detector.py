@@ -1,7 +1,6 @@
 """GPU detector-stack tests: RoIAlign kernel vs CPU oracle, tiny detector
 forward on GPU, end-to-end serving worker on the hipGraph path."""
 
-import numpy as np
 import pytest
 import torch
 

@@ -2,7 +2,6 @@
 
 import torch
 
-from vilbert_multi_task_amd.config import ViLBertConfig
 from vilbert_multi_task_amd.data import (
     ConceptCapLoaderTrain,
     ConceptCapLoaderVal,

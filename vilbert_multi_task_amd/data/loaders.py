@@ -10,8 +10,7 @@ eval tensor shapes; a real-corpus backend can be slotted in by replacing
 
 from __future__ import annotations
 
-import math
-from typing import Dict, Iterator, Optional
+from typing import Dict, Iterator
 
 import torch
 

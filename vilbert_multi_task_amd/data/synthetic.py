@@ -14,7 +14,7 @@ from typing import Dict
 
 import torch
 
-from ..tasks import FEATURE_DIM, MAX_SEQ_LENGTH, NUM_REGIONS, SPATIAL_DIM
+from ..tasks import FEATURE_DIM, MAX_SEQ_LENGTH, NUM_REGIONS
 
 
 def synthetic_batch(

@@ -22,7 +22,6 @@ real tokens.
 from __future__ import annotations
 
 import os
-import re
 import unicodedata
 from typing import Dict, List, Optional, Tuple
 

@@ -9,8 +9,8 @@ default; `DetectorConfig.tiny()` shrinks everything for CPU tests.
 
 from __future__ import annotations
 
-from dataclasses import dataclass, field
-from typing import Dict, List, Sequence, Tuple
+from dataclasses import dataclass
+from typing import Dict, List, Tuple
 
 import torch
 import torch.nn as nn

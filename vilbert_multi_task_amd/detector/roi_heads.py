@@ -8,7 +8,6 @@ standard FPN level assignment k = floor(4 + log2(sqrt(area)/224)).
 
 from __future__ import annotations
 
-import math
 from typing import List, Tuple
 
 import torch

@@ -12,7 +12,7 @@ On CPU (tests) the runner degrades to eager execution with the same API.
 from __future__ import annotations
 
 import os
-from typing import Dict, Optional, Tuple
+from typing import Dict, Tuple
 
 import torch
 

@@ -15,7 +15,7 @@ with region features projected into the text stream.
 
 from __future__ import annotations
 
-from typing import Optional, Tuple
+from typing import Optional
 
 import torch
 import torch.nn as nn

@@ -40,11 +40,6 @@ namespace {
 
 hipStream_t cur_stream() { return c10::hip::getCurrentHIPStream().stream(); }
 
-void check_same(const at::Tensor& a, const at::Tensor& b, const char* msg) {
-  TORCH_CHECK(a.scalar_type() == b.scalar_type(), msg, ": dtype mismatch");
-  TORCH_CHECK(a.is_cuda() && b.is_cuda(), msg, ": tensors must be on GPU");
-}
-
 at::Tensor residual_layer_norm(const at::Tensor& x,
                                const c10::optional<at::Tensor>& res,
                                const at::Tensor& w, const at::Tensor& b,

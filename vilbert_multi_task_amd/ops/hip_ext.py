@@ -8,7 +8,6 @@ the GPU box) by ``python -m vilbert_multi_task_amd.ops.build`` or by
 from __future__ import annotations
 
 import os
-import types
 
 import torch
 

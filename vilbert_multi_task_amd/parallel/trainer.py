@@ -21,7 +21,7 @@ worker decodes — worker.py:295-386):
 from __future__ import annotations
 
 import os
-from typing import Dict, Optional, Tuple
+from typing import Tuple
 
 import torch
 import torch.nn.functional as F

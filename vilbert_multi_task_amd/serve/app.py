@@ -59,7 +59,7 @@ async def _parse_form(request: Request):
 
 from .broker import Broker, vilbert_task
 from .db import Database
-from .push import PushClient, PushHub, log_to_terminal
+from .push import PushHub, log_to_terminal
 
 _INDEX_HTML = """<!doctype html>
 <html><head><title>ViLBERT Multi-Task Demo (MI355X)</title></head>

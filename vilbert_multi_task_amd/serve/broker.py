@@ -20,7 +20,6 @@ competing-consumers model: N workers polling one queue.
 from __future__ import annotations
 
 import json
-import os
 import sqlite3
 import time
 import uuid

@@ -11,7 +11,6 @@ Tables:
 from __future__ import annotations
 
 import json
-import os
 import sqlite3
 import threading
 import time

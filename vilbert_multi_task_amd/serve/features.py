@@ -21,7 +21,7 @@ from typing import Dict, List, Sequence
 
 import torch
 
-from ..tasks import FEATURE_DIM, MAX_SEQ_LENGTH, NUM_REGIONS
+from ..tasks import FEATURE_DIM, NUM_REGIONS
 
 
 class SyntheticFeatureProvider:
