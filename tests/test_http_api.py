@@ -86,3 +86,15 @@ def test_demo_images_endpoint(client, tmp_path):
     r = c.get("/demo_images/")
     assert r.status_code == 200
     assert "images" in r.json()
+
+
+def test_admin_surface(client):
+    c, _ = client
+    c.post("/", data={"socket_id": "adm", "task_id": "1", "question": "Q?",
+                      "image_list[]": ["demo/x.jpg"]})
+    r = c.get("/admin/")
+    assert r.status_code == 200
+    body = r.json()
+    assert len(body["tasks"]) == 9  # the 9 registry task ids
+    assert body["queue"]["ready"] >= 1
+    assert body["recent_questions"] == []  # worker inserts the row, not the app

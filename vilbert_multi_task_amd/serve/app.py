@@ -121,6 +121,21 @@ def create_app(
         pool = sorted(os.listdir(demo_dir)) if os.path.isdir(demo_dir) else []
         return JSONResponse({"images": [f"demo/{p}" for p in pool[:6]]})
 
+    @app.get("/admin/")
+    async def admin_index():
+        """Read-only admin surface (the reference exposes Django admin list
+        views for Tasks and QuestionAnswer — demo/admin.py:1-34)."""
+        return JSONResponse(
+            {
+                "tasks": app.state.db.list_tasks(),
+                "recent_questions": app.state.db.recent_questions(25),
+                "queue": {
+                    "ready": app.state.broker.depth(),
+                    "dead": app.state.broker.dead_count(),
+                },
+            }
+        )
+
     @app.get("/media/{path:path}")
     async def media(path: str):
         from fastapi.responses import FileResponse
@@ -170,6 +185,7 @@ def create_app(
             dst = os.path.join(media_root, "demo", name)
             with open(dst, "wb") as out:
                 out.write(payload)
+            app.state.db.add_attachment(dst)  # Attachment row (models.py:45-46)
             file_paths.append(dst)
         return JSONResponse({"file_paths": file_paths})
 

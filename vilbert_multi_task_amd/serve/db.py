@@ -28,6 +28,12 @@ CREATE TABLE IF NOT EXISTS tasks (
     num_of_images INTEGER DEFAULT 1,
     example TEXT DEFAULT ''
 );
+CREATE TABLE IF NOT EXISTS attachment (
+    id INTEGER PRIMARY KEY AUTOINCREMENT,
+    created TEXT NOT NULL,
+    modified TEXT NOT NULL,
+    attachment TEXT
+);
 CREATE TABLE IF NOT EXISTS questionanswer (
     id INTEGER PRIMARY KEY AUTOINCREMENT,
     created TEXT NOT NULL,
@@ -108,6 +114,25 @@ class Database:
                 " WHERE id=?",
                 (answer_text, json.dumps(answer_images or []), _now(), qa_id),
             )
+
+    def add_attachment(self, path: str) -> int:
+        """Attachment rows (demo/models.py:45-46): uploaded-file records."""
+        with self._conn() as c:
+            cur = c.execute(
+                "INSERT INTO attachment (created, modified, attachment) VALUES (?,?,?)",
+                (_now(), _now(), path),
+            )
+            return int(cur.lastrowid)
+
+    def recent_questions(self, limit: int = 50) -> List[Dict[str, Any]]:
+        rows = self._conn().execute(
+            "SELECT * FROM questionanswer ORDER BY id DESC LIMIT ?", (limit,)
+        ).fetchall()
+        return [dict(r) for r in rows]
+
+    def list_tasks(self) -> List[Dict[str, Any]]:
+        rows = self._conn().execute("SELECT * FROM tasks ORDER BY unique_id").fetchall()
+        return [dict(r) for r in rows]
 
     def get_question(self, qa_id: int) -> Optional[Dict[str, Any]]:
         row = self._conn().execute(
