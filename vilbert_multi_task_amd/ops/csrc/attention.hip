@@ -59,7 +59,7 @@ DEV bf16x8 lds_b128(const char* p) {
 
 DEV void lds_store_b128(char* p, uint4 v) { *reinterpret_cast<uint4*>(p) = v; }
 
-template <int D, bool KGLOBAL>
+template <int D, bool KGLOBAL, int NTMAX>
 __global__ __launch_bounds__(256) void attn_kernel(
     const bf16* __restrict__ q, const bf16* __restrict__ kg,
     const bf16* __restrict__ vg, const bf16* __restrict__ mask,
@@ -82,7 +82,7 @@ __global__ __launch_bounds__(256) void attn_kernel(
   const int wid = wave_id();
 
   const int LK_PAD = (Lk + 31) & ~31;
-  const int NT = LK_PAD / 16;  // <= 8 score tiles per stripe
+  const int NT = LK_PAD / 16;  // <= NTMAX score tiles per stripe
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
   // KGLOBAL: K fragments read straight from global (L2-resident slice) —
@@ -143,11 +143,11 @@ __global__ __launch_bounds__(256) void attn_kernel(
     }
 
     // ---- S = Q K^T --------------------------------------------------------
-    f32x4 acc_s[8];
+    f32x4 acc_s[NTMAX];
 #pragma unroll
-    for (int nt = 0; nt < 8; ++nt) acc_s[nt] = {0.f, 0.f, 0.f, 0.f};
+    for (int nt = 0; nt < NTMAX; ++nt) acc_s[nt] = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll
-    for (int nt = 0; nt < 8; ++nt) {
+    for (int nt = 0; nt < NTMAX; ++nt) {
       if (nt >= NT) break;
       const int key = nt * 16 + (lane & 15);
       if (KGLOBAL) {
@@ -179,7 +179,7 @@ __global__ __launch_bounds__(256) void attn_kernel(
       const int row = qrow0 + (lane >> 4) * 4 + r;
       float mx = -3.0e38f;
 #pragma unroll
-      for (int nt = 0; nt < 8; ++nt) {
+      for (int nt = 0; nt < NTMAX; ++nt) {
         if (nt >= NT) break;
         const int col = nt * 16 + col0;
         float sv = acc_s[nt][r] * scale;
@@ -197,7 +197,7 @@ __global__ __launch_bounds__(256) void attn_kernel(
       const float mrow = group16_max(mx);
       float sum = 0.f;
 #pragma unroll
-      for (int nt = 0; nt < 8; ++nt) {
+      for (int nt = 0; nt < NTMAX; ++nt) {
         if (nt >= NT) break;
         const int col = nt * 16 + col0;
         const float p = (col < Lk) ? __expf(acc_s[nt][r] - mrow) : 0.f;
@@ -214,7 +214,7 @@ __global__ __launch_bounds__(256) void attn_kernel(
       char* prow_base = P_lds + prow * (LK_PAD * 2);
       const int psw = SWZ(prow);
 #pragma unroll
-      for (int nt = 0; nt < 8; ++nt) {
+      for (int nt = 0; nt < NTMAX; ++nt) {
         if (nt >= NT) break;
         const int col = nt * 16 + col0;
         *reinterpret_cast<short*>(prow_base + ((col * 2) ^ psw)) =
@@ -232,7 +232,7 @@ __global__ __launch_bounds__(256) void attn_kernel(
     const char* pa_base = P_lds + parow * (LK_PAD * 2);
     const int pasw = SWZ(parow);
 #pragma unroll
-    for (int kk = 0; kk < 4; ++kk) {  // LK_PAD/32 <= 4
+    for (int kk = 0; kk < NTMAX / 2; ++kk) {  // LK_PAD/32 <= NTMAX/2
       if (kk * 32 >= LK_PAD) break;
       const bf16x8 ap =
           lds_b128(pa_base + (((kk * 64) + ((lane >> 4) * 16)) ^ pasw));
@@ -323,21 +323,19 @@ void launch_attention(const bf16* q, const bf16* k, const bf16* v,
   const size_t lds = sizeof(bf16) *
       (size_t)((kglobal_env ? 1 : 2) * LK_PAD * D + 4 * 16 * LK_PAD);
   const dim3 grid(B * H * nsplit);
+#define LAUNCH_ATTN(DD, KG, NTM)                                              \
+  hipLaunchKernelGGL((attn_kernel<DD, KG, NTM>), grid, dim3(256), lds, stream, \
+                     q, k, v, mask, out, B, H, Lq, Lk, mask_mode, scale, qs,  \
+                     ks, vs, nsplit)
+  const bool small = LK_PAD <= 64;  // NTMAX=4 halves the accumulator VGPRs
   if (D == 64) {
-    if (kglobal_env)
-      hipLaunchKernelGGL((attn_kernel<64, true>), grid, dim3(256), lds, stream, q, k, v,
-                         mask, out, B, H, Lq, Lk, mask_mode, scale, qs, ks, vs, nsplit);
-    else
-      hipLaunchKernelGGL((attn_kernel<64, false>), grid, dim3(256), lds, stream, q, k, v,
-                         mask, out, B, H, Lq, Lk, mask_mode, scale, qs, ks, vs, nsplit);
+    if (kglobal_env) { if (small) LAUNCH_ATTN(64, true, 4); else LAUNCH_ATTN(64, true, 8); }
+    else             { if (small) LAUNCH_ATTN(64, false, 4); else LAUNCH_ATTN(64, false, 8); }
   } else if (D == 128) {
-    if (kglobal_env)
-      hipLaunchKernelGGL((attn_kernel<128, true>), grid, dim3(256), lds, stream, q, k, v,
-                         mask, out, B, H, Lq, Lk, mask_mode, scale, qs, ks, vs, nsplit);
-    else
-      hipLaunchKernelGGL((attn_kernel<128, false>), grid, dim3(256), lds, stream, q, k, v,
-                         mask, out, B, H, Lq, Lk, mask_mode, scale, qs, ks, vs, nsplit);
+    if (kglobal_env) { if (small) LAUNCH_ATTN(128, true, 4); else LAUNCH_ATTN(128, true, 8); }
+    else             { if (small) LAUNCH_ATTN(128, false, 4); else LAUNCH_ATTN(128, false, 8); }
   }
+#undef LAUNCH_ATTN
 }
 
 void launch_mfma_probe(const bf16* a, const bf16* b, float* c, hipStream_t stream) {
