@@ -52,6 +52,9 @@ class BucketedDataParallel:
         self._buckets: List[_Bucket] = []
         self._param_bucket: Dict[torch.nn.Parameter, _Bucket] = {}
         self._hooks = []
+        # gradient accumulation: defer all reduction to finalize_backward()
+        # (hook-time reduction would all-reduce PARTIAL micro-batch grads)
+        self.defer_reduction = False
         if self.enabled:
             if broadcast_params:
                 for p in module.parameters():
@@ -88,6 +91,8 @@ class BucketedDataParallel:
 
     # ------------------------------------------------------------------
     def _on_grad(self, p: torch.nn.Parameter) -> None:
+        if self.defer_reduction:
+            return
         b = self._param_bucket[p]
         b.pending += 1
         if b.pending == len(b.params):
@@ -119,8 +124,9 @@ class BucketedDataParallel:
         if not self.enabled:
             return
         for b in self._buckets:
-            if b.pending and b.work is None:
-                # partially-filled bucket (unused task heads) or late tail
+            if b.work is None and (b.pending or self.defer_reduction):
+                # deferred (grad accumulation), partially-filled bucket
+                # (unused task heads), or late tail
                 self._reduce_bucket(b)
         for b in self._buckets:
             if b.work is not None:

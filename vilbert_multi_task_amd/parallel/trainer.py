@@ -120,6 +120,9 @@ class MultiTaskTrainer:
         seq_len: int = 37,
         regions: int = 101,
         grad_clip: float = 1.0,
+        warmup_steps: int = 0,
+        total_steps: int = 0,
+        grad_accum: int = 1,
     ):
         self.cfg = cfg
         self.device = device
@@ -131,24 +134,54 @@ class MultiTaskTrainer:
         self.ddp = BucketedDataParallel(self.model, bucket_bytes=bucket_bytes)
         self.opt = torch.optim.AdamW(self.model.parameters(), lr=lr, weight_decay=0.01)
         self.sampler = RoundRobinTaskSampler(rank=rank, world_size=world_size)
+        self.base_lr = lr
+        self.warmup_steps = warmup_steps
+        self.total_steps = total_steps
+        self.grad_accum = max(grad_accum, 1)
+
+    def _lr_at(self, step: int) -> float:
+        """Warmup-linear schedule (the 12-in-1 recipe): linear ramp over
+        warmup_steps, then linear decay to 0 at total_steps (constant when
+        total_steps == 0)."""
+        if self.warmup_steps and step < self.warmup_steps:
+            return self.base_lr * (step + 1) / self.warmup_steps
+        if self.total_steps:
+            frac = max(0.0, 1.0 - (step - self.warmup_steps) / max(
+                self.total_steps - self.warmup_steps, 1))
+            return self.base_lr * frac
+        return self.base_lr
 
     def train_step(self) -> Tuple[str, float]:
+        """One optimizer step = grad_accum micro-batches of ONE task
+        (accumulation stays within a task so the gradient layout is stable
+        for the bucketed all-reduce)."""
         self.model.train()
         dataset = self.sampler.next_task()
-        seed = self.sampler.shard_seed(dataset)
-        batch, targets = make_training_batch(
-            dataset, self.batch_size, self.cfg, seed, self.device,
-            self.seq_len, self.regions,
-        )
         self.ddp.zero_grad()
-        out = self.ddp(*forward_args(batch))
-        loss = task_loss(dataset, out, targets)
-        loss.backward()
-        self.ddp.finalize_backward()
+        self.ddp.defer_reduction = self.grad_accum > 1
+        total = 0.0
+        for micro in range(self.grad_accum):
+            seed = self.sampler.shard_seed(dataset) + micro * 1000003
+            batch, targets = make_training_batch(
+                dataset, self.batch_size, self.cfg, seed, self.device,
+                self.seq_len, self.regions,
+            )
+            out = self.ddp(*forward_args(batch))
+            loss = task_loss(dataset, out, targets) / self.grad_accum
+            if micro < self.grad_accum - 1:
+                # defer the all-reduce hooks' finalize to the last micro-batch
+                loss.backward()
+            else:
+                loss.backward()
+                self.ddp.finalize_backward()
+            total += float(loss.detach())
         if self.grad_clip:
             torch.nn.utils.clip_grad_norm_(self.model.parameters(), self.grad_clip)
+        lr = self._lr_at(self.sampler.state.step)
+        for g in self.opt.param_groups:
+            g["lr"] = lr
         self.opt.step()
-        return dataset, float(loss.detach())
+        return dataset, total
 
     # -- checkpoint (model + optimizer + sampler: SURVEY.md §5) -----------
     def save_checkpoint(self, path: str, upstream_layout: bool = True) -> None:
