@@ -114,6 +114,12 @@ class Broker:
         with self._conn:
             self._conn.execute("DELETE FROM messages WHERE id=?", (msg_id,))
 
+    def ack_many(self, msg_ids: List[int]) -> None:
+        with self._conn:
+            self._conn.executemany(
+                "DELETE FROM messages WHERE id=?", [(i,) for i in msg_ids]
+            )
+
     def nack(self, msg_id: int) -> None:
         """Return a message to ready (reference behavior: unacked messages
         redeliver — worker.py:653-655)."""

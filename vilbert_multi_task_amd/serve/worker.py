@@ -82,6 +82,7 @@ class ServingWorker:
         self.gqa_vocab = gqa_vocab or AnswerVocab(1533)
         self.max_batch_rows = max_batch_rows
         self.queue = queue
+        self._tok_cache: Dict[str, tuple] = {}
 
     # ------------------------------------------------------------------
     def _parse(self, d: Delivery) -> Optional[_Request]:
@@ -135,7 +136,12 @@ class ServingWorker:
         q_rows, mask_rows, seg_rows, task_rows = [], [], [], []
         infos_all: List[Dict] = []
         for r in reqs:
-            ids, mask, seg = self.tokenizer.encode_for_serving(r.question, MAX_SEQ_LENGTH)
+            cached = self._tok_cache.get(r.question)
+            if cached is None:
+                cached = self.tokenizer.encode_for_serving(r.question, MAX_SEQ_LENGTH)
+                if len(self._tok_cache) < 8192:
+                    self._tok_cache[r.question] = cached
+            ids, mask, seg = cached
             r.infos = self.provider.extract(r.image_paths)
             infos_all.extend(r.infos)
             for _ in range(r.num_rows):  # text replicated per image row
@@ -191,10 +197,15 @@ class ServingWorker:
             return 0
         m = get_metrics()
         trace = RequestTrace()
-        for r in reqs:
-            r.qa_id = self.db.create_question(
-                r.task_id, r.question, r.image_paths, r.socket_id
+        with trace.stage("db_insert"):
+            qa_ids = self.db.create_questions(
+                [
+                    (r.task_id, r.question, json.dumps(r.image_paths), r.socket_id)
+                    for r in reqs
+                ]
             )
+        for r, qa in zip(reqs, qa_ids):
+            r.qa_id = qa
             log_to_terminal(
                 self.push, r.socket_id, {"terminal": "Processing request..."}
             )
@@ -210,23 +221,29 @@ class ServingWorker:
                 m.requests_total.labels(str(r.task_id), "error").inc()
             return 0
         served = 0
+        answers = []
+        acks = []
         with trace.stage("decode"):
             for r in reqs:
                 try:
                     result = self.decode_request(r, outputs, batch)
-                    self.db.save_answer(r.qa_id, json.dumps(result))
-                    log_to_terminal(self.push, r.socket_id, {"terminal": json.dumps(result)})
-                    log_to_terminal(self.push, r.socket_id, {"result": json.dumps(result)})
+                    payload = json.dumps(result)
+                    answers.append((r.qa_id, payload))
+                    log_to_terminal(self.push, r.socket_id, {"terminal": payload})
+                    log_to_terminal(self.push, r.socket_id, {"result": payload})
                     log_to_terminal(
                         self.push, r.socket_id, {"terminal": "Completed VilBERT task"}
                     )
-                    self.broker.ack(r.delivery.msg_id)
+                    acks.append(r.delivery.msg_id)
                     m.requests_total.labels(str(r.task_id), "ok").inc()
                     served += 1
                 except Exception:
                     traceback.print_exc()
                     self.broker.nack(r.delivery.msg_id)
                     m.requests_total.labels(str(r.task_id), "error").inc()
+        with trace.stage("db_commit"):
+            self.db.save_answers(answers)
+            self.broker.ack_many(acks)
         rows = sum(r.num_rows for r in reqs)
         m.batch_rows.observe(rows)
         m.request_latency.observe(trace.total_ms() / 1e3)
