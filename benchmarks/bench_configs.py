@@ -107,9 +107,14 @@ def bench_mixed_serving(device: str, requests: int = 512, max_batch: int = 64):
     runner = GraphRunner(m, device=device, use_graphs=device.startswith("cuda"),
                          serving_fast=True)
     with tempfile.TemporaryDirectory() as td:
+        from vilbert_multi_task_amd.serve.features import SyntheticFeatureProvider
+
         broker = Broker(os.path.join(td, "q.sqlite3"))
         db = Database(os.path.join(td, "db.sqlite3"))
-        worker = ServingWorker(runner, broker, db, NullPush(), max_batch_rows=max_batch)
+        worker = ServingWorker(
+            runner, broker, db, NullPush(), max_batch_rows=max_batch,
+            provider=SyntheticFeatureProvider(device=device),
+        )
         tasks = [1, 15, 13, 11, 12]
         n = 0
         i = 0

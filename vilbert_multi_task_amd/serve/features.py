@@ -27,11 +27,15 @@ from ..tasks import FEATURE_DIM, NUM_REGIONS
 class SyntheticFeatureProvider:
     """Deterministic pseudo-features keyed by image path (no detector)."""
 
-    def __init__(self, num_boxes: int = 100, feat_dim: int = FEATURE_DIM, cache_size: int = 1024):
+    def __init__(self, num_boxes: int = 100, feat_dim: int = FEATURE_DIM,
+                 cache_size: int = 1024, device: str = "cpu"):
         self.num_boxes = num_boxes
         self.feat_dim = feat_dim
         self._cache: Dict[str, Dict] = {}
         self._cache_size = cache_size
+        self.device = device  # "cuda": features cached on-GPU, so batch
+        # assembly (tensorize_regions) runs on the GPU — the serving boxes
+        # have weak CPUs and the 60+ MB host-side stack dominated the batch
 
     def extract(self, image_paths: Sequence[str]) -> List[Dict]:
         out = []
@@ -49,6 +53,9 @@ class SyntheticFeatureProvider:
             bbox = torch.cat([c - wh / 2, c + wh / 2], dim=1).clamp(min=0)
             bbox[:, 2].clamp_(max=w)
             bbox[:, 3].clamp_(max=h)
+            if self.device != "cpu":
+                feats = feats.to(self.device)
+                bbox = bbox.to(self.device)
             info = {
                 "features": feats,
                 "bbox": bbox,
@@ -97,19 +104,21 @@ def tensorize_regions(
     nb = {int(i["num_boxes"]) for i in infos}
     if nb == {num_regions - 1}:
         # fast path (the common serving case: every image has exactly 100
-        # boxes) — batched tensor ops instead of a per-image Python loop
+        # boxes) — batched tensor ops on the features' own device (GPU when
+        # the provider caches there; detector features are already on-GPU)
+        dev = infos[0]["features"].device
         f = torch.stack([i["features"] for i in infos]).float()  # [N,100,F]
-        bbox = torch.stack([i["bbox"] for i in infos]).float()   # [N,100,4]
+        bbox = torch.stack([i["bbox"].to(dev) for i in infos]).float()
         wh = torch.tensor(
-            [[i["image_width"], i["image_height"]] for i in infos]
+            [[i["image_width"], i["image_height"]] for i in infos], device=dev
         ).view(n, 1, 2)
-        features = torch.empty(n, num_regions, feat_dim)
+        features = torch.empty(n, num_regions, feat_dim, device=dev)
         features[:, 0] = f.mean(dim=1)
         features[:, 1:] = f
         norm = bbox / wh.repeat(1, 1, 2)
         area = (norm[..., 2] - norm[..., 0]) * (norm[..., 3] - norm[..., 1])
-        spatials = torch.empty(n, num_regions, 5)
-        spatials[:, 0] = torch.tensor([0.0, 0.0, 1.0, 1.0, 1.0])
+        spatials = torch.empty(n, num_regions, 5, device=dev)
+        spatials[:, 0] = torch.tensor([0.0, 0.0, 1.0, 1.0, 1.0], device=dev)
         spatials[:, 1:, :4] = norm
         spatials[:, 1:, 4] = area
         return {

@@ -66,7 +66,7 @@ def build_worker(args):
 
         provider = DetectorFeatureProvider(device=device)
     else:
-        provider = SyntheticFeatureProvider()
+        provider = SyntheticFeatureProvider(device=device)
 
     return ServingWorker(
         runner,
