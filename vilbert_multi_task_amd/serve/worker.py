@@ -192,11 +192,12 @@ class ServingWorker:
         """One drain-batch-infer-respond cycle; returns #requests served."""
         from ..utils.trace import RequestTrace, get_metrics
 
-        reqs = self.gather_batch(max_wait_s)
+        trace = RequestTrace()
+        with trace.stage("gather"):
+            reqs = self.gather_batch(max_wait_s)
         if not reqs:
             return 0
         m = get_metrics()
-        trace = RequestTrace()
         with trace.stage("db_insert"):
             qa_ids = self.db.create_questions(
                 [
