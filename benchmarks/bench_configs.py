@@ -116,14 +116,20 @@ def bench_mixed_serving(device: str, requests: int = 512, max_batch: int = 64):
             provider=SyntheticFeatureProvider(device=device),
         )
         tasks = [1, 15, 13, 11, 12]
-        n = 0
-        i = 0
-        while n < requests:
-            t = tasks[i % len(tasks)]
-            imgs = ["/a.jpg", "/b.jpg"] if t == 12 else ["/a.jpg"]
-            vilbert_task(broker, imgs, f"question {i}", t, f"s{i}")
-            n += 1
-            i += 1
+
+        def enqueue(count, tag):
+            i = 0
+            while i < count:
+                t = tasks[i % len(tasks)]
+                imgs = ["/a.jpg", "/b.jpg"] if t == 12 else ["/a.jpg"]
+                vilbert_task(broker, imgs, f"question {i}", t, f"{tag}{i}")
+                i += 1
+
+        # warmup: captures the hipGraph buckets before the timed window
+        enqueue(2 * max_batch, "w")
+        while broker.depth() > 0:
+            worker.process_once()
+        enqueue(requests, "s")
         lat = []
         t0 = time.perf_counter()
         served = 0
