@@ -143,3 +143,31 @@ def test_feature_provider_end_to_end(tmp_path, tiny_detector):
     assert reg["spatials"].shape == (2, 11, 5)
     assert (reg["spatials"][:, :, :4] <= 1.001).all()
     assert reg["image_mask"][0, 0] == 1  # global region always valid
+
+
+def test_detector_checkpoint_roundtrip_and_foreign_names(tmp_path, tiny_detector):
+    import torch
+    from vilbert_multi_task_amd.detector.checkpoint import (
+        load_detectron_checkpoint,
+        save_checkpoint,
+    )
+    from vilbert_multi_task_amd.detector import DetectionModel, DetectorConfig
+
+    path = str(tmp_path / "model_final.pth")
+    save_checkpoint(tiny_detector, path)
+    torch.manual_seed(99)
+    m2 = DetectionModel(DetectorConfig.tiny())
+    rep = load_detectron_checkpoint(m2, path)
+    assert rep["missing"] == [] and rep["unexpected"] == []
+    for (k1, v1), (k2, v2) in zip(
+        tiny_detector.state_dict().items(), m2.state_dict().items()
+    ):
+        assert torch.equal(v1, v2), k1
+
+    # foreign naming (module. prefixes + renamed keys) -> shape matching
+    sd = {"module.some.foreign." + k: v for k, v in tiny_detector.state_dict().items()}
+    torch.save({"model": sd}, path)
+    torch.manual_seed(100)
+    m3 = DetectionModel(DetectorConfig.tiny())
+    rep = load_detectron_checkpoint(m3, path)
+    assert rep["missing"] == []  # every param found a same-shape source
