@@ -85,183 +85,180 @@ __global__ __launch_bounds__(256) void attn_kernel(
   const int NT = LK_PAD / 16;  // <= NTMAX score tiles per stripe
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  // Two-phase LDS reuse (PMC: 80 KB/WG fit only 1 WG/CU — mean occupancy
-  // 2 waves/CU): K and the transposed V SHARE one 32 KB tile. Phase A
-  // stages K and runs QK^T+softmax for every stripe (P -> per-stripe LDS,
-  // normalizer folded into P); a barrier retires K; phase B stages V over
-  // it and runs P·V. Total = tile + P = 64 KB (D=128) -> 2 WG/CU.
-  char* KV_lds = smem;                          // phase A: K [LK_PAD][D];
-                                                // phase B: VT [D][LK_PAD]
-  char* K_lds = KV_lds;
-  char* V_lds = KV_lds;
-  // per-(wave, local-stripe) P tiles: [16][LK_PAD] bf16, <=2 stripes/wave
-  char* P_base = smem + LK_PAD * D * 2;
+  // KGLOBAL: K fragments read straight from global (L2-resident slice) —
+  // no K_lds, LDS drops to V+P = 48 KB for D=128 -> 3 workgroups/CU.
+  char* K_lds = smem;                          // [LK_PAD][D] bf16, XOR-swizzled
+  char* V_lds = smem + (KGLOBAL ? 0 : LK_PAD * D * 2);  // VT [D][LK_PAD]
+  char* P_lds = V_lds + LK_PAD * D * 2 + wid * 16 * LK_PAD * 2;  // per-wave
 
-  // ---- helpers to stage one tensor into the shared tile ------------------
-  const int rows_per_pass = blockDim.x / KCH;  // 16 (D=128) or 32 (D=64)
-  const int r0 = tid / KCH;
-  const int cch = tid % KCH;
-  const long kbase0 = (long)b * Lk * ks + (long)h * D;
-  const long vbase0 = (long)b * Lk * vs + (long)h * D;
-
-  // ---- phase A: stage K (swizzled row-major) -----------------------------
-  if (!KGLOBAL) {
+  // ---- stage K (swizzled row-major) and V (tr-readable subtiles) ---------
+  {
+    const int rows_per_pass = blockDim.x / KCH;  // 16 (D=128) or 32 (D=64)
+    const int r0 = tid / KCH;
+    const int c = tid % KCH;
+    const long kbase0 = (long)b * Lk * ks + (long)h * D;
+    const long vbase0 = (long)b * Lk * vs + (long)h * D;
     for (int r = r0; r < LK_PAD; r += rows_per_pass) {
-      uint4 kraw = {0, 0, 0, 0};
-      if (r < Lk)
-        kraw = *reinterpret_cast<const uint4*>(kg + kbase0 + (long)r * ks + cch * 8);
-      lds_store_b128(K_lds + r * (D * 2) + ((cch * 16) ^ SWZ(r)), kraw);
+      uint4 kraw = {0, 0, 0, 0}, vraw = {0, 0, 0, 0};
+      if (r < Lk) {
+        if (!KGLOBAL)
+          kraw = *reinterpret_cast<const uint4*>(kg + kbase0 + (long)r * ks + c * 8);
+        vraw = *reinterpret_cast<const uint4*>(vg + vbase0 + (long)r * vs + c * 8);
+      }
+      if (!KGLOBAL)
+        lds_store_b128(K_lds + r * (D * 2) + ((c * 16) ^ SWZ(r)), kraw);
+      // V transposed image [D][LK_PAD] (XOR-swizzled rows): the PV MFMA
+      // B-fragment wants per-lane contiguous keys at fixed d, so transpose
+      // at staging. Writes are conflict-free: for fixed j, consecutive
+      // threads (consecutive r) write contiguous bytes of one VT row.
+      // (ds_read_b64_tr_b16 was measured unusable here: per 16-lane group
+      // it honors only lanes 0/4/8/12's addresses — 16 distinct values per
+      // read vs the 128 a 16x16x32 B-fragment needs; see tr16_probe.)
+      {
+        union { uint4 u; short s[8]; } vv;
+        vv.u = vraw;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          const int d = c * 8 + j;
+          *reinterpret_cast<short*>(
+              V_lds + d * (LK_PAD * 2) + ((r * 2) ^ SWZ(d))) = vv.s[j];
+        }
+      }
     }
   }
   __syncthreads();
 
+  // ---- per-wave stripes of 16 query rows ---------------------------------
   const int nstripes = (Lq + 15) / 16;
-  const int PSTRIDE = 16 * LK_PAD * 2;  // bytes per P tile
+  for (int s = split * (blockDim.x / WAVE) + wid; s < nstripes;
+       s += nsplit * (blockDim.x / WAVE)) {
+    const int qrow0 = s * 16;
+    // Q A-fragments straight from global
+    bf16x8 aq[D / 32];
+    {
+      const int row = min(qrow0 + (lane & 15), Lq - 1);
+      const long qoff = ((long)b * Lq + row) * qs + (long)h * D + (lane >> 4) * 8;
+#pragma unroll
+      for (int kk = 0; kk < D / 32; ++kk) aq[kk] = load_bf16x8(q + qoff + kk * 32);
+    }
 
-  // ---- phase A: S = QK^T, softmax, normalized P -> LDS -------------------
-  {
-    int ls = 0;
-    for (int s = wid; s < nstripes; s += blockDim.x / WAVE, ++ls) {
-      const int qrow0 = s * 16;
-      bf16x8 aq[D / 32];
-      {
-        const int row = min(qrow0 + (lane & 15), Lq - 1);
-        const long qoff = ((long)b * Lq + row) * qs + (long)h * D + (lane >> 4) * 8;
+    // ---- S = Q K^T --------------------------------------------------------
+    f32x4 acc_s[NTMAX];
 #pragma unroll
-        for (int kk = 0; kk < D / 32; ++kk) aq[kk] = load_bf16x8(q + qoff + kk * 32);
+    for (int nt = 0; nt < NTMAX; ++nt) acc_s[nt] = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+    for (int nt = 0; nt < NTMAX; ++nt) {
+      if (nt >= NT) break;
+      const int key = nt * 16 + (lane & 15);
+      if (KGLOBAL) {
+        const long kb = (long)b * Lk * ks + (long)h * D +
+                        (long)min(key, Lk - 1) * ks + (lane >> 4) * 8;
+#pragma unroll
+        for (int kk = 0; kk < D / 32; ++kk) {
+          const bf16x8 bk = load_bf16x8(kg + kb + kk * 32);
+          acc_s[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              (bf16x8)aq[kk], bk, acc_s[nt], 0, 0, 0);
+        }
+      } else {
+        const char* kbase = K_lds + key * (D * 2);
+        const int ksw = SWZ(key);
+#pragma unroll
+        for (int kk = 0; kk < D / 32; ++kk) {
+          const bf16x8 bk = lds_b128(kbase + (((kk * 64) + ((lane >> 4) * 16)) ^ ksw));
+          acc_s[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              (bf16x8)aq[kk], bk, acc_s[nt], 0, 0, 0);
+        }
       }
-      f32x4 acc_s[NTMAX];
+    }
+
+    // ---- mask + softmax (rows live across the 16-lane group) -------------
+    const int col0 = lane & 15;
+    float inv_l[4];
 #pragma unroll
-      for (int nt = 0; nt < NTMAX; ++nt) acc_s[nt] = {0.f, 0.f, 0.f, 0.f};
+    for (int r = 0; r < 4; ++r) {
+      const int row = qrow0 + (lane >> 4) * 4 + r;
+      float mx = -3.0e38f;
 #pragma unroll
       for (int nt = 0; nt < NTMAX; ++nt) {
         if (nt >= NT) break;
-        const int key = nt * 16 + (lane & 15);
-        if (KGLOBAL) {
-          const long kb = (long)b * Lk * ks + (long)h * D +
-                          (long)min(key, Lk - 1) * ks + (lane >> 4) * 8;
-#pragma unroll
-          for (int kk = 0; kk < D / 32; ++kk) {
-            const bf16x8 bk = load_bf16x8(kg + kb + kk * 32);
-            acc_s[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                (bf16x8)aq[kk], bk, acc_s[nt], 0, 0, 0);
-          }
+        const int col = nt * 16 + col0;
+        float sv = acc_s[nt][r] * scale;
+        if (col < Lk) {
+          if (mask_mode == 1)
+            sv += bf2f(mask[(long)b * Lk + col]);
+          else if (mask_mode == 2)
+            sv += bf2f(mask[((long)b * Lq + min(row, Lq - 1)) * Lk + col]);
+          acc_s[nt][r] = sv;
+          mx = fmaxf(mx, sv);
         } else {
-          const char* kbase = K_lds + key * (D * 2);
-          const int ksw = SWZ(key);
-#pragma unroll
-          for (int kk = 0; kk < D / 32; ++kk) {
-            const bf16x8 bk = lds_b128(kbase + (((kk * 64) + ((lane >> 4) * 16)) ^ ksw));
-            acc_s[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                (bf16x8)aq[kk], bk, acc_s[nt], 0, 0, 0);
-          }
+          acc_s[nt][r] = -3.0e38f;
         }
       }
+      const float mrow = group16_max(mx);
+      float sum = 0.f;
+#pragma unroll
+      for (int nt = 0; nt < NTMAX; ++nt) {
+        if (nt >= NT) break;
+        const int col = nt * 16 + col0;
+        const float p = (col < Lk) ? __expf(acc_s[nt][r] - mrow) : 0.f;
+        acc_s[nt][r] = p;
+        sum += p;
+      }
+      inv_l[r] = 1.0f / group16_sum(sum);
+    }
 
-      // mask + softmax; the 1/l normalizer is folded into P so phase B
-      // needs no per-row epilogue state
-      const int col0 = lane & 15;
-      char* P_tile = P_base + (wid * 2 + ls) * PSTRIDE;
+    // ---- P -> LDS (bf16, swizzled row-major; A-frag readable) -------------
 #pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        const int row = qrow0 + (lane >> 4) * 4 + r;
-        float mx = -3.0e38f;
+    for (int r = 0; r < 4; ++r) {
+      const int prow = (lane >> 4) * 4 + r;
+      char* prow_base = P_lds + prow * (LK_PAD * 2);
+      const int psw = SWZ(prow);
 #pragma unroll
-        for (int nt = 0; nt < NTMAX; ++nt) {
-          if (nt >= NT) break;
-          const int col = nt * 16 + col0;
-          float sv = acc_s[nt][r] * scale;
-          if (col < Lk) {
-            if (mask_mode == 1)
-              sv += bf2f(mask[(long)b * Lk + col]);
-            else if (mask_mode == 2)
-              sv += bf2f(mask[((long)b * Lq + min(row, Lq - 1)) * Lk + col]);
-            acc_s[nt][r] = sv;
-            mx = fmaxf(mx, sv);
-          } else {
-            acc_s[nt][r] = -3.0e38f;
-          }
-        }
-        const float mrow = group16_max(mx);
-        float sum = 0.f;
-#pragma unroll
-        for (int nt = 0; nt < NTMAX; ++nt) {
-          if (nt >= NT) break;
-          const int col = nt * 16 + col0;
-          const float pv = (col < Lk) ? __expf(acc_s[nt][r] - mrow) : 0.f;
-          acc_s[nt][r] = pv;
-          sum += pv;
-        }
-        const float inv_l = 1.0f / group16_sum(sum);
-        const int prow = (lane >> 4) * 4 + r;
-        char* prow_base = P_tile + prow * (LK_PAD * 2);
-        const int psw = SWZ(prow);
-#pragma unroll
-        for (int nt = 0; nt < NTMAX; ++nt) {
-          if (nt >= NT) break;
-          const int col = nt * 16 + col0;
-          *reinterpret_cast<short*>(prow_base + ((col * 2) ^ psw)) =
-              (short)f2us(acc_s[nt][r] * inv_l);
-        }
+      for (int nt = 0; nt < NTMAX; ++nt) {
+        if (nt >= NT) break;
+        const int col = nt * 16 + col0;
+        *reinterpret_cast<short*>(prow_base + ((col * 2) ^ psw)) =
+            (short)f2us(acc_s[nt][r]);
       }
     }
-  }
-  __syncthreads();  // K retired; P complete
+    // per-wave P buffer: same wave writes then reads (compiler inserts the
+    // lgkm waits through the address dependence); no cross-wave sharing.
 
-  // ---- phase B: stage V transposed over the shared tile ------------------
-  for (int r = r0; r < LK_PAD; r += rows_per_pass) {
-    uint4 vraw = {0, 0, 0, 0};
-    if (r < Lk)
-      vraw = *reinterpret_cast<const uint4*>(vg + vbase0 + (long)r * vs + cch * 8);
-    union { uint4 u; short sh[8]; } vv;
-    vv.u = vraw;
+    // ---- O = P V (V fragments via hardware transpose reads) ---------------
+    f32x4 acc_o[D / 16];
 #pragma unroll
-    for (int j = 0; j < 8; ++j) {
-      const int d = cch * 8 + j;
-      *reinterpret_cast<short*>(
-          V_lds + d * (LK_PAD * 2) + ((r * 2) ^ SWZ(d))) = vv.sh[j];
-    }
-  }
-  __syncthreads();
-
-  // ---- phase B: O = P V --------------------------------------------------
-  {
-    int ls = 0;
-    for (int s = wid; s < nstripes; s += blockDim.x / WAVE, ++ls) {
-      const int qrow0 = s * 16;
-      const char* P_tile = P_base + (wid * 2 + ls) * PSTRIDE;
-      f32x4 acc_o[D / 16];
+    for (int nt = 0; nt < D / 16; ++nt) acc_o[nt] = {0.f, 0.f, 0.f, 0.f};
+    const int parow = lane & 15;
+    const char* pa_base = P_lds + parow * (LK_PAD * 2);
+    const int pasw = SWZ(parow);
 #pragma unroll
-      for (int nt = 0; nt < D / 16; ++nt) acc_o[nt] = {0.f, 0.f, 0.f, 0.f};
-      const int parow = lane & 15;
-      const char* pa_base = P_tile + parow * (LK_PAD * 2);
-      const int pasw = SWZ(parow);
+    for (int kk = 0; kk < NTMAX / 2; ++kk) {  // LK_PAD/32 <= NTMAX/2
+      if (kk * 32 >= LK_PAD) break;
+      const bf16x8 ap =
+          lds_b128(pa_base + (((kk * 64) + ((lane >> 4) * 16)) ^ pasw));
+      const int keyoff = (kk * 64) + ((lane >> 4) * 16);  // byte offset of keys
 #pragma unroll
-      for (int kk = 0; kk < NTMAX / 2; ++kk) {  // LK_PAD/32 <= NTMAX/2
-        if (kk * 32 >= LK_PAD) break;
-        const bf16x8 ap =
-            lds_b128(pa_base + (((kk * 64) + ((lane >> 4) * 16)) ^ pasw));
-        const int keyoff = (kk * 64) + ((lane >> 4) * 16);  // byte offset of keys
-#pragma unroll
-        for (int nt = 0; nt < D / 16; ++nt) {
-          const int d = nt * 16 + (lane & 15);
-          const bf16x8 bv =
-              lds_b128(V_lds + d * (LK_PAD * 2) + (keyoff ^ SWZ(d)));
-          acc_o[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ap, bv, acc_o[nt], 0, 0, 0);
-        }
-      }
-#pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        const int row = qrow0 + (lane >> 4) * 4 + r;
-        if (row < Lq) {
-          const long obase = ((long)b * Lq + row) * HD + (long)h * D;
-          const int col0 = lane & 15;
-#pragma unroll
-          for (int nt = 0; nt < D / 16; ++nt)
-            out[obase + nt * 16 + col0] = f2bf(acc_o[nt][r]);
-        }
+      for (int nt = 0; nt < D / 16; ++nt) {
+        const int d = nt * 16 + (lane & 15);
+        const bf16x8 bv =
+            lds_b128(V_lds + d * (LK_PAD * 2) + (keyoff ^ SWZ(d)));
+        acc_o[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ap, bv, acc_o[nt], 0, 0, 0);
       }
     }
+
+    // ---- normalize + store O ---------------------------------------------
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int row = qrow0 + (lane >> 4) * 4 + r;
+      if (row < Lq) {
+        const long obase = ((long)b * Lq + row) * HD + (long)h * D;
+#pragma unroll
+        for (int nt = 0; nt < D / 16; ++nt)
+          out[obase + nt * 16 + col0] = f2bf(acc_o[nt][r] * inv_l[r]);
+      }
+    }
+    // NO __syncthreads here: P_lds is per-wave and stripe counts differ
+    // across waves (a block-wide barrier inside this loop would deadlock).
   }
 }
 
@@ -322,10 +319,9 @@ void launch_attention(const bf16* q, const bf16* k, const bf16* v,
   }();
   // measured on MI355X (B=256): nsplit>1 duplicates staging and loses
   int nsplit = nsplit_env > 0 ? nsplit_env : 1;
-  // shared K/V tile + per-(wave,stripe) P tiles (bf16); Lq<=128 -> <=2
-  // stripes per wave
+  // (K if staged) + V + 4x per-wave P (all bf16)
   const size_t lds = sizeof(bf16) *
-      (size_t)(LK_PAD * D + 4 * 2 * 16 * LK_PAD);
+      (size_t)((kglobal_env ? 1 : 2) * LK_PAD * D + 4 * 16 * LK_PAD);
   const dim3 grid(B * H * nsplit);
 #define LAUNCH_ATTN(DD, KG, NTM)                                              \
   hipLaunchKernelGGL((attn_kernel<DD, KG, NTM>), grid, dim3(256), lds, stream, \
