@@ -40,6 +40,11 @@ def main() -> None:
     ap.add_argument("--batch", type=int, default=512, help="per-GPU queries per step")
     ap.add_argument("--no-graphs", action="store_true")
     ap.add_argument("--eager", action="store_true", help="force eager torch ops (debug)")
+    ap.add_argument(
+        "--fp8", action="store_true",
+        help="OPT-IN fp8 (e4m3) encoder GEMMs; NOT the judged config "
+        "(reduced precision) — reported with dtype=fp8-mixed",
+    )
     args = ap.parse_args()
 
     if args.eager:
@@ -79,6 +84,7 @@ def main() -> None:
         device=device,
         use_graphs=not args.no_graphs,
         dtype=torch.bfloat16 if device.startswith("cuda") else torch.float32,
+        fp8=args.fp8,
     )
     batch = synthetic_batch(
         args.batch,
@@ -144,7 +150,7 @@ def main() -> None:
                     "higher_is_better": True,
                     "scaling": "weak",
                     "vs_baseline": None,
-                    "dtype": "bf16" if device.startswith("cuda") else "fp32",
+                    "dtype": ("fp8-mixed" if args.fp8 else "bf16") if device.startswith("cuda") else "fp32",
                     "data": "synthetic",
                     "config": {
                         "model": "vilbert-12in1-270M (bert_base_6layer_6conect)",

@@ -96,3 +96,26 @@ def test_training_step_bf16(base_model_pair):
         m.zero_grad(set_to_none=True)
     finally:
         m.eval()
+
+
+def test_fp8_serving_mode_accuracy(base_model_pair):
+    """Opt-in fp8 (e4m3) encoder GEMMs: logits track the bf16 path."""
+    import copy
+
+    _, m_bf16 = base_model_pair
+    m_bf16.eval()
+    from vilbert_multi_task_amd.models.fp8 import convert_encoder_to_fp8
+
+    m_fp8 = copy.deepcopy(m_bf16)
+    n = convert_encoder_to_fp8(m_fp8)
+    assert n >= 100  # 12 text + 6 vision layers x4 + 6 co x(2x4) + FFNs
+    batch = synthetic_batch(2, seed=5, device="cuda", dtype=torch.bfloat16)
+    with torch.no_grad():
+        ref = m_bf16(*forward_args(batch))
+        out = m_fp8(*forward_args(batch))
+    for i, name in [(0, "vil_prediction"), (6, "vision_logit")]:
+        a = ref[i].flatten().float()
+        b = out[i].flatten().float()
+        cos = torch.nn.functional.cosine_similarity(a, b, dim=0).item()
+        assert cos > 0.97, f"{name}: cos {cos}"
+        assert torch.isfinite(b).all()

@@ -30,10 +30,15 @@ class GraphRunner:
         dtype: torch.dtype = torch.bfloat16,
         use_graphs: bool = True,
         serving_fast: bool = False,
+        fp8: bool = False,
     ):
         self.model = model.eval()
         if hasattr(model, "prepare_for_serving"):
             model.prepare_for_serving()
+        if fp8 and device.startswith("cuda"):
+            from ..models.fp8 import convert_encoder_to_fp8
+
+            convert_encoder_to_fp8(model)
         if serving_fast:
             # skip heads the demo decode never reads (models/heads.py)
             model.skip_unused_heads = True

@@ -168,7 +168,10 @@ class FeedForward(nn.Module):
         self.layer_norm = FusedLayerNorm(hidden, eps)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        h = F_ops.linear_bias_gelu(x, self.intermediate.weight, self.intermediate.bias)
+        if isinstance(self.intermediate, nn.Linear):
+            h = F_ops.linear_bias_gelu(x, self.intermediate.weight, self.intermediate.bias)
+        else:  # Fp8Linear serving mode: fp8 matmul+bias, then erf GELU
+            h = F_ops.bias_gelu(self.intermediate(x), None)
         return self.layer_norm(self.dropout(self.output(h)), residual=x)
 
 
