@@ -79,15 +79,22 @@ def convert_encoder_to_fp8(model) -> int:
     n = 0
     bert = model.bert if hasattr(model, "bert") else model
     for m in bert.modules():
-        if isinstance(m, (MultiHeadSelfAttention, CrossAttention)):
-            for name in ("query", "key", "value", "out"):
-                setattr(m, name, Fp8Linear(getattr(m, name)))
-                n += 1
-            m._wqkv = None  # disable the fused bf16 path
-            m._bqkv = None
-            if hasattr(m, "_wkv"):
-                m._wkv = None
-                m._bkv = None
+        if isinstance(m, MultiHeadSelfAttention):
+            # fused QKV in fp8: ONE activation quantization + ONE _scaled_mm
+            w = torch.cat([m.query.weight, m.key.weight, m.value.weight], dim=0)
+            m._wqkv8, m._wqkv_scale = quantize_weight(w.detach())
+            m._bqkv = torch.cat([m.query.bias, m.key.bias, m.value.bias]).detach()
+            m._wqkv = None  # bf16 fused path off
+            m.out = Fp8Linear(m.out)
+            n += 4
+        elif isinstance(m, CrossAttention):
+            w = torch.cat([m.key.weight, m.value.weight], dim=0)
+            m._wkv8, m._wkv_scale = quantize_weight(w.detach())
+            m._bkv = torch.cat([m.key.bias, m.value.bias]).detach()
+            m._wkv = None
+            m.query = Fp8Linear(m.query)
+            m.out = Fp8Linear(m.out)
+            n += 4
         elif isinstance(m, FeedForward):
             m.intermediate = Fp8Linear(m.intermediate)
             m.output = Fp8Linear(m.output)

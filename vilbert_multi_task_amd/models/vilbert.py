@@ -129,6 +129,8 @@ class MultiHeadSelfAttention(nn.Module):
         self.layer_norm = FusedLayerNorm(hidden, eps)
         self._wqkv = None  # serving-time fused projection (prepare_serving)
         self._bqkv = None
+        self._wqkv8 = None  # fp8 serving mode (models/fp8.py)
+        self._wqkv_scale = None
 
     def prepare_serving(self) -> None:
         """Fuse Q/K/V projections into one GEMM for the inference path
@@ -145,7 +147,14 @@ class MultiHeadSelfAttention(nn.Module):
     def forward(
         self, x: torch.Tensor, mask_bias: Optional[torch.Tensor], need_probs: bool
     ) -> Tuple[torch.Tensor, Optional[torch.Tensor]]:
-        if self._wqkv is not None and not self.training and not torch.is_grad_enabled():
+        infer = not self.training and not torch.is_grad_enabled()
+        if self._wqkv8 is not None and infer:
+            from .fp8 import fp8_linear
+
+            h = self.heads * self.head_dim
+            qkv = fp8_linear(x, self._wqkv8, self._wqkv_scale, self._bqkv)
+            q, k, v = qkv[..., :h], qkv[..., h : 2 * h], qkv[..., 2 * h :]
+        elif self._wqkv is not None and infer:
             h = self.heads * self.head_dim
             qkv = torch.nn.functional.linear(x, self._wqkv, self._bqkv)
             q, k, v = qkv[..., :h], qkv[..., h : 2 * h], qkv[..., 2 * h :]
@@ -206,6 +215,8 @@ class CrossAttention(nn.Module):
         self.layer_norm = FusedLayerNorm(q_hidden, eps)
         self._wkv = None  # serving-time fused K/V projection
         self._bkv = None
+        self._wkv8 = None  # fp8 serving mode
+        self._wkv_scale = None
 
     def prepare_serving(self) -> None:
         with torch.no_grad():
@@ -213,7 +224,14 @@ class CrossAttention(nn.Module):
             self._bkv = torch.cat([self.key.bias, self.value.bias], dim=0).contiguous()
 
     def forward(self, x_q, x_kv, mask_bias, need_probs=False):
-        if self._wkv is not None and not self.training and not torch.is_grad_enabled():
+        infer = not self.training and not torch.is_grad_enabled()
+        if self._wkv8 is not None and infer:
+            from .fp8 import fp8_linear
+
+            h = self.heads * self.head_dim
+            kv = fp8_linear(x_kv, self._wkv8, self._wkv_scale, self._bkv)
+            k, v = kv[..., :h], kv[..., h:]
+        elif self._wkv is not None and infer:
             h = self.heads * self.head_dim
             kv = torch.nn.functional.linear(x_kv, self._wkv, self._bkv)
             k, v = kv[..., :h], kv[..., h:]
