@@ -95,11 +95,17 @@ def main() -> None:
     ap.add_argument("--gqa-answers", default="save/gqa/cache/trainval_label2ans.pkl")
     ap.add_argument("--detector", action="store_true", help="full Faster R-CNN features")
     ap.add_argument("--max-batch", type=int, default=64)
+    ap.add_argument("--metrics-port", type=int, default=0,
+                    help="Prometheus /metrics port for the worker (0 = off)")
     args = ap.parse_args()
 
     if args.role == "app":
         run_app(args)
     elif args.role == "worker":
+        if args.metrics_port:
+            from ..utils.trace import start_metrics_server
+
+            start_metrics_server(args.metrics_port)
         build_worker(args).run_forever()
     else:
         t = threading.Thread(target=lambda: build_worker(args).run_forever(), daemon=True)
