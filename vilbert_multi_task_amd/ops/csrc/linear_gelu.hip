@@ -17,6 +17,7 @@
 #include <hipblaslt/hipblaslt.h>
 
 #include <map>
+#include <vector>
 #include <mutex>
 #include <stdexcept>
 #include <tuple>
@@ -36,6 +37,7 @@ struct Plan {
   hipblasLtMatrixLayout_t a{}, b{}, c{};
   hipblasLtMatmulAlgo_t algo{};
   bool has_algo = false;
+  std::vector<hipblasLtMatmulHeuristicResult_t> candidates;
 };
 
 hipblasLtHandle_t handle_once() {
@@ -83,17 +85,55 @@ Plan& get_plan(long M, long N, long K, void* workspace, size_t ws_bytes) {
   HIPBLASLT_CHECK(hipblasLtMatmulPreferenceCreate(&pref));
   HIPBLASLT_CHECK(hipblasLtMatmulPreferenceSetAttribute(
       pref, HIPBLASLT_MATMUL_PREF_MAX_WORKSPACE_BYTES, &ws_bytes, sizeof(ws_bytes)));
-  hipblasLtMatmulHeuristicResult_t results[4];
+  hipblasLtMatmulHeuristicResult_t results[8];
   int found = 0;
   HIPBLASLT_CHECK(hipblasLtMatmulAlgoGetHeuristic(
-      handle_once(), p.op, p.a, p.b, p.c, p.c, pref, 4, results, &found));
+      handle_once(), p.op, p.a, p.b, p.c, p.c, pref, 8, results, &found));
   hipblasLtMatmulPreferenceDestroy(pref);
   if (found > 0) {
     p.algo = results[0].algo;
     p.has_algo = true;
+    p.candidates.assign(results, results + found);
   }
   auto r = cache.emplace(key, p);
   return r.first->second;
+}
+
+// Time each heuristic candidate once (3 reps) and keep the fastest. Runs at
+// plan creation — i.e. during eager warmup BEFORE hipGraph capture, where
+// stream synchronization is legal. Scratch in/out buffers come from the
+// caller's tensors (the timing matmuls write the real y, which the caller
+// recomputes right after with the winning algo).
+static void autotune(Plan& p, const void* x, const void* w, const void* bias,
+                     void* y, void* workspace, size_t ws_bytes,
+                     hipStream_t stream) {
+  if (p.candidates.size() <= 1) return;
+  float alpha = 1.0f, beta = 0.0f;
+  hipEvent_t ev0, ev1;
+  if (hipEventCreate(&ev0) != hipSuccess) return;
+  if (hipEventCreate(&ev1) != hipSuccess) { hipEventDestroy(ev0); return; }
+  float best = 1e30f;
+  hipblasLtMatmulAlgo_t best_algo = p.algo;
+  for (auto& cand : p.candidates) {
+    // correctness probe first
+    if (hipblasLtMatmul(handle_once(), p.op, &alpha, w, p.a, x, p.b, &beta, y,
+                        p.c, y, p.c, &cand.algo, workspace, ws_bytes,
+                        stream) != HIPBLAS_STATUS_SUCCESS)
+      continue;
+    hipEventRecord(ev0, stream);
+    for (int r = 0; r < 3; ++r)
+      hipblasLtMatmul(handle_once(), p.op, &alpha, w, p.a, x, p.b, &beta, y,
+                      p.c, y, p.c, &cand.algo, workspace, ws_bytes, stream);
+    hipEventRecord(ev1, stream);
+    hipEventSynchronize(ev1);
+    float ms = 1e30f;
+    hipEventElapsedTime(&ms, ev0, ev1);
+    if (ms < best) { best = ms; best_algo = cand.algo; }
+  }
+  p.algo = best_algo;
+  p.candidates.clear();  // tuned once
+  hipEventDestroy(ev0);
+  hipEventDestroy(ev1);
 }
 
 }  // namespace
@@ -108,6 +148,12 @@ int hipblaslt_linear_gelu(const void* x, const void* w, const void* bias,
     if (!p.has_algo) return 1;
     HIPBLASLT_CHECK(hipblasLtMatmulDescSetAttribute(
         p.op, HIPBLASLT_MATMUL_DESC_BIAS_POINTER, &bias, sizeof(bias)));
+    int capturing = 0;
+    hipStreamCaptureStatus st = hipStreamCaptureStatusNone;
+    if (hipStreamIsCapturing(stream, &st) == hipSuccess &&
+        st != hipStreamCaptureStatusNone)
+      capturing = 1;
+    if (!capturing) autotune(p, x, w, bias, y, workspace, ws_bytes, stream);
     float alpha = 1.0f, beta = 0.0f;
     HIPBLASLT_CHECK(hipblasLtMatmul(
         handle_once(), p.op, &alpha, w, p.a, x, p.b, &beta, y, p.c, y, p.c,
