@@ -244,3 +244,15 @@ def test_linear_bias_gelu_fused(ext):
     ref = torch.nn.functional.gelu(x.float() @ w.float().T + b.float())
     err = (y.float() - ref).abs() - ref.abs() / 64
     assert err.max() < 3e-2, err.max().item()
+
+
+def test_linear_bias_residual_fused(ext):
+    """GEMM + bias + residual in one hipBLASLt call (beta=1 epilogue)."""
+    x = _rand_bf16(37 * 8, 3072, seed=80)
+    w = _rand_bf16(768, 3072, seed=81, scale=0.02)
+    b = _rand_bf16(768, seed=82, scale=0.1)
+    res = _rand_bf16(37 * 8, 768, seed=83)
+    y = torch.ops.vilbert_amd.linear_bias_residual(x, w, b, res)
+    ref = x.float() @ w.float().T + b.float() + res.float()
+    err = (y.float() - ref).abs() - ref.abs() / 64
+    assert err.max() < 3e-2, err.max().item()

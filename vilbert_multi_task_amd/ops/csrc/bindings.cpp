@@ -32,6 +32,9 @@ void launch_nms_multiclass(const float*, const float*, const long*, float*, int,
 void launch_tr16_probe(short*, int, hipStream_t);
 int hipblaslt_linear_gelu(const void*, const void*, const void*, void*, long,
                           long, long, void*, size_t, hipStream_t);
+int hipblaslt_linear_bias_add(const void*, const void*, const void*,
+                              const void*, void*, long, long, long, void*,
+                              size_t, hipStream_t);
 template <typename T>
 void launch_roi_align(const T*, const float*, T*, int, int, int, int, int, int,
                       int, float, int, hipStream_t);
@@ -215,6 +218,29 @@ at::Tensor linear_bias_gelu(const at::Tensor& x, const at::Tensor& w,
   return y;
 }
 
+at::Tensor linear_bias_residual(const at::Tensor& x, const at::Tensor& w,
+                                const at::Tensor& bias,
+                                const at::Tensor& residual) {
+  TORCH_CHECK(x.is_cuda() && x.scalar_type() == at::kBFloat16, "linear_bias_residual: bf16");
+  TORCH_CHECK(w.dim() == 2 && bias.dim() == 1 && w.size(0) == bias.size(0));
+  auto xc = x.contiguous();
+  auto rc = residual.contiguous();
+  const long K = x.size(-1), N = w.size(0);
+  const long M = x.numel() / K;
+  TORCH_CHECK(w.size(1) == K && rc.numel() == M * N);
+  auto sizes = x.sizes().vec();
+  sizes.back() = N;
+  auto y = at::empty(sizes, x.options());
+  constexpr size_t kWs = 32L * 1024 * 1024;
+  hipStream_t stream = cur_stream();
+  void* ws = ws_for_stream(stream, x.options(), kWs);
+  int rc2 = hipblaslt_linear_bias_add(
+      xc.data_ptr(), w.contiguous().data_ptr(), bias.contiguous().data_ptr(),
+      rc.data_ptr(), y.data_ptr(), M, N, K, ws, kWs, stream);
+  TORCH_CHECK(rc2 == 0, "hipblaslt_linear_bias_add failed (no algo)");
+  return y;
+}
+
 at::Tensor roi_align(const at::Tensor& input, const at::Tensor& rois,
                      int64_t ph, int64_t pw, double spatial_scale,
                      int64_t sampling_ratio) {
@@ -271,6 +297,7 @@ TORCH_LIBRARY(vilbert_amd, m) {
   m.def("nms_multiclass(Tensor boxes, Tensor scores, float iou_thr, float score_thr) -> Tensor");
   m.def("roi_align(Tensor input, Tensor rois, int ph, int pw, float spatial_scale, int sampling_ratio) -> Tensor");
   m.def("linear_bias_gelu(Tensor x, Tensor w, Tensor bias) -> Tensor");
+  m.def("linear_bias_residual(Tensor x, Tensor w, Tensor bias, Tensor residual) -> Tensor");
 }
 
 TORCH_LIBRARY_IMPL(vilbert_amd, CUDA, m) {
@@ -283,4 +310,5 @@ TORCH_LIBRARY_IMPL(vilbert_amd, CUDA, m) {
   m.impl("nms_multiclass", nms_multiclass);
   m.impl("roi_align", roi_align);
   m.impl("linear_bias_gelu", linear_bias_gelu);
+  m.impl("linear_bias_residual", linear_bias_residual);
 }

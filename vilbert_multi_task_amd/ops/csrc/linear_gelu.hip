@@ -37,6 +37,7 @@ struct Plan {
   hipblasLtMatrixLayout_t a{}, b{}, c{};
   hipblasLtMatmulAlgo_t algo{};
   bool has_algo = false;
+  float beta = 0.0f;
   std::vector<hipblasLtMatmulHeuristicResult_t> candidates;
 };
 
@@ -49,8 +50,8 @@ hipblasLtHandle_t handle_once() {
   return h;
 }
 
-std::map<std::tuple<long, long, long>, Plan>& plan_cache() {
-  static std::map<std::tuple<long, long, long>, Plan> m;
+std::map<std::tuple<long, long, long, int>, Plan>& plan_cache() {
+  static std::map<std::tuple<long, long, long, int>, Plan> m;
   return m;
 }
 std::mutex& plan_mu() {
@@ -58,8 +59,10 @@ std::mutex& plan_mu() {
   return m;
 }
 
-Plan& get_plan(long M, long N, long K, void* workspace, size_t ws_bytes) {
-  auto key = std::make_tuple(M, N, K);
+// kind 0: epilogue GELU_BIAS, beta=0.  kind 1: epilogue BIAS, beta=1 (the
+// C operand carries the residual: D = x@W^T + bias + residual).
+Plan& get_plan(long M, long N, long K, int kind, void* workspace, size_t ws_bytes) {
+  auto key = std::make_tuple(M, N, K, kind);
   auto& cache = plan_cache();
   auto it = cache.find(key);
   if (it != cache.end()) return it->second;
@@ -71,7 +74,8 @@ Plan& get_plan(long M, long N, long K, void* workspace, size_t ws_bytes) {
       p.op, HIPBLASLT_MATMUL_DESC_TRANSA, &ta, sizeof(ta)));
   HIPBLASLT_CHECK(hipblasLtMatmulDescSetAttribute(
       p.op, HIPBLASLT_MATMUL_DESC_TRANSB, &tb, sizeof(tb)));
-  hipblasLtEpilogue_t epi = HIPBLASLT_EPILOGUE_GELU_BIAS;
+  hipblasLtEpilogue_t epi =
+      kind == 0 ? HIPBLASLT_EPILOGUE_GELU_BIAS : HIPBLASLT_EPILOGUE_BIAS;
   HIPBLASLT_CHECK(hipblasLtMatmulDescSetAttribute(
       p.op, HIPBLASLT_MATMUL_DESC_EPILOGUE, &epi, sizeof(epi)));
   // A = w [K,N] col-major view of row-major [N,K], opA = T -> [N,K]
@@ -93,6 +97,7 @@ Plan& get_plan(long M, long N, long K, void* workspace, size_t ws_bytes) {
   if (found > 0) {
     p.algo = results[0].algo;
     p.has_algo = true;
+    p.beta = kind == 0 ? 0.0f : 1.0f;
     p.candidates.assign(results, results + found);
   }
   auto r = cache.emplace(key, p);
@@ -105,10 +110,10 @@ Plan& get_plan(long M, long N, long K, void* workspace, size_t ws_bytes) {
 // caller's tensors (the timing matmuls write the real y, which the caller
 // recomputes right after with the winning algo).
 static void autotune(Plan& p, const void* x, const void* w, const void* bias,
-                     void* y, void* workspace, size_t ws_bytes,
-                     hipStream_t stream) {
+                     const void* cmat, void* y, void* workspace,
+                     size_t ws_bytes, hipStream_t stream) {
   if (p.candidates.size() <= 1) return;
-  float alpha = 1.0f, beta = 0.0f;
+  float alpha = 1.0f, beta = p.beta;
   hipEvent_t ev0, ev1;
   if (hipEventCreate(&ev0) != hipSuccess) return;
   if (hipEventCreate(&ev1) != hipSuccess) { hipEventDestroy(ev0); return; }
@@ -116,14 +121,14 @@ static void autotune(Plan& p, const void* x, const void* w, const void* bias,
   hipblasLtMatmulAlgo_t best_algo = p.algo;
   for (auto& cand : p.candidates) {
     // correctness probe first
-    if (hipblasLtMatmul(handle_once(), p.op, &alpha, w, p.a, x, p.b, &beta, y,
-                        p.c, y, p.c, &cand.algo, workspace, ws_bytes,
+    if (hipblasLtMatmul(handle_once(), p.op, &alpha, w, p.a, x, p.b, &beta,
+                        cmat, p.c, y, p.c, &cand.algo, workspace, ws_bytes,
                         stream) != HIPBLAS_STATUS_SUCCESS)
       continue;
     hipEventRecord(ev0, stream);
     for (int r = 0; r < 3; ++r)
-      hipblasLtMatmul(handle_once(), p.op, &alpha, w, p.a, x, p.b, &beta, y,
-                      p.c, y, p.c, &cand.algo, workspace, ws_bytes, stream);
+      hipblasLtMatmul(handle_once(), p.op, &alpha, w, p.a, x, p.b, &beta,
+                      cmat, p.c, y, p.c, &cand.algo, workspace, ws_bytes, stream);
     hipEventRecord(ev1, stream);
     hipEventSynchronize(ev1);
     float ms = 1e30f;
@@ -138,28 +143,45 @@ static void autotune(Plan& p, const void* x, const void* w, const void* bias,
 
 }  // namespace
 
-// Returns 0 on success, nonzero on failure (caller falls back to unfused).
-int hipblaslt_linear_gelu(const void* x, const void* w, const void* bias,
-                          void* y, long M, long N, long K, void* workspace,
-                          size_t ws_bytes, hipStream_t stream) {
+// Shared driver. kind 0: gelu(x@W^T+b); kind 1: x@W^T+b+residual.
+static int run_epilogue_gemm(int kind, const void* x, const void* w,
+                             const void* bias, const void* residual, void* y,
+                             long M, long N, long K, void* workspace,
+                             size_t ws_bytes, hipStream_t stream) {
   try {
     std::lock_guard<std::mutex> lock(plan_mu());
-    Plan& p = get_plan(M, N, K, workspace, ws_bytes);
+    Plan& p = get_plan(M, N, K, kind, workspace, ws_bytes);
     if (!p.has_algo) return 1;
     HIPBLASLT_CHECK(hipblasLtMatmulDescSetAttribute(
         p.op, HIPBLASLT_MATMUL_DESC_BIAS_POINTER, &bias, sizeof(bias)));
+    const void* cmat = kind == 0 ? y : residual;
     int capturing = 0;
     hipStreamCaptureStatus st = hipStreamCaptureStatusNone;
     if (hipStreamIsCapturing(stream, &st) == hipSuccess &&
         st != hipStreamCaptureStatusNone)
       capturing = 1;
-    if (!capturing) autotune(p, x, w, bias, y, workspace, ws_bytes, stream);
-    float alpha = 1.0f, beta = 0.0f;
+    if (!capturing) autotune(p, x, w, bias, cmat, y, workspace, ws_bytes, stream);
+    float alpha = 1.0f, beta = p.beta;
     HIPBLASLT_CHECK(hipblasLtMatmul(
-        handle_once(), p.op, &alpha, w, p.a, x, p.b, &beta, y, p.c, y, p.c,
+        handle_once(), p.op, &alpha, w, p.a, x, p.b, &beta, cmat, p.c, y, p.c,
         &p.algo, workspace, ws_bytes, stream));
     return 0;
   } catch (const std::exception&) {
     return 1;
   }
+}
+
+int hipblaslt_linear_gelu(const void* x, const void* w, const void* bias,
+                          void* y, long M, long N, long K, void* workspace,
+                          size_t ws_bytes, hipStream_t stream) {
+  return run_epilogue_gemm(0, x, w, bias, nullptr, y, M, N, K, workspace,
+                           ws_bytes, stream);
+}
+
+int hipblaslt_linear_bias_add(const void* x, const void* w, const void* bias,
+                              const void* residual, void* y, long M, long N,
+                              long K, void* workspace, size_t ws_bytes,
+                              hipStream_t stream) {
+  return run_epilogue_gemm(1, x, w, bias, residual, y, M, N, K, workspace,
+                           ws_bytes, stream);
 }

@@ -119,6 +119,24 @@ def linear_bias_gelu(x: torch.Tensor, w: torch.Tensor, bias: torch.Tensor) -> to
     return bias_gelu(h, bias)
 
 
+_LINEAR_RES_OK = True
+
+
+def linear_bias_residual(
+    x: torch.Tensor, w: torch.Tensor, bias: torch.Tensor, residual: torch.Tensor
+) -> torch.Tensor:
+    """y = x @ w.T + bias + residual — the residual add fused into the GEMM
+    epilogue (beta=1) so the following LayerNorm reads ONE tensor."""
+    global _LINEAR_RES_OK
+    if _LINEAR_RES_OK and _want_hip(x, w, bias, residual):
+        ext = _load_extension()
+        try:
+            return ext.linear_bias_residual(x, w, bias, residual)
+        except RuntimeError:
+            _LINEAR_RES_OK = False
+    return torch.nn.functional.linear(x, w, bias) + residual
+
+
 # --------------------------------------------------------------------------
 # Multi-head scaled-dot-product attention with additive mask.
 # Flattened [B, L, H*D] layout so the GPU path needs no transpose copies.
