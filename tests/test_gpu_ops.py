@@ -247,7 +247,9 @@ def test_linear_bias_gelu_fused(ext):
 
 
 def test_linear_bias_residual_fused(ext):
-    """GEMM + bias + residual in one hipBLASLt call (beta=1 epilogue)."""
+    """GEMM + bias + residual in one hipBLASLt call (beta=1 epilogue).
+    NOT wired into the model: intermittent hipBLASLt write-faults at some
+    shapes (see models/vilbert.py note); kept as the round-2 start point."""
     x = _rand_bf16(37 * 8, 3072, seed=80)
     w = _rand_bf16(768, 3072, seed=81, scale=0.02)
     b = _rand_bf16(768, seed=82, scale=0.1)
