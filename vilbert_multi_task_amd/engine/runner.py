@@ -112,6 +112,19 @@ class GraphRunner:
             out = self._forward(inp)
         self._graphs[bucket] = (g, inp, out)
 
+    def warmup_buckets(self, max_rows: int) -> None:
+        """Capture every power-of-two bucket up to max_rows upfront so no
+        serving request pays first-use capture latency (~1-3 s)."""
+        if not self.use_graphs:
+            return
+        b = 1
+        while True:
+            self.capture(b)
+            if b >= max_rows:
+                break
+            b <<= 1
+        torch.cuda.synchronize()
+
     # -- execution ---------------------------------------------------------
     @torch.no_grad()
     def run(self, batch: dict):

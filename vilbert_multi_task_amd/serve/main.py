@@ -106,7 +106,10 @@ def main() -> None:
             from ..utils.trace import start_metrics_server
 
             start_metrics_server(args.metrics_port)
-        build_worker(args).run_forever()
+        w = build_worker(args)
+        if hasattr(w.runner, "warmup_buckets"):
+            w.runner.warmup_buckets(args.max_batch)
+        w.run_forever()
     else:
         t = threading.Thread(target=lambda: build_worker(args).run_forever(), daemon=True)
         t.start()
