@@ -156,7 +156,7 @@ class MultiHeadSelfAttention(nn.Module):
             q, k, v = qkv[..., :h], qkv[..., h : 2 * h], qkv[..., 2 * h :]
         elif self._wqkv is not None and infer:
             h = self.heads * self.head_dim
-            qkv = torch.nn.functional.linear(x, self._wqkv, self._bqkv)
+            qkv = F_ops.linear_bias(x, self._wqkv, self._bqkv)
             q, k, v = qkv[..., :h], qkv[..., h : 2 * h], qkv[..., 2 * h :]
         else:
             q, k, v = self.query(x), self.key(x), self.value(x)
@@ -168,7 +168,11 @@ class MultiHeadSelfAttention(nn.Module):
         # (linear_bias_residual, beta=1) intermittently faults in hipBLASLt
         # at some shapes ("write access to a read-only page", B=512 warmup)
         # — wiring reverted; the op remains for round-2 investigation.
-        y = self.layer_norm(self.dropout(self.out(ctx)), residual=x)
+        if infer and isinstance(self.out, nn.Linear):
+            o = F_ops.linear_bias(ctx, self.out.weight, self.out.bias)
+        else:
+            o = self.out(ctx)
+        y = self.layer_norm(self.dropout(o), residual=x)
         return y, probs
 
 
@@ -185,6 +189,13 @@ class FeedForward(nn.Module):
             h = F_ops.linear_bias_gelu(x, self.intermediate.weight, self.intermediate.bias)
         else:  # Fp8Linear serving mode: fp8 matmul+bias, then erf GELU
             h = F_ops.bias_gelu(self.intermediate(x), None)
+        if (
+            not self.training
+            and not torch.is_grad_enabled()
+            and isinstance(self.output, nn.Linear)
+        ):
+            o = F_ops.linear_bias(h, self.output.weight, self.output.bias)
+            return self.layer_norm(self.dropout(o), residual=x)
         return self.layer_norm(self.dropout(self.output(h)), residual=x)
 
 
@@ -237,7 +248,7 @@ class CrossAttention(nn.Module):
             k, v = kv[..., :h], kv[..., h:]
         elif self._wkv is not None and infer:
             h = self.heads * self.head_dim
-            kv = torch.nn.functional.linear(x_kv, self._wkv, self._bkv)
+            kv = F_ops.linear_bias(x_kv, self._wkv, self._bkv)
             k, v = kv[..., :h], kv[..., h:]
         else:
             k, v = self.key(x_kv), self.value(x_kv)
@@ -245,7 +256,11 @@ class CrossAttention(nn.Module):
             self.query(x_q), k, v, self.heads,
             mask_bias, self.attn_dropout_p, self.training, need_probs,
         )
-        y = self.layer_norm(self.dropout(self.out(ctx)), residual=x_q)
+        if infer and isinstance(self.out, nn.Linear):
+            o = F_ops.linear_bias(ctx, self.out.weight, self.out.bias)
+        else:
+            o = self.out(ctx)
+        y = self.layer_norm(self.dropout(o), residual=x_q)
         return y, probs
 
 

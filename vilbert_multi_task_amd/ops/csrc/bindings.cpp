@@ -35,6 +35,8 @@ int hipblaslt_linear_gelu(const void*, const void*, const void*, void*, long,
 int hipblaslt_linear_bias_add(const void*, const void*, const void*,
                               const void*, void*, long, long, long, void*,
                               size_t, hipStream_t);
+int hipblaslt_linear_bias(const void*, const void*, const void*, void*, long,
+                          long, long, void*, size_t, hipStream_t);
 template <typename T>
 void launch_roi_align(const T*, const float*, T*, int, int, int, int, int, int,
                       int, float, int, hipStream_t);
@@ -218,6 +220,27 @@ at::Tensor linear_bias_gelu(const at::Tensor& x, const at::Tensor& w,
   return y;
 }
 
+at::Tensor linear_bias(const at::Tensor& x, const at::Tensor& w,
+                       const at::Tensor& bias) {
+  TORCH_CHECK(x.is_cuda() && x.scalar_type() == at::kBFloat16, "linear_bias: bf16");
+  TORCH_CHECK(w.dim() == 2 && bias.dim() == 1 && w.size(0) == bias.size(0));
+  auto xc = x.contiguous();
+  const long K = x.size(-1), N = w.size(0);
+  const long M = x.numel() / K;
+  TORCH_CHECK(w.size(1) == K);
+  auto sizes = x.sizes().vec();
+  sizes.back() = N;
+  auto y = at::empty(sizes, x.options());
+  constexpr size_t kWs = 32L * 1024 * 1024;
+  hipStream_t stream = cur_stream();
+  void* ws = ws_for_stream(stream, x.options(), kWs);
+  int rc = hipblaslt_linear_bias(
+      xc.data_ptr(), w.contiguous().data_ptr(), bias.contiguous().data_ptr(),
+      y.data_ptr(), M, N, K, ws, kWs, stream);
+  TORCH_CHECK(rc == 0, "hipblaslt_linear_bias failed (no algo)");
+  return y;
+}
+
 at::Tensor linear_bias_residual(const at::Tensor& x, const at::Tensor& w,
                                 const at::Tensor& bias,
                                 const at::Tensor& residual) {
@@ -298,6 +321,7 @@ TORCH_LIBRARY(vilbert_amd, m) {
   m.def("roi_align(Tensor input, Tensor rois, int ph, int pw, float spatial_scale, int sampling_ratio) -> Tensor");
   m.def("linear_bias_gelu(Tensor x, Tensor w, Tensor bias) -> Tensor");
   m.def("linear_bias_residual(Tensor x, Tensor w, Tensor bias, Tensor residual) -> Tensor");
+  m.def("linear_bias(Tensor x, Tensor w, Tensor bias) -> Tensor");
 }
 
 TORCH_LIBRARY_IMPL(vilbert_amd, CUDA, m) {
@@ -311,4 +335,5 @@ TORCH_LIBRARY_IMPL(vilbert_amd, CUDA, m) {
   m.impl("roi_align", roi_align);
   m.impl("linear_bias_gelu", linear_bias_gelu);
   m.impl("linear_bias_residual", linear_bias_residual);
+  m.impl("linear_bias", linear_bias);
 }

@@ -61,6 +61,7 @@ std::mutex& plan_mu() {
 
 // kind 0: epilogue GELU_BIAS, beta=0.  kind 1: epilogue BIAS, beta=1 (the
 // C operand carries the residual: D = x@W^T + bias + residual).
+// kind 2: epilogue BIAS, beta=0 (plain autotuned linear).
 Plan& get_plan(long M, long N, long K, int kind, void* workspace, size_t ws_bytes) {
   auto key = std::make_tuple(M, N, K, kind);
   auto& cache = plan_cache();
@@ -76,6 +77,7 @@ Plan& get_plan(long M, long N, long K, int kind, void* workspace, size_t ws_byte
       p.op, HIPBLASLT_MATMUL_DESC_TRANSB, &tb, sizeof(tb)));
   hipblasLtEpilogue_t epi =
       kind == 0 ? HIPBLASLT_EPILOGUE_GELU_BIAS : HIPBLASLT_EPILOGUE_BIAS;
+  (void)epi;
   HIPBLASLT_CHECK(hipblasLtMatmulDescSetAttribute(
       p.op, HIPBLASLT_MATMUL_DESC_EPILOGUE, &epi, sizeof(epi)));
   // A = w [K,N] col-major view of row-major [N,K], opA = T -> [N,K]
@@ -97,7 +99,7 @@ Plan& get_plan(long M, long N, long K, int kind, void* workspace, size_t ws_byte
   if (found > 0) {
     p.algo = results[0].algo;
     p.has_algo = true;
-    p.beta = kind == 0 ? 0.0f : 1.0f;
+    p.beta = kind == 1 ? 1.0f : 0.0f;
     p.candidates.assign(results, results + found);
   }
   auto r = cache.emplace(key, p);
@@ -183,5 +185,12 @@ int hipblaslt_linear_bias_add(const void* x, const void* w, const void* bias,
                               long K, void* workspace, size_t ws_bytes,
                               hipStream_t stream) {
   return run_epilogue_gemm(1, x, w, bias, residual, y, M, N, K, workspace,
+                           ws_bytes, stream);
+}
+
+int hipblaslt_linear_bias(const void* x, const void* w, const void* bias,
+                          void* y, long M, long N, long K, void* workspace,
+                          size_t ws_bytes, hipStream_t stream) {
+  return run_epilogue_gemm(2, x, w, bias, nullptr, y, M, N, K, workspace,
                            ws_bytes, stream);
 }

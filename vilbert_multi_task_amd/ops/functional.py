@@ -119,6 +119,22 @@ def linear_bias_gelu(x: torch.Tensor, w: torch.Tensor, bias: torch.Tensor) -> to
     return bias_gelu(h, bias)
 
 
+_LINEAR_OK = True
+
+
+def linear_bias(x: torch.Tensor, w: torch.Tensor, bias: torch.Tensor) -> torch.Tensor:
+    """Autotuned hipBLASLt linear (bias epilogue) for the serving path;
+    plain F.linear otherwise."""
+    global _LINEAR_OK
+    if _LINEAR_OK and _want_hip(x, w, bias):
+        ext = _load_extension()
+        try:
+            return ext.linear_bias(x, w, bias)
+        except RuntimeError:
+            _LINEAR_OK = False
+    return torch.nn.functional.linear(x, w, bias)
+
+
 _LINEAR_RES_OK = True
 
 
