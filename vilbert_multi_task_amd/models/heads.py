@@ -129,6 +129,9 @@ class VILBertForVLTasks(nn.Module):
             linguisic_prediction = self.linguistic_prediction(self.dropout(t))
             linguisic_logit = self.linguistic_logit(self.dropout(t))
 
+        if getattr(self, "_fp8_ctx", None) is not None and not self.training:
+            self._fp8_ctx.update()  # delayed-scaling refresh (capturable)
+
         return (
             vil_prediction,
             vil_prediction_gqa,

@@ -164,6 +164,10 @@ class MultiHeadSelfAttention(nn.Module):
             q, k, v, self.heads,
             mask_bias, self.attn_dropout_p, self.training, need_probs,
         )
+        if infer and getattr(self, "_fp8_ctx_site", None) is not None:
+            from .fp8 import attach_quant_pack
+
+            attach_quant_pack(ctx, self._fp8_ctx_obj, self._fp8_ctx_site)
         # NOTE: folding the residual into the out-proj GEMM epilogue
         # (linear_bias_residual, beta=1) intermittently faults in hipBLASLt
         # at some shapes ("write access to a read-only page", B=512 warmup)
@@ -187,6 +191,15 @@ class FeedForward(nn.Module):
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         if isinstance(self.intermediate, nn.Linear):
             h = F_ops.linear_bias_gelu(x, self.intermediate.weight, self.intermediate.bias)
+        elif getattr(self, "_fp8_gelu_site", None) is not None and x.is_cuda:
+            # fp8 mode: GELU kernel also emits the e4m3 pack for the
+            # FFN-output GEMM (producer-fused quantization)
+            h, h8 = torch.ops.vilbert_amd.bias_gelu_fp8(
+                self.intermediate(x).contiguous(), None,
+                self._fp8_ctx_obj.scales, self._fp8_ctx_obj.amaxes,
+                self._fp8_gelu_site,
+            )
+            h._fp8 = (h8, self._fp8_ctx_obj.scales[self._fp8_gelu_site])
         else:  # Fp8Linear serving mode: fp8 matmul+bias, then erf GELU
             h = F_ops.bias_gelu(self.intermediate(x), None)
         if (
@@ -256,6 +269,10 @@ class CrossAttention(nn.Module):
             self.query(x_q), k, v, self.heads,
             mask_bias, self.attn_dropout_p, self.training, need_probs,
         )
+        if infer and getattr(self, "_fp8_ctx_site", None) is not None:
+            from .fp8 import attach_quant_pack
+
+            attach_quant_pack(ctx, self._fp8_ctx_obj, self._fp8_ctx_site)
         if infer and isinstance(self.out, nn.Linear):
             o = F_ops.linear_bias(ctx, self.out.weight, self.out.bias)
         else:

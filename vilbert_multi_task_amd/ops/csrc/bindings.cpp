@@ -37,6 +37,15 @@ int hipblaslt_linear_bias_add(const void*, const void*, const void*,
                               size_t, hipStream_t);
 int hipblaslt_linear_bias(const void*, const void*, const void*, void*, long,
                           long, long, void*, size_t, hipStream_t);
+void launch_residual_ln_fp8(const bf16*, const bf16*, const bf16*, const bf16*,
+                            bf16*, long, int, float, unsigned char*,
+                            const float*, float*, int, hipStream_t);
+void launch_bias_gelu_fp8(const bf16*, const bf16*, bf16*, long, int,
+                          unsigned char*, const float*, float*, int,
+                          hipStream_t);
+void launch_quantize_fp8(const bf16*, unsigned char*, const float*, float*, int,
+                         long, hipStream_t);
+void launch_update_fp8_scales(float*, float*, int, hipStream_t);
 template <typename T>
 void launch_roi_align(const T*, const float*, T*, int, int, int, int, int, int,
                       int, float, int, hipStream_t);
@@ -220,6 +229,59 @@ at::Tensor linear_bias_gelu(const at::Tensor& x, const at::Tensor& w,
   return y;
 }
 
+std::tuple<at::Tensor, at::Tensor> residual_layer_norm_fp8(
+    const at::Tensor& x, const c10::optional<at::Tensor>& res,
+    const at::Tensor& w, const at::Tensor& b, double eps,
+    const at::Tensor& scales, const at::Tensor& amaxes, int64_t site) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous() && x.scalar_type() == at::kBFloat16);
+  const int dim = (int)x.size(-1);
+  const long rows = x.numel() / dim;
+  auto y = at::empty_like(x);
+  auto y8 = at::empty(x.sizes(), x.options().dtype(at::kFloat8_e4m3fn));
+  const bool hr = res.has_value();
+  launch_residual_ln_fp8(
+      (const bf16*)x.data_ptr(), hr ? (const bf16*)res->data_ptr() : nullptr,
+      (const bf16*)w.data_ptr(), (const bf16*)b.data_ptr(), (bf16*)y.data_ptr(),
+      rows, dim, (float)eps, (unsigned char*)y8.data_ptr(),
+      scales.data_ptr<float>(), amaxes.data_ptr<float>(), (int)site,
+      cur_stream());
+  return {y, y8};
+}
+
+std::tuple<at::Tensor, at::Tensor> bias_gelu_fp8(
+    const at::Tensor& x, const c10::optional<at::Tensor>& bias,
+    const at::Tensor& scales, const at::Tensor& amaxes, int64_t site) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous() && x.scalar_type() == at::kBFloat16);
+  const int dim = (int)x.size(-1);
+  const long n = x.numel();
+  auto y = at::empty_like(x);
+  auto y8 = at::empty(x.sizes(), x.options().dtype(at::kFloat8_e4m3fn));
+  const bool hb = bias.has_value();
+  launch_bias_gelu_fp8((const bf16*)x.data_ptr(),
+                       hb ? (const bf16*)bias->data_ptr() : nullptr,
+                       (bf16*)y.data_ptr(), n, dim,
+                       (unsigned char*)y8.data_ptr(), scales.data_ptr<float>(),
+                       amaxes.data_ptr<float>(), (int)site, cur_stream());
+  return {y, y8};
+}
+
+at::Tensor quantize_fp8(const at::Tensor& x, const at::Tensor& scales,
+                        const at::Tensor& amaxes, int64_t site) {
+  TORCH_CHECK(x.is_cuda() && x.scalar_type() == at::kBFloat16);
+  auto xc = x.contiguous();
+  auto y8 = at::empty(x.sizes(), x.options().dtype(at::kFloat8_e4m3fn));
+  launch_quantize_fp8((const bf16*)xc.data_ptr(), (unsigned char*)y8.data_ptr(),
+                      scales.data_ptr<float>(), amaxes.data_ptr<float>(),
+                      (int)site, x.numel(), cur_stream());
+  return y8;
+}
+
+void update_fp8_scales(at::Tensor& scales, at::Tensor& amaxes) {
+  TORCH_CHECK(scales.is_cuda() && scales.scalar_type() == at::kFloat);
+  launch_update_fp8_scales(scales.data_ptr<float>(), amaxes.data_ptr<float>(),
+                           (int)scales.numel(), cur_stream());
+}
+
 at::Tensor linear_bias(const at::Tensor& x, const at::Tensor& w,
                        const at::Tensor& bias) {
   TORCH_CHECK(x.is_cuda() && x.scalar_type() == at::kBFloat16, "linear_bias: bf16");
@@ -322,6 +384,10 @@ TORCH_LIBRARY(vilbert_amd, m) {
   m.def("linear_bias_gelu(Tensor x, Tensor w, Tensor bias) -> Tensor");
   m.def("linear_bias_residual(Tensor x, Tensor w, Tensor bias, Tensor residual) -> Tensor");
   m.def("linear_bias(Tensor x, Tensor w, Tensor bias) -> Tensor");
+  m.def("residual_layer_norm_fp8(Tensor x, Tensor? res, Tensor w, Tensor b, float eps, Tensor scales, Tensor amaxes, int site) -> (Tensor, Tensor)");
+  m.def("bias_gelu_fp8(Tensor x, Tensor? bias, Tensor scales, Tensor amaxes, int site) -> (Tensor, Tensor)");
+  m.def("quantize_fp8(Tensor x, Tensor scales, Tensor amaxes, int site) -> Tensor");
+  m.def("update_fp8_scales(Tensor(a!) scales, Tensor(b!) amaxes) -> ()");
 }
 
 TORCH_LIBRARY_IMPL(vilbert_amd, CUDA, m) {
@@ -336,4 +402,8 @@ TORCH_LIBRARY_IMPL(vilbert_amd, CUDA, m) {
   m.impl("linear_bias_gelu", linear_bias_gelu);
   m.impl("linear_bias_residual", linear_bias_residual);
   m.impl("linear_bias", linear_bias);
+  m.impl("residual_layer_norm_fp8", residual_layer_norm_fp8);
+  m.impl("bias_gelu_fp8", bias_gelu_fp8);
+  m.impl("quantize_fp8", quantize_fp8);
+  m.impl("update_fp8_scales", update_fp8_scales);
 }

@@ -7,7 +7,19 @@
 // ~2-2.5x slower), f32 accumulation, fused residual add so the tensor is
 // read once from HBM instead of twice.
 
+#include <hip/hip_fp8.h>
+
 #include "common.h"
+
+#define E4M3_MAX_E 448.0f
+
+DEV void atomic_max_f32_nonneg_e(float* addr, float v) {
+  atomicMax(reinterpret_cast<unsigned int*>(addr), __float_as_uint(v));
+}
+DEV unsigned char f2e4m3_e(float v) {
+  __hip_fp8_e4m3 q(fminf(fmaxf(v, -E4M3_MAX_E), E4M3_MAX_E));
+  return q.__x;
+}
 
 // ---------------------------------------------------------------------------
 // residual + LayerNorm.  One wave per row; rows assigned grid-stride.
@@ -63,13 +75,22 @@ struct VecIO<float> {
 // beyond that re-read from L2 in the normalize pass).
 #define LN_MAX_CHUNKS 4
 
-template <typename T, bool HAS_RES>
+template <typename T, bool HAS_RES, bool FP8OUT = false>
 __global__ void residual_ln_kernel(const T* __restrict__ x,
                                    const T* __restrict__ res,
                                    const T* __restrict__ w,
                                    const T* __restrict__ b,
                                    T* __restrict__ y,
-                                   long rows, int dim, float eps) {
+                                   long rows, int dim, float eps,
+                                   unsigned char* __restrict__ y8 = nullptr,
+                                   const float* __restrict__ scales = nullptr,
+                                   float* __restrict__ amaxes = nullptr,
+                                   int site = 0) {
+  // FP8OUT: also emit an e4m3 copy of y scaled by scales[site] (delayed
+  // scaling: previous step's scale) and accumulate this step's amax —
+  // makes fp8 GEMM-input quantization free (docs/ROADMAP.md item 1).
+  const float fp8_inv = FP8OUT ? 1.0f / scales[site] : 0.f;
+  float fp8_amax = 0.f;
   const int lane = lane_id();
   const int wid = wave_id();
   const int waves_per_blk = blockDim.x / WAVE;
@@ -130,16 +151,35 @@ __global__ void residual_ln_kernel(const T* __restrict__ x,
           for (int i = 0; i < 8; ++i)
             o8[i] = (cache[c * 8 + i] - mean) * rstd * w8[i] + b8[i];
           VecIO<T>::store8(yr + j, o8);
+          if (FP8OUT) {
+            unsigned char q[8];
+#pragma unroll
+            for (int i = 0; i < 8; ++i) {
+              fp8_amax = fmaxf(fp8_amax, fabsf(o8[i]));
+              q[i] = f2e4m3_e(o8[i] * fp8_inv);
+            }
+            *reinterpret_cast<uint2*>(y8 + row * dim + j) =
+                *reinterpret_cast<uint2*>(q);
+          }
         }
       }
     } else {
       for (int j = lane; j < dim; j += WAVE) {
         float v = VecIO<T>::ld(xr + j);
         if (HAS_RES) v += VecIO<T>::ld(rr + j);
-        VecIO<T>::st(yr + j,
-                     (v - mean) * rstd * VecIO<T>::ld(w + j) + VecIO<T>::ld(b + j));
+        const float o = (v - mean) * rstd * VecIO<T>::ld(w + j) + VecIO<T>::ld(b + j);
+        VecIO<T>::st(yr + j, o);
+        if (FP8OUT) {
+          fp8_amax = fmaxf(fp8_amax, fabsf(o));
+          y8[row * dim + j] = f2e4m3_e(o * fp8_inv);
+        }
       }
     }
+  }
+  if (FP8OUT) {
+    fp8_amax = wave_max(fp8_amax);
+    if (lane_id() == 0 && fp8_amax > 0.f)
+      atomic_max_f32_nonneg_e(&amaxes[site], fp8_amax);
   }
 }
 
@@ -147,11 +187,17 @@ __global__ void residual_ln_kernel(const T* __restrict__ x,
 // bias + exact GELU:  y = 0.5*(x+b)*(1+erf((x+b)/sqrt(2)))
 // ---------------------------------------------------------------------------
 
-template <typename T, bool HAS_BIAS>
+template <typename T, bool HAS_BIAS, bool FP8OUT = false>
 __global__ void bias_gelu_kernel(const T* __restrict__ x,
                                  const T* __restrict__ bias,
                                  T* __restrict__ y,
-                                 long n, int dim) {
+                                 long n, int dim,
+                                 unsigned char* __restrict__ y8 = nullptr,
+                                 const float* __restrict__ scales = nullptr,
+                                 float* __restrict__ amaxes = nullptr,
+                                 int site = 0) {
+  const float fp8_inv = FP8OUT ? 1.0f / scales[site] : 0.f;
+  float fp8_amax = 0.f;
   const long i8 = ((long)blockIdx.x * blockDim.x + threadIdx.x) * 8;
   const long stride8 = (long)gridDim.x * blockDim.x * 8;
   constexpr float kInvSqrt2 = 0.70710678118654752440f;
@@ -169,13 +215,32 @@ __global__ void bias_gelu_kernel(const T* __restrict__ x,
       for (int j = 0; j < 8; ++j)
         v[j] = 0.5f * v[j] * (1.0f + erff(v[j] * kInvSqrt2));
       VecIO<T>::store8(y + i, v);
+      if (FP8OUT) {
+        unsigned char q[8];
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          fp8_amax = fmaxf(fp8_amax, fabsf(v[j]));
+          q[j] = f2e4m3_e(v[j] * fp8_inv);
+        }
+        *reinterpret_cast<uint2*>(y8 + i) = *reinterpret_cast<uint2*>(q);
+      }
     } else {
       for (long k = i; k < min(i + 8, n); ++k) {
         float v = VecIO<T>::ld(x + k);
         if (HAS_BIAS) v += VecIO<T>::ld(bias + (k % dim));
-        VecIO<T>::st(y + k, 0.5f * v * (1.0f + erff(v * kInvSqrt2)));
+        const float g = 0.5f * v * (1.0f + erff(v * kInvSqrt2));
+        VecIO<T>::st(y + k, g);
+        if (FP8OUT) {
+          fp8_amax = fmaxf(fp8_amax, fabsf(g));
+          y8[k] = f2e4m3_e(g * fp8_inv);
+        }
       }
     }
+  }
+  if (FP8OUT) {
+    fp8_amax = wave_max(fp8_amax);
+    if (lane_id() == 0 && fp8_amax > 0.f)
+      atomic_max_f32_nonneg_e(&amaxes[site], fp8_amax);
   }
 }
 
@@ -267,6 +332,24 @@ __global__ void embedding_ln_kernel(const long* __restrict__ ids,
 // host-side launchers (called from bindings.cpp)
 // ---------------------------------------------------------------------------
 
+void launch_residual_ln_fp8(const bf16* x, const bf16* res, const bf16* w,
+                            const bf16* b, bf16* y, long rows, int dim,
+                            float eps, unsigned char* y8, const float* scales,
+                            float* amaxes, int site, hipStream_t stream) {
+  const int block = 256;
+  const int waves = block / WAVE;
+  long want = (rows + waves - 1) / waves;
+  const int grid = (int)(want < 2048 ? want : 2048);
+  if (res)
+    hipLaunchKernelGGL((residual_ln_kernel<bf16, true, true>), dim3(grid),
+                       dim3(block), 0, stream, x, res, w, b, y, rows, dim, eps,
+                       y8, scales, amaxes, site);
+  else
+    hipLaunchKernelGGL((residual_ln_kernel<bf16, false, true>), dim3(grid),
+                       dim3(block), 0, stream, x, res, w, b, y, rows, dim, eps,
+                       y8, scales, amaxes, site);
+}
+
 template <typename T>
 void launch_residual_ln(const T* x, const T* res, const T* w, const T* b, T* y,
                         long rows, int dim, float eps, hipStream_t stream) {
@@ -279,6 +362,22 @@ void launch_residual_ln(const T* x, const T* res, const T* w, const T* b, T* y,
   else
     hipLaunchKernelGGL((residual_ln_kernel<T, false>), dim3(grid), dim3(block), 0,
                        stream, x, res, w, b, y, rows, dim, eps);
+}
+
+void launch_bias_gelu_fp8(const bf16* x, const bf16* bias, bf16* y, long n,
+                          int dim, unsigned char* y8, const float* scales,
+                          float* amaxes, int site, hipStream_t stream) {
+  const int block = 256;
+  const long want = (n + 8 * block - 1) / (8 * block);
+  const int grid = (int)(want < 2048 ? want : 2048);
+  if (bias)
+    hipLaunchKernelGGL((bias_gelu_kernel<bf16, true, true>), dim3(grid),
+                       dim3(block), 0, stream, x, bias, y, n, dim, y8, scales,
+                       amaxes, site);
+  else
+    hipLaunchKernelGGL((bias_gelu_kernel<bf16, false, true>), dim3(grid),
+                       dim3(block), 0, stream, x, bias, y, n, dim, y8, scales,
+                       amaxes, site);
 }
 
 template <typename T>
