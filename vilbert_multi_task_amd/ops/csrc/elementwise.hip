@@ -20,6 +20,24 @@ DEV unsigned char f2e4m3_e(float v) {
   __hip_fp8_e4m3 q(fminf(fmaxf(v, -E4M3_MAX_E), E4M3_MAX_E));
   return q.__x;
 }
+// hardware packed convert: 8 floats -> 8 e4m3 bytes in 4 v_cvt_pk_fp8_f32
+// (the __hip_fp8_e4m3 constructor is a software path — it made the FP8OUT
+// LayerNorm 2.8x slower than plain; profiles/prof_fp8)
+DEV uint2 pack8_e4m3(const float* v, float inv) {
+  float c[8];
+#pragma unroll
+  for (int i = 0; i < 8; ++i)
+    c[i] = fminf(fmaxf(v[i] * inv, -E4M3_MAX_E), E4M3_MAX_E);
+  int lo = 0, hi = 0;
+  lo = __builtin_amdgcn_cvt_pk_fp8_f32(c[0], c[1], lo, false);
+  lo = __builtin_amdgcn_cvt_pk_fp8_f32(c[2], c[3], lo, true);
+  hi = __builtin_amdgcn_cvt_pk_fp8_f32(c[4], c[5], hi, false);
+  hi = __builtin_amdgcn_cvt_pk_fp8_f32(c[6], c[7], hi, true);
+  uint2 r;
+  r.x = (unsigned int)lo;
+  r.y = (unsigned int)hi;
+  return r;
+}
 
 // ---------------------------------------------------------------------------
 // residual + LayerNorm.  One wave per row; rows assigned grid-stride.
@@ -152,14 +170,11 @@ __global__ void residual_ln_kernel(const T* __restrict__ x,
             o8[i] = (cache[c * 8 + i] - mean) * rstd * w8[i] + b8[i];
           VecIO<T>::store8(yr + j, o8);
           if (FP8OUT) {
-            unsigned char q[8];
 #pragma unroll
-            for (int i = 0; i < 8; ++i) {
+            for (int i = 0; i < 8; ++i)
               fp8_amax = fmaxf(fp8_amax, fabsf(o8[i]));
-              q[i] = f2e4m3_e(o8[i] * fp8_inv);
-            }
             *reinterpret_cast<uint2*>(y8 + row * dim + j) =
-                *reinterpret_cast<uint2*>(q);
+                pack8_e4m3(o8, fp8_inv);
           }
         }
       }
@@ -216,13 +231,10 @@ __global__ void bias_gelu_kernel(const T* __restrict__ x,
         v[j] = 0.5f * v[j] * (1.0f + erff(v[j] * kInvSqrt2));
       VecIO<T>::store8(y + i, v);
       if (FP8OUT) {
-        unsigned char q[8];
 #pragma unroll
-        for (int j = 0; j < 8; ++j) {
+        for (int j = 0; j < 8; ++j)
           fp8_amax = fmaxf(fp8_amax, fabsf(v[j]));
-          q[j] = f2e4m3_e(v[j] * fp8_inv);
-        }
-        *reinterpret_cast<uint2*>(y8 + i) = *reinterpret_cast<uint2*>(q);
+        *reinterpret_cast<uint2*>(y8 + i) = pack8_e4m3(v, fp8_inv);
       }
     } else {
       for (long k = i; k < min(i + 8, n); ++k) {

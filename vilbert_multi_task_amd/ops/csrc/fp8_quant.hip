@@ -26,6 +26,22 @@ DEV unsigned char f2e4m3(float v) {
   return q.__x;
 }
 
+DEV uint2 pack8_e4m3_q(const float* v, float inv) {
+  float c[8];
+#pragma unroll
+  for (int i = 0; i < 8; ++i)
+    c[i] = fminf(fmaxf(v[i] * inv, -E4M3_MAX), E4M3_MAX);
+  int lo = 0, hi = 0;
+  lo = __builtin_amdgcn_cvt_pk_fp8_f32(c[0], c[1], lo, false);
+  lo = __builtin_amdgcn_cvt_pk_fp8_f32(c[2], c[3], lo, true);
+  hi = __builtin_amdgcn_cvt_pk_fp8_f32(c[4], c[5], hi, false);
+  hi = __builtin_amdgcn_cvt_pk_fp8_f32(c[6], c[7], hi, true);
+  uint2 r;
+  r.x = (unsigned int)lo;
+  r.y = (unsigned int)hi;
+  return r;
+}
+
 DEV void VecIO_q_load(const bf16* p, float* out) {
   const uint4 raw = *reinterpret_cast<const uint4*>(p);
   const unsigned int w[4] = {raw.x, raw.y, raw.z, raw.w};
@@ -53,14 +69,9 @@ __global__ void quantize_fp8_kernel(const bf16* __restrict__ x,
     if (i + 8 <= n) {
       float v[8];
       VecIO_q_load(x + i, v);
-      unsigned char q[8];
 #pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        const float a = fabsf(v[j]);
-        local = fmaxf(local, a);
-        q[j] = f2e4m3(fminf(fmaxf(v[j] * inv, -E4M3_MAX), E4M3_MAX));
-      }
-      *reinterpret_cast<uint2*>(y8 + i) = *reinterpret_cast<uint2*>(q);
+      for (int j = 0; j < 8; ++j) local = fmaxf(local, fabsf(v[j]));
+      *reinterpret_cast<uint2*>(y8 + i) = pack8_e4m3_q(v, inv);
     } else {
       for (long k = i; k < n; ++k) {
         const float v = bf2f(x[k]);
