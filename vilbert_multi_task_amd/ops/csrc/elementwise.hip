@@ -14,6 +14,10 @@
 #define E4M3_MAX_E 448.0f
 
 DEV void atomic_max_f32_nonneg_e(float* addr, float v) {
+  // read-guard: skip the atomic once the global amax already covers v —
+  // without it every wave serializes on one L2 dword (measured: the FP8OUT
+  // LN was 2.8x slower than plain LN at unchanged occupancy/conversions)
+  if (v <= *reinterpret_cast<volatile float*>(addr)) return;
   atomicMax(reinterpret_cast<unsigned int*>(addr), __float_as_uint(v));
 }
 DEV unsigned char f2e4m3_e(float v) {

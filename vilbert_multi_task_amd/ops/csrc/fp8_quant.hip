@@ -17,7 +17,9 @@
 #define E4M3_MAX 448.0f
 
 DEV void atomic_max_f32_nonneg(float* addr, float v) {
-  // monotone uint ordering holds for non-negative floats
+  // monotone uint ordering holds for non-negative floats; read-guard skips
+  // the atomic when the global amax already covers v (contention fix)
+  if (v <= *reinterpret_cast<volatile float*>(addr)) return;
   atomicMax(reinterpret_cast<unsigned int*>(addr), __float_as_uint(v));
 }
 
