@@ -38,11 +38,14 @@ class Fp8Context:
 
     def __init__(self, nsites: int, device):
         self.scales = torch.ones(nsites, dtype=torch.float32, device=device)
+        self.inv_scales = torch.ones(nsites, dtype=torch.float32, device=device)
         self.amaxes = torch.zeros(nsites, dtype=torch.float32, device=device)
         self.nsites = nsites
 
     def update(self) -> None:
-        torch.ops.vilbert_amd.update_fp8_scales(self.scales, self.amaxes)
+        torch.ops.vilbert_amd.update_fp8_scales(
+            self.scales, self.inv_scales, self.amaxes
+        )
 
 
 class _SiteAllocator:
@@ -83,6 +86,12 @@ def fp8_mm(
         x8 = x8.reshape(-1, shape[-1])
     else:
         x8, x_scale = _dynamic_quant(x.reshape(-1, shape[-1]))
+    if bias is not None:
+        try:
+            y = torch.ops.vilbert_amd.fp8_linear(x8, w8, bias, w_scale, x_scale)
+            return y.reshape(*shape[:-1], y.shape[-1])
+        except RuntimeError:
+            pass  # no hipBLASLt algo: _scaled_mm fallback
     y = torch._scaled_mm(
         x8, w8.t(), scale_a=x_scale, scale_b=w_scale, bias=bias,
         out_dtype=torch.bfloat16,

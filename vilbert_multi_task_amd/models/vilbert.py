@@ -192,14 +192,38 @@ class FeedForward(nn.Module):
         if isinstance(self.intermediate, nn.Linear):
             h = F_ops.linear_bias_gelu(x, self.intermediate.weight, self.intermediate.bias)
         elif getattr(self, "_fp8_gelu_site", None) is not None and x.is_cuda:
-            # fp8 mode: GELU kernel also emits the e4m3 pack for the
-            # FFN-output GEMM (producer-fused quantization)
-            h, h8 = torch.ops.vilbert_amd.bias_gelu_fp8(
-                self.intermediate(x).contiguous(), None,
-                self._fp8_ctx_obj.scales, self._fp8_ctx_obj.amaxes,
-                self._fp8_gelu_site,
-            )
-            h._fp8 = (h8, self._fp8_ctx_obj.scales[self._fp8_gelu_site])
+            # full-fp8 FFN chain: ONE hipBLASLt call does fp8 GEMM + bias +
+            # GELU and emits the e4m3 intermediate directly (D-scale =
+            # delayed inv-scale, AMAX_D -> the site's amax slot); fallback
+            # to the two-step path if no algo supports the combo
+            ctx8 = self._fp8_ctx_obj
+            site = self._fp8_gelu_site
+            pack = getattr(x, "_fp8", None)
+            h = None
+            if pack is not None:
+                from .fp8 import fp8_mm  # noqa: F401 (fallback path below)
+
+                x8, x_scale = pack
+                inter = self.intermediate  # Fp8Linear
+                try:
+                    h8 = torch.ops.vilbert_amd.fp8_linear_gelu_fp8out(
+                        x8.reshape(-1, x.shape[-1]), inter.w8, inter.bias,
+                        inter.w_scale, x_scale,
+                        ctx8.inv_scales[site], ctx8.amaxes[site],
+                    )
+                    h8 = h8.reshape(*x.shape[:-1], inter.out_features)
+                    # the bf16 intermediate never materializes: h8 itself is
+                    # the carrier — the output Fp8Linear only reads ._fp8
+                    h = h8
+                    h._fp8 = (h8, ctx8.scales[site])
+                except RuntimeError:
+                    h = None
+            if h is None:
+                h, h8 = torch.ops.vilbert_amd.bias_gelu_fp8(
+                    self.intermediate(x).contiguous(), None,
+                    ctx8.scales, ctx8.amaxes, site,
+                )
+                h._fp8 = (h8, ctx8.scales[site])
         else:  # Fp8Linear serving mode: fp8 matmul+bias, then erf GELU
             h = F_ops.bias_gelu(self.intermediate(x), None)
         if (

@@ -45,7 +45,14 @@ void launch_bias_gelu_fp8(const bf16*, const bf16*, bf16*, long, int,
                           hipStream_t);
 void launch_quantize_fp8(const bf16*, unsigned char*, const float*, float*, int,
                          long, hipStream_t);
-void launch_update_fp8_scales(float*, float*, int, hipStream_t);
+void launch_update_fp8_scales(float*, float*, float*, int, hipStream_t);
+int hipblaslt_fp8_linear(const void*, const void*, const void*, const void*,
+                         const void*, void*, long, long, long, void*, size_t,
+                         hipStream_t);
+int hipblaslt_fp8_linear_gelu_fp8out(const void*, const void*, const void*,
+                                     const void*, const void*, const void*,
+                                     void*, void*, long, long, long, void*,
+                                     size_t, hipStream_t);
 template <typename T>
 void launch_roi_align(const T*, const float*, T*, int, int, int, int, int, int,
                       int, float, int, hipStream_t);
@@ -276,10 +283,55 @@ at::Tensor quantize_fp8(const at::Tensor& x, const at::Tensor& scales,
   return y8;
 }
 
-void update_fp8_scales(at::Tensor& scales, at::Tensor& amaxes) {
+void update_fp8_scales(at::Tensor& scales, at::Tensor& inv_scales,
+                       at::Tensor& amaxes) {
   TORCH_CHECK(scales.is_cuda() && scales.scalar_type() == at::kFloat);
-  launch_update_fp8_scales(scales.data_ptr<float>(), amaxes.data_ptr<float>(),
-                           (int)scales.numel(), cur_stream());
+  launch_update_fp8_scales(scales.data_ptr<float>(),
+                           inv_scales.data_ptr<float>(),
+                           amaxes.data_ptr<float>(), (int)scales.numel(),
+                           cur_stream());
+}
+
+at::Tensor fp8_linear(const at::Tensor& x8, const at::Tensor& w8,
+                      const at::Tensor& bias, const at::Tensor& w_scale,
+                      const at::Tensor& x_scale) {
+  TORCH_CHECK(x8.is_cuda() && x8.scalar_type() == at::kFloat8_e4m3fn);
+  const long K = x8.size(-1), N = w8.size(0);
+  const long M = x8.numel() / K;
+  auto sizes = x8.sizes().vec();
+  sizes.back() = N;
+  auto y = at::empty(sizes, x8.options().dtype(at::kBFloat16));
+  constexpr size_t kWs = 32L * 1024 * 1024;
+  hipStream_t stream = cur_stream();
+  void* ws = ws_for_stream(stream, x8.options(), kWs);
+  int rc = hipblaslt_fp8_linear(
+      x8.data_ptr(), w8.data_ptr(), bias.data_ptr(), w_scale.data_ptr(),
+      x_scale.data_ptr(), y.data_ptr(), M, N, K, ws, kWs, stream);
+  TORCH_CHECK(rc == 0, "hipblaslt_fp8_linear: no algo");
+  return y;
+}
+
+at::Tensor fp8_linear_gelu_fp8out(const at::Tensor& x8, const at::Tensor& w8,
+                                  const at::Tensor& bias,
+                                  const at::Tensor& w_scale,
+                                  const at::Tensor& x_scale,
+                                  const at::Tensor& d_inv_scale,
+                                  at::Tensor& amax_out) {
+  TORCH_CHECK(x8.is_cuda() && x8.scalar_type() == at::kFloat8_e4m3fn);
+  const long K = x8.size(-1), N = w8.size(0);
+  const long M = x8.numel() / K;
+  auto sizes = x8.sizes().vec();
+  sizes.back() = N;
+  auto y8 = at::empty(sizes, x8.options());
+  constexpr size_t kWs = 32L * 1024 * 1024;
+  hipStream_t stream = cur_stream();
+  void* ws = ws_for_stream(stream, x8.options(), kWs);
+  int rc = hipblaslt_fp8_linear_gelu_fp8out(
+      x8.data_ptr(), w8.data_ptr(), bias.data_ptr(), w_scale.data_ptr(),
+      x_scale.data_ptr(), d_inv_scale.data_ptr(), amax_out.data_ptr(),
+      y8.data_ptr(), M, N, K, ws, kWs, stream);
+  TORCH_CHECK(rc == 0, "hipblaslt_fp8_linear_gelu_fp8out: no algo");
+  return y8;
 }
 
 at::Tensor linear_bias(const at::Tensor& x, const at::Tensor& w,
@@ -387,7 +439,9 @@ TORCH_LIBRARY(vilbert_amd, m) {
   m.def("residual_layer_norm_fp8(Tensor x, Tensor? res, Tensor w, Tensor b, float eps, Tensor scales, Tensor amaxes, int site) -> (Tensor, Tensor)");
   m.def("bias_gelu_fp8(Tensor x, Tensor? bias, Tensor scales, Tensor amaxes, int site) -> (Tensor, Tensor)");
   m.def("quantize_fp8(Tensor x, Tensor scales, Tensor amaxes, int site) -> Tensor");
-  m.def("update_fp8_scales(Tensor(a!) scales, Tensor(b!) amaxes) -> ()");
+  m.def("update_fp8_scales(Tensor(a!) scales, Tensor(b!) inv_scales, Tensor(c!) amaxes) -> ()");
+  m.def("fp8_linear(Tensor x8, Tensor w8, Tensor bias, Tensor w_scale, Tensor x_scale) -> Tensor");
+  m.def("fp8_linear_gelu_fp8out(Tensor x8, Tensor w8, Tensor bias, Tensor w_scale, Tensor x_scale, Tensor d_inv_scale, Tensor(a!) amax_out) -> Tensor");
 }
 
 TORCH_LIBRARY_IMPL(vilbert_amd, CUDA, m) {
@@ -406,4 +460,6 @@ TORCH_LIBRARY_IMPL(vilbert_amd, CUDA, m) {
   m.impl("bias_gelu_fp8", bias_gelu_fp8);
   m.impl("quantize_fp8", quantize_fp8);
   m.impl("update_fp8_scales", update_fp8_scales);
+  m.impl("fp8_linear", fp8_linear);
+  m.impl("fp8_linear_gelu_fp8out", fp8_linear_gelu_fp8out);
 }

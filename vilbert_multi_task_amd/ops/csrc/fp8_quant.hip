@@ -92,12 +92,17 @@ __global__ void quantize_fp8_kernel(const bf16* __restrict__ x,
 // ---------------------------------------------------------------------------
 
 __global__ void update_fp8_scales_kernel(float* __restrict__ scales,
+                                         float* __restrict__ inv_scales,
                                          float* __restrict__ amaxes,
                                          int nsites) {
   const int i = blockIdx.x * blockDim.x + threadIdx.x;
   if (i < nsites) {
     const float a = amaxes[i];
-    if (a > 0.f) scales[i] = fmaxf(a, 1e-8f) / E4M3_MAX;
+    if (a > 0.f) {
+      const float s = fmaxf(a, 1e-8f) / E4M3_MAX;
+      scales[i] = s;
+      inv_scales[i] = 1.0f / s;  // hipBLASLt D_SCALE multiplies before cast
+    }
     amaxes[i] = 0.f;
   }
 }
@@ -111,8 +116,8 @@ void launch_quantize_fp8(const bf16* x, unsigned char* y8, const float* scales,
                      y8, scales, amaxes, site, n);
 }
 
-void launch_update_fp8_scales(float* scales, float* amaxes, int nsites,
-                              hipStream_t stream) {
+void launch_update_fp8_scales(float* scales, float* inv_scales, float* amaxes,
+                              int nsites, hipStream_t stream) {
   hipLaunchKernelGGL(update_fp8_scales_kernel, dim3(ceil_div(nsites, 64)),
-                     dim3(64), 0, stream, scales, amaxes, nsites);
+                     dim3(64), 0, stream, scales, inv_scales, amaxes, nsites);
 }

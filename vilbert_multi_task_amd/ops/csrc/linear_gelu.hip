@@ -62,12 +62,18 @@ std::mutex& plan_mu() {
 // kind 0: epilogue GELU_BIAS, beta=0.  kind 1: epilogue BIAS, beta=1 (the
 // C operand carries the residual: D = x@W^T + bias + residual).
 // kind 2: epilogue BIAS, beta=0 (plain autotuned linear).
+// kind 3: fp8(e4m3) A/B -> bf16 D, epilogue BIAS, per-tensor scales.
+// kind 4: fp8 A/B -> fp8 D, epilogue GELU_BIAS, D scaled by *d_scale with
+//         AMAX_D written to the site's amax slot (full-fp8 FFN chain).
 Plan& get_plan(long M, long N, long K, int kind, void* workspace, size_t ws_bytes) {
   auto key = std::make_tuple(M, N, K, kind);
   auto& cache = plan_cache();
   auto it = cache.find(key);
   if (it != cache.end()) return it->second;
 
+  const bool fp8_in = kind >= 3;
+  const hipDataType ab_t = fp8_in ? HIP_R_8F_E4M3 : HIP_R_16BF;
+  const hipDataType d_t = (kind == 4) ? HIP_R_8F_E4M3 : HIP_R_16BF;
   Plan p;
   HIPBLASLT_CHECK(hipblasLtMatmulDescCreate(&p.op, HIPBLAS_COMPUTE_32F, HIP_R_32F));
   hipblasOperation_t ta = HIPBLAS_OP_T, tb = HIPBLAS_OP_N;
@@ -76,16 +82,21 @@ Plan& get_plan(long M, long N, long K, int kind, void* workspace, size_t ws_byte
   HIPBLASLT_CHECK(hipblasLtMatmulDescSetAttribute(
       p.op, HIPBLASLT_MATMUL_DESC_TRANSB, &tb, sizeof(tb)));
   hipblasLtEpilogue_t epi =
-      kind == 0 ? HIPBLASLT_EPILOGUE_GELU_BIAS : HIPBLASLT_EPILOGUE_BIAS;
-  (void)epi;
+      (kind == 0 || kind == 4) ? HIPBLASLT_EPILOGUE_GELU_BIAS
+                               : HIPBLASLT_EPILOGUE_BIAS;
   HIPBLASLT_CHECK(hipblasLtMatmulDescSetAttribute(
       p.op, HIPBLASLT_MATMUL_DESC_EPILOGUE, &epi, sizeof(epi)));
+  if (fp8_in) {
+    hipDataType bias_t = HIP_R_16BF;
+    HIPBLASLT_CHECK(hipblasLtMatmulDescSetAttribute(
+        p.op, HIPBLASLT_MATMUL_DESC_BIAS_DATA_TYPE, &bias_t, sizeof(bias_t)));
+  }
   // A = w [K,N] col-major view of row-major [N,K], opA = T -> [N,K]
-  HIPBLASLT_CHECK(hipblasLtMatrixLayoutCreate(&p.a, HIP_R_16BF, K, N, K));
+  HIPBLASLT_CHECK(hipblasLtMatrixLayoutCreate(&p.a, ab_t, K, N, K));
   // B = x [K,M] col-major view of row-major [M,K]
-  HIPBLASLT_CHECK(hipblasLtMatrixLayoutCreate(&p.b, HIP_R_16BF, K, M, K));
+  HIPBLASLT_CHECK(hipblasLtMatrixLayoutCreate(&p.b, ab_t, K, M, K));
   // C = y [N,M] col-major view of row-major [M,N]
-  HIPBLASLT_CHECK(hipblasLtMatrixLayoutCreate(&p.c, HIP_R_16BF, N, M, N));
+  HIPBLASLT_CHECK(hipblasLtMatrixLayoutCreate(&p.c, d_t, N, M, N));
 
   hipblasLtMatmulPreference_t pref;
   HIPBLASLT_CHECK(hipblasLtMatmulPreferenceCreate(&pref));
@@ -193,4 +204,64 @@ int hipblaslt_linear_bias(const void* x, const void* w, const void* bias,
                           size_t ws_bytes, hipStream_t stream) {
   return run_epilogue_gemm(2, x, w, bias, nullptr, y, M, N, K, workspace,
                            ws_bytes, stream);
+}
+
+// fp8 matmuls: per-tensor scale pointers (device); kind 4 also writes the
+// e4m3 D with *d_inv_scale applied and AMAX(D) to *amax_out.
+static int run_fp8_gemm(int kind, const void* x8, const void* w8,
+                        const void* bias, const void* w_scale,
+                        const void* x_scale, const void* d_inv_scale,
+                        void* amax_out, void* y, long M, long N, long K,
+                        void* workspace, size_t ws_bytes, hipStream_t stream) {
+  try {
+    std::lock_guard<std::mutex> lock(plan_mu());
+    Plan& p = get_plan(M, N, K, kind, workspace, ws_bytes);
+    if (!p.has_algo) return 1;
+    HIPBLASLT_CHECK(hipblasLtMatmulDescSetAttribute(
+        p.op, HIPBLASLT_MATMUL_DESC_BIAS_POINTER, &bias, sizeof(bias)));
+    HIPBLASLT_CHECK(hipblasLtMatmulDescSetAttribute(
+        p.op, HIPBLASLT_MATMUL_DESC_A_SCALE_POINTER, &w_scale, sizeof(w_scale)));
+    HIPBLASLT_CHECK(hipblasLtMatmulDescSetAttribute(
+        p.op, HIPBLASLT_MATMUL_DESC_B_SCALE_POINTER, &x_scale, sizeof(x_scale)));
+    if (kind == 4) {
+      HIPBLASLT_CHECK(hipblasLtMatmulDescSetAttribute(
+          p.op, HIPBLASLT_MATMUL_DESC_D_SCALE_POINTER, &d_inv_scale,
+          sizeof(d_inv_scale)));
+      HIPBLASLT_CHECK(hipblasLtMatmulDescSetAttribute(
+          p.op, HIPBLASLT_MATMUL_DESC_AMAX_D_POINTER, &amax_out,
+          sizeof(amax_out)));
+    }
+    int capturing = 0;
+    hipStreamCaptureStatus st = hipStreamCaptureStatusNone;
+    if (hipStreamIsCapturing(stream, &st) == hipSuccess &&
+        st != hipStreamCaptureStatusNone)
+      capturing = 1;
+    if (!capturing) autotune(p, x8, w8, bias, y, y, workspace, ws_bytes, stream);
+    float alpha = 1.0f, beta = 0.0f;
+    HIPBLASLT_CHECK(hipblasLtMatmul(
+        handle_once(), p.op, &alpha, w8, p.a, x8, p.b, &beta, y, p.c, y, p.c,
+        &p.algo, workspace, ws_bytes, stream));
+    return 0;
+  } catch (const std::exception&) {
+    return 1;
+  }
+}
+
+int hipblaslt_fp8_linear(const void* x8, const void* w8, const void* bias,
+                         const void* w_scale, const void* x_scale, void* y,
+                         long M, long N, long K, void* workspace,
+                         size_t ws_bytes, hipStream_t stream) {
+  return run_fp8_gemm(3, x8, w8, bias, w_scale, x_scale, nullptr, nullptr, y,
+                      M, N, K, workspace, ws_bytes, stream);
+}
+
+int hipblaslt_fp8_linear_gelu_fp8out(const void* x8, const void* w8,
+                                     const void* bias, const void* w_scale,
+                                     const void* x_scale,
+                                     const void* d_inv_scale, void* amax_out,
+                                     void* y8, long M, long N, long K,
+                                     void* workspace, size_t ws_bytes,
+                                     hipStream_t stream) {
+  return run_fp8_gemm(4, x8, w8, bias, w_scale, x_scale, d_inv_scale, amax_out,
+                      y8, M, N, K, workspace, ws_bytes, stream);
 }
