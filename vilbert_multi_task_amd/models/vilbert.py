@@ -160,14 +160,14 @@ class MultiHeadSelfAttention(nn.Module):
             q, k, v = qkv[..., :h], qkv[..., h : 2 * h], qkv[..., 2 * h :]
         else:
             q, k, v = self.query(x), self.key(x), self.value(x)
+        fp8_out = None
+        if infer and getattr(self, "_fp8_ctx_site", None) is not None:
+            fp8_out = (self._fp8_ctx_obj, self._fp8_ctx_site)
         ctx, probs = F_ops.attention(
             q, k, v, self.heads,
             mask_bias, self.attn_dropout_p, self.training, need_probs,
+            fp8_out=fp8_out,
         )
-        if infer and getattr(self, "_fp8_ctx_site", None) is not None:
-            from .fp8 import attach_quant_pack
-
-            attach_quant_pack(ctx, self._fp8_ctx_obj, self._fp8_ctx_site)
         # NOTE: folding the residual into the out-proj GEMM epilogue
         # (linear_bias_residual, beta=1) intermittently faults in hipBLASLt
         # at some shapes ("write access to a read-only page", B=512 warmup)
@@ -289,14 +289,14 @@ class CrossAttention(nn.Module):
             k, v = kv[..., :h], kv[..., h:]
         else:
             k, v = self.key(x_kv), self.value(x_kv)
+        fp8_out = None
+        if infer and getattr(self, "_fp8_ctx_site", None) is not None:
+            fp8_out = (self._fp8_ctx_obj, self._fp8_ctx_site)
         ctx, probs = F_ops.attention(
             self.query(x_q), k, v, self.heads,
             mask_bias, self.attn_dropout_p, self.training, need_probs,
+            fp8_out=fp8_out,
         )
-        if infer and getattr(self, "_fp8_ctx_site", None) is not None:
-            from .fp8 import attach_quant_pack
-
-            attach_quant_pack(ctx, self._fp8_ctx_obj, self._fp8_ctx_site)
         if infer and isinstance(self.out, nn.Linear):
             o = F_ops.linear_bias(ctx, self.out.weight, self.out.bias)
         else:

@@ -32,6 +32,13 @@
 #include "common.h"
 #include <cstdlib>
 
+// e4m3 epilogue support (fp8 serving mode): emit the context tensor's
+// quantized copy during the O store (delayed scaling; read-guarded amax)
+DEV void attn_atomic_max_f32(float* addr, float v) {
+  if (v <= *reinterpret_cast<volatile float*>(addr)) return;
+  atomicMax(reinterpret_cast<unsigned int*>(addr), __float_as_uint(v));
+}
+
 #define ATTN_MAX_L 128  // max Lq/Lk this kernel serves (serving shapes <=101+pad)
 // LDS XOR swizzle: include the row's HIGH bits so sibling rows 8 apart land
 // on different 16B slots. PMC on MI355X showed the VT transpose writes were
@@ -59,12 +66,17 @@ DEV bf16x8 lds_b128(const char* p) {
 
 DEV void lds_store_b128(char* p, uint4 v) { *reinterpret_cast<uint4*>(p) = v; }
 
-template <int D, bool KGLOBAL, int NTMAX>
+template <int D, bool KGLOBAL, int NTMAX, bool FP8OUT = false>
 __global__ __launch_bounds__(256) void attn_kernel(
     const bf16* __restrict__ q, const bf16* __restrict__ kg,
     const bf16* __restrict__ vg, const bf16* __restrict__ mask,
     bf16* __restrict__ out, int B, int H, int Lq, int Lk, int mask_mode,
-    float scale, int qs, int ks, int vs, int nsplit) {
+    float scale, int qs, int ks, int vs, int nsplit,
+    unsigned char* __restrict__ out8 = nullptr,
+    const float* __restrict__ fp8_scales = nullptr,
+    float* __restrict__ fp8_amaxes = nullptr, int fp8_site = 0) {
+  const float fp8_inv = FP8OUT ? 1.0f / fp8_scales[fp8_site] : 0.f;
+  float fp8_amax = 0.f;
   // qs/ks/vs: row strides (elems) of q/k/v — [B,L,H*D] views into a fused
   // QKV (or KV) projection pass without any copy (stride 3*HD / 2*HD).
   // nsplit: query-stripe splits per (b,h) — each workgroup stages K/V and
@@ -253,12 +265,25 @@ __global__ __launch_bounds__(256) void attn_kernel(
       if (row < Lq) {
         const long obase = ((long)b * Lq + row) * HD + (long)h * D;
 #pragma unroll
-        for (int nt = 0; nt < D / 16; ++nt)
-          out[obase + nt * 16 + col0] = f2bf(acc_o[nt][r] * inv_l[r]);
+        for (int nt = 0; nt < D / 16; ++nt) {
+          const float o = acc_o[nt][r] * inv_l[r];
+          out[obase + nt * 16 + col0] = f2bf(o);
+          if (FP8OUT) {
+            fp8_amax = fmaxf(fp8_amax, fabsf(o));
+            const float c = fminf(fmaxf(o * fp8_inv, -448.f), 448.f);
+            int w = __builtin_amdgcn_cvt_pk_fp8_f32(c, c, 0, false);
+            out8[obase + nt * 16 + col0] = (unsigned char)(w & 0xff);
+          }
+        }
       }
     }
     // NO __syncthreads here: P_lds is per-wave and stripe counts differ
     // across waves (a block-wide barrier inside this loop would deadlock).
+  }
+  if (FP8OUT) {
+    fp8_amax = wave_max(fp8_amax);
+    if (lane_id() == 0 && fp8_amax > 0.f)
+      attn_atomic_max_f32(&fp8_amaxes[fp8_site], fp8_amax);
   }
 }
 
@@ -303,10 +328,12 @@ __global__ void tr16_probe_kernel(short* __restrict__ outv, int mode) {
 // launchers
 // ---------------------------------------------------------------------------
 
-void launch_attention(const bf16* q, const bf16* k, const bf16* v,
-                      const bf16* mask, bf16* out, int B, int H, int Lq, int Lk,
-                      int D, int mask_mode, int qs, int ks, int vs,
-                      hipStream_t stream) {
+void launch_attention_impl(const bf16* q, const bf16* k, const bf16* v,
+                           const bf16* mask, bf16* out, int B, int H, int Lq,
+                           int Lk, int D, int mask_mode, int qs, int ks,
+                           int vs, unsigned char* out8,
+                           const float* fp8_scales, float* fp8_amaxes,
+                           int fp8_site, hipStream_t stream) {
   const float scale = 1.0f / sqrtf((float)D);
   const int LK_PAD = (Lk + 31) & ~31;
   static const int kglobal_env = [] {
@@ -324,9 +351,17 @@ void launch_attention(const bf16* q, const bf16* k, const bf16* v,
       (size_t)((kglobal_env ? 1 : 2) * LK_PAD * D + 4 * 16 * LK_PAD);
   const dim3 grid(B * H * nsplit);
 #define LAUNCH_ATTN(DD, KG, NTM)                                              \
-  hipLaunchKernelGGL((attn_kernel<DD, KG, NTM>), grid, dim3(256), lds, stream, \
-                     q, k, v, mask, out, B, H, Lq, Lk, mask_mode, scale, qs,  \
-                     ks, vs, nsplit)
+  do {                                                                        \
+    if (out8)                                                                 \
+      hipLaunchKernelGGL((attn_kernel<DD, KG, NTM, true>), grid, dim3(256),   \
+                         lds, stream, q, k, v, mask, out, B, H, Lq, Lk,       \
+                         mask_mode, scale, qs, ks, vs, nsplit, out8,          \
+                         fp8_scales, fp8_amaxes, fp8_site);                   \
+    else                                                                      \
+      hipLaunchKernelGGL((attn_kernel<DD, KG, NTM, false>), grid, dim3(256),  \
+                         lds, stream, q, k, v, mask, out, B, H, Lq, Lk,       \
+                         mask_mode, scale, qs, ks, vs, nsplit);               \
+  } while (0)
   const bool small = LK_PAD <= 64;  // NTMAX=4 halves the accumulator VGPRs
   if (D == 64) {
     if (kglobal_env) { if (small) LAUNCH_ATTN(64, true, 4); else LAUNCH_ATTN(64, true, 8); }
@@ -336,6 +371,23 @@ void launch_attention(const bf16* q, const bf16* k, const bf16* v,
     else             { if (small) LAUNCH_ATTN(128, false, 4); else LAUNCH_ATTN(128, false, 8); }
   }
 #undef LAUNCH_ATTN
+}
+
+void launch_attention(const bf16* q, const bf16* k, const bf16* v,
+                      const bf16* mask, bf16* out, int B, int H, int Lq, int Lk,
+                      int D, int mask_mode, int qs, int ks, int vs,
+                      hipStream_t stream) {
+  launch_attention_impl(q, k, v, mask, out, B, H, Lq, Lk, D, mask_mode, qs, ks,
+                        vs, nullptr, nullptr, nullptr, 0, stream);
+}
+
+void launch_attention_fp8out(const bf16* q, const bf16* k, const bf16* v,
+                             const bf16* mask, bf16* out, unsigned char* out8,
+                             const float* scales, float* amaxes, int site,
+                             int B, int H, int Lq, int Lk, int D, int mask_mode,
+                             int qs, int ks, int vs, hipStream_t stream) {
+  launch_attention_impl(q, k, v, mask, out, B, H, Lq, Lk, D, mask_mode, qs, ks,
+                        vs, out8, scales, amaxes, site, stream);
 }
 
 void launch_mfma_probe(const bf16* a, const bf16* b, float* c, hipStream_t stream) {

@@ -26,6 +26,10 @@ void launch_embedding_ln(const long*, const long*, const long*, const T*,
                          float, hipStream_t);
 void launch_attention(const bf16*, const bf16*, const bf16*, const bf16*, bf16*,
                       int, int, int, int, int, int, int, int, int, hipStream_t);
+void launch_attention_fp8out(const bf16*, const bf16*, const bf16*, const bf16*,
+                             bf16*, unsigned char*, const float*, float*, int,
+                             int, int, int, int, int, int, int, int, int,
+                             hipStream_t);
 void launch_mfma_probe(const bf16*, const bf16*, float*, hipStream_t);
 void launch_nms_multiclass(const float*, const float*, const long*, float*, int,
                            int, float, float, hipStream_t);
@@ -149,6 +153,39 @@ at::Tensor attention(const at::Tensor& q, const at::Tensor& k,
                    (bf16*)out.data_ptr(), B, H, Lq, Lk, D, mask_mode,
                    qs, kss, vss, cur_stream());
   return out;
+}
+
+std::tuple<at::Tensor, at::Tensor> attention_fp8out(
+    const at::Tensor& q, const at::Tensor& k, const at::Tensor& v,
+    int64_t heads, const c10::optional<at::Tensor>& mask,
+    const at::Tensor& scales, const at::Tensor& amaxes, int64_t site) {
+  TORCH_CHECK(q.is_cuda() && q.scalar_type() == at::kBFloat16);
+  const int B = (int)q.size(0), Lq = (int)q.size(1), HD = (int)q.size(2);
+  const int Lk = (int)k.size(1);
+  const int H = (int)heads;
+  const int D = HD / H;
+  TORCH_CHECK(HD % H == 0 && (D == 64 || D == 128));
+  TORCH_CHECK(Lq <= 128 && Lk <= 128);
+  const int qs = row_stride_of(q, "q"), kss = row_stride_of(k, "k"),
+            vss = row_stride_of(v, "v");
+  int mask_mode = 0;
+  const bf16* mptr = nullptr;
+  if (mask.has_value() && mask->defined()) {
+    const long mn = mask->numel();
+    if (mn == (long)B * Lk) mask_mode = 1;
+    else if (mn == (long)B * Lq * Lk) mask_mode = 2;
+    else TORCH_CHECK(false, "attention_fp8out: bad mask shape");
+    mptr = (const bf16*)mask->data_ptr();
+  }
+  auto out = at::empty({q.size(0), q.size(1), q.size(2)}, q.options());
+  auto out8 = at::empty(out.sizes(), out.options().dtype(at::kFloat8_e4m3fn));
+  launch_attention_fp8out(
+      (const bf16*)q.data_ptr(), (const bf16*)k.data_ptr(),
+      (const bf16*)v.data_ptr(), mptr, (bf16*)out.data_ptr(),
+      (unsigned char*)out8.data_ptr(), scales.data_ptr<float>(),
+      amaxes.data_ptr<float>(), (int)site, B, H, Lq, Lk, D, mask_mode, qs,
+      kss, vss, cur_stream());
+  return {out, out8};
 }
 
 at::Tensor embedding_ln(const at::Tensor& ids, const at::Tensor& pos_ids,
@@ -428,6 +465,7 @@ TORCH_LIBRARY(vilbert_amd, m) {
   m.def("residual_layer_norm(Tensor x, Tensor? res, Tensor w, Tensor b, float eps) -> Tensor");
   m.def("bias_gelu(Tensor x, Tensor? bias) -> Tensor");
   m.def("attention(Tensor q, Tensor k, Tensor v, int heads, Tensor? mask) -> Tensor");
+  m.def("attention_fp8out(Tensor q, Tensor k, Tensor v, int heads, Tensor? mask, Tensor scales, Tensor(a!) amaxes, int site) -> (Tensor, Tensor)");
   m.def("embedding_ln(Tensor ids, Tensor pos, Tensor type, Tensor word_w, Tensor pos_w, Tensor type_w, Tensor ln_w, Tensor ln_b, float eps) -> Tensor");
   m.def("mfma_probe(Tensor a, Tensor b) -> Tensor");
   m.def("tr16_probe(Tensor dummy, int mode) -> Tensor");
@@ -448,6 +486,7 @@ TORCH_LIBRARY_IMPL(vilbert_amd, CUDA, m) {
   m.impl("residual_layer_norm", residual_layer_norm);
   m.impl("bias_gelu", bias_gelu);
   m.impl("attention", attention);
+  m.impl("attention_fp8out", attention_fp8out);
   m.impl("embedding_ln", embedding_ln);
   m.impl("mfma_probe", mfma_probe);
   m.impl("tr16_probe", tr16_probe);

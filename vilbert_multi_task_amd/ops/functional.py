@@ -167,6 +167,7 @@ def attention(
     dropout_p: float = 0.0,
     training: bool = False,
     need_probs: bool = False,
+    fp8_out=None,  # (Fp8Context, site): emit the e4m3 ctx pack in-kernel
 ) -> Tuple[torch.Tensor, Optional[torch.Tensor]]:
     """q: [B,Lq,H*D], k/v: [B,Lk,H*D]; mask_bias additive [B,1,1,Lk] or
     [B,1,Lq,Lk] (0 keep, large-negative masked). Returns
@@ -183,6 +184,13 @@ def attention(
             mb = mb.contiguous()
         # q/k/v may be strided views into a fused QKV projection (dim-1
         # stride 3*HD); the kernel reads strides directly — no copies.
+        if fp8_out is not None:
+            ctx8, site = fp8_out
+            out, out8 = ext.attention_fp8out(
+                q, k, v, num_heads, mb, ctx8.scales, ctx8.amaxes, site
+            )
+            out._fp8 = (out8, ctx8.scales[site])
+            return out, None
         return ext.attention(q, k, v, num_heads, mb), None
 
     b, lq, hd = q.shape
