@@ -58,7 +58,7 @@ def build_worker(args):
     device = "cuda" if torch.cuda.is_available() else "cpu"
     if device == "cuda":
         model = model.to(device=device, dtype=torch.bfloat16)
-    runner = GraphRunner(model, device=device, use_graphs=device == "cuda", serving_fast=True)
+    runner = GraphRunner(model, device=device, use_graphs=device == "cuda", serving_fast=True, fp8=args.fp8)
 
     provider = None
     if args.detector:
@@ -94,6 +94,9 @@ def main() -> None:
     ap.add_argument("--vqa-answers", default="save/VQA/cache/trainval_label2ans.pkl")
     ap.add_argument("--gqa-answers", default="save/gqa/cache/trainval_label2ans.pkl")
     ap.add_argument("--detector", action="store_true", help="full Faster R-CNN features")
+    ap.add_argument("--fp8", action="store_true",
+                    help="fp8 (e4m3) encoder GEMMs: +9-14%% throughput, "
+                    "logit cosine > 0.97 vs bf16 (docs/PERFORMANCE.md)")
     ap.add_argument("--max-batch", type=int, default=64)
     ap.add_argument("--metrics-port", type=int, default=0,
                     help="Prometheus /metrics port for the worker (0 = off)")
