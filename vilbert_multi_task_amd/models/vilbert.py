@@ -161,13 +161,24 @@ class MultiHeadSelfAttention(nn.Module):
         else:
             q, k, v = self.query(x), self.key(x), self.value(x)
         fp8_out = None
+        fp8_post = False
         if infer and getattr(self, "_fp8_ctx_site", None) is not None:
-            fp8_out = (self._fp8_ctx_obj, self._fp8_ctx_site)
+            # measured: the in-kernel e4m3 epilogue pays off for the D=128
+            # kernels; the issue-bound D=64 text kernel is faster with a
+            # standalone quantize pass (profiles/prof_fp8b)
+            if self.head_dim == 128:
+                fp8_out = (self._fp8_ctx_obj, self._fp8_ctx_site)
+            else:
+                fp8_post = True
         ctx, probs = F_ops.attention(
             q, k, v, self.heads,
             mask_bias, self.attn_dropout_p, self.training, need_probs,
             fp8_out=fp8_out,
         )
+        if fp8_post:
+            from .fp8 import attach_quant_pack
+
+            attach_quant_pack(ctx, self._fp8_ctx_obj, self._fp8_ctx_site)
         # NOTE: folding the residual into the out-proj GEMM epilogue
         # (linear_bias_residual, beta=1) intermittently faults in hipBLASLt
         # at some shapes ("write access to a read-only page", B=512 warmup)
