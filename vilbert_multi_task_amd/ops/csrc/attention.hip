@@ -110,13 +110,29 @@ __global__ __launch_bounds__(256) void attn_kernel(
     const int c = tid % KCH;
     const long kbase0 = (long)b * Lk * ks + (long)h * D;
     const long vbase0 = (long)b * Lk * vs + (long)h * D;
-    for (int r = r0; r < LK_PAD; r += rows_per_pass) {
-      uint4 kraw = {0, 0, 0, 0}, vraw = {0, 0, 0, 0};
+    // issue EVERY staging load first (up to 8 passes x K+V), then the LDS
+    // store pass: one batch of HBM round trips instead of one per pass
+    const int npass = (LK_PAD + rows_per_pass - 1) / rows_per_pass;  // <= 8
+    uint4 kr[8], vr[8];
+#pragma unroll
+    for (int pi = 0; pi < 8; ++pi) {
+      if (pi >= npass) break;
+      const int r = r0 + pi * rows_per_pass;
+      kr[pi] = make_uint4(0, 0, 0, 0);
+      vr[pi] = make_uint4(0, 0, 0, 0);
       if (r < Lk) {
         if (!KGLOBAL)
-          kraw = *reinterpret_cast<const uint4*>(kg + kbase0 + (long)r * ks + c * 8);
-        vraw = *reinterpret_cast<const uint4*>(vg + vbase0 + (long)r * vs + c * 8);
+          kr[pi] = *reinterpret_cast<const uint4*>(kg + kbase0 + (long)r * ks + c * 8);
+        vr[pi] = *reinterpret_cast<const uint4*>(vg + vbase0 + (long)r * vs + c * 8);
       }
+    }
+#pragma unroll
+    for (int pi = 0; pi < 8; ++pi) {
+      if (pi >= npass) break;
+      const int r = r0 + pi * rows_per_pass;
+      if (r >= LK_PAD) break;
+      const uint4 kraw = kr[pi];
+      const uint4 vraw = vr[pi];
       if (!KGLOBAL)
         lds_store_b128(K_lds + r * (D * 2) + ((c * 16) ^ SWZ(r)), kraw);
       // V transposed image [D][LK_PAD] (XOR-swizzled rows): the PV MFMA
