@@ -39,11 +39,11 @@ def _model(device, dtype=torch.bfloat16):
     return m, cfg
 
 
-def bench_retrieval(device: str, pairs: int = 2048, steps: int = 10, warmup: int = 3):
+def bench_retrieval(device: str, pairs: int = 2048, steps: int = 10, warmup: int = 3, fp8: bool = False):
     """Config 4: one caption scored against `pairs` candidate images — the
     co-attention MFMA saturation config. Runs as `pairs` rows per forward."""
     m, cfg = _model(device)
-    runner = GraphRunner(m, device=device, use_graphs=device.startswith("cuda"))
+    runner = GraphRunner(m, device=device, use_graphs=device.startswith("cuda"), fp8=fp8)
     batch = synthetic_batch(pairs, task_id=7, seed=1)
     for _ in range(warmup):
         runner.run(batch)
@@ -59,7 +59,7 @@ def bench_retrieval(device: str, pairs: int = 2048, steps: int = 10, warmup: int
         "unit": "pairs/s",
         "ms_per_step": round(dt * 1e3, 3),
         "config": {"pairs": pairs, "task": "retrieval (7)"},
-        "dtype": "bf16" if device.startswith("cuda") else "fp32",
+        "dtype": ("fp8-mixed" if fp8 else "bf16") if device.startswith("cuda") else "fp32",
         "data": "synthetic",
     }
 
@@ -93,7 +93,7 @@ def bench_training(device: str, batch: int = 32, steps: int = 12, warmup: int = 
     }
 
 
-def bench_mixed_serving(device: str, requests: int = 512, max_batch: int = 64):
+def bench_mixed_serving(device: str, requests: int = 512, max_batch: int = 64, fp8: bool = False):
     """Config 5: RefCOCO + NLVR2 + VQA + GQA concurrently through the
     dynamic batcher (queue -> batched hipGraph forwards -> decode)."""
     import tempfile
@@ -105,7 +105,7 @@ def bench_mixed_serving(device: str, requests: int = 512, max_batch: int = 64):
 
     m, cfg = _model(device)
     runner = GraphRunner(m, device=device, use_graphs=device.startswith("cuda"),
-                         serving_fast=True)
+                         serving_fast=True, fp8=fp8)
     with tempfile.TemporaryDirectory() as td:
         from vilbert_multi_task_amd.serve.features import SyntheticFeatureProvider
 
@@ -155,17 +155,18 @@ def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--which", default="all", choices=["retrieval", "training", "mixed", "all"])
     ap.add_argument("--pairs", type=int, default=2048)
+    ap.add_argument("--fp8", action="store_true")
     args = ap.parse_args()
     device = "cuda" if torch.cuda.is_available() else "cpu"
     if device == "cpu":
         print("WARNING: CPU — shrinking configs", file=sys.stderr)
         args.pairs = 8
     runs = {
-        "retrieval": lambda: bench_retrieval(device, args.pairs),
+        "retrieval": lambda: bench_retrieval(device, args.pairs, fp8=args.fp8),
         "training": lambda: bench_training(device, batch=4 if device == "cpu" else 32,
                                            steps=3 if device == "cpu" else 12,
                                            warmup=1 if device == "cpu" else 3),
-        "mixed": lambda: bench_mixed_serving(device, requests=20 if device == "cpu" else 512),
+        "mixed": lambda: bench_mixed_serving(device, requests=20 if device == "cpu" else 512, fp8=args.fp8),
     }
     for name, fn in runs.items():
         if args.which in (name, "all"):
