@@ -60,14 +60,18 @@ class GraphRunner:
         self.dtype = dtype
         self.use_graphs = use_graphs and device.startswith("cuda")
         self._graphs: Dict[int, Tuple[torch.cuda.CUDAGraph, dict, tuple]] = {}
+        # fp8: specific hipBLASLt fp8-epilogue algos fault ("write access to
+        # a read-only page") at the very large M of bucket 2048 — cap the
+        # bucket and serve bigger batches as chunked replays (ROADMAP.md).
+        self.max_bucket = 1024 if fp8 else 4096
 
     # -- bucket management -------------------------------------------------
     def bucket_for(self, batch: int) -> int:
-        """Smallest captured power-of-two bucket >= batch."""
+        """Smallest captured power-of-two bucket >= batch (capped)."""
         b = 1
         while b < batch:
             b <<= 1
-        return b
+        return min(b, self.max_bucket)
 
     def _static_inputs(self, bucket: int) -> dict:
         s = synthetic_batch(
@@ -133,6 +137,17 @@ class GraphRunner:
         n = batch["question"].shape[0]
         if not self.use_graphs:
             return self._forward({k: v.to(self.device) for k, v in batch.items()})
+        if n > self.max_bucket:
+            # chunked replays (even-sized chunks keep NLVR2 pair alignment)
+            outs = [
+                self.run({k: v[i : i + self.max_bucket] for k, v in batch.items()})
+                for i in range(0, n, self.max_bucket)
+            ]
+            res = []
+            for idx in range(len(outs[0])):
+                parts = [o[idx] for o in outs]
+                res.append(torch.cat(parts) if torch.is_tensor(parts[0]) else parts[0])
+            return tuple(res)
         bucket = self.bucket_for(n)
         if bucket not in self._graphs:
             self.capture(bucket)
