@@ -152,3 +152,15 @@ def test_prepare_for_serving_matches_eager(tiny_model, tiny_config):
         fused = tiny_model(*forward_args(batch))
     for i in range(9):
         assert torch.allclose(base[i], fused[i], atol=1e-5), i
+
+
+def test_dynamic_attention_flag_rejected(tiny_config):
+    import dataclasses
+
+    import pytest as _pytest
+
+    from vilbert_multi_task_amd.models.vilbert import ViLBertModel
+
+    cfg = dataclasses.replace(tiny_config, dynamic_attention=True)
+    with _pytest.raises(NotImplementedError):
+        ViLBertModel(cfg)

@@ -215,3 +215,18 @@ def test_multi_worker_competing_consumers(tmp_path, tiny_model, tiny_config):
     assert served == 8
     assert broker_a.depth() == 0
     assert len([p for s, p in push.messages if "result" in p]) == 8
+
+
+def test_guesswhat_dialog_rewrite():
+    from vilbert_multi_task_amd.serve.worker import guesswhat_dialog_rewrite
+
+    # the rewrite the reference computes (then discards) at worker.py:391-400
+    q = "q: is it a person? a: yes q: on the left? a: no"
+    assert guesswhat_dialog_rewrite(q) == (
+        "start is it a person? answer yes stop "
+        "start on the left? answer no stop"
+    )
+    # no dialog markers -> unchanged (plain questions pass through)
+    assert guesswhat_dialog_rewrite("what is this") == "what is this"
+    # unanswered trailing question
+    assert guesswhat_dialog_rewrite("q: is it red?") == "start is it red? answer  stop"

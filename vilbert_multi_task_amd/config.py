@@ -55,9 +55,16 @@ class ViLBertConfig:
 
     # ---- behavioral flags used by the reference worker ----
     task_specific_tokens: bool = True   # worker.py:516-517
-    dynamic_attention: bool = False     # worker.py:519
-    visualization: bool = False         # worker.py:522 (True in demo path)
-    predict_feature: bool = False       # worker.py:509-514
+    dynamic_attention: bool = False     # worker.py:519 — dead flag in the demo
+                                        # path; model raises if set (vilbert.py)
+    visualization: bool = False         # worker.py:522 (True in demo path).
+                                        # Advisory here: the native path returns
+                                        # attention maps whenever the caller
+                                        # passes output_all_attention_masks=True
+                                        # (a superset of the upstream gate).
+    predict_feature: bool = False       # worker.py:509-514 — demo forces False
+                                        # (1601-way masked-region classification;
+                                        # heads.py vision_prediction)
     fast_mode: bool = False
     fusion_method: str = "mul"          # pooled fusion: t_pooled * v_pooled
     in_batch_pairs: bool = False

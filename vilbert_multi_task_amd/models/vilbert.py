@@ -358,6 +358,16 @@ class ViLBertModel(nn.Module):
 
     def __init__(self, config: ViLBertConfig):
         super().__init__()
+        if getattr(config, "dynamic_attention", False):
+            # The demo path hard-codes this off (worker.py:484; the CLI flag
+            # at worker.py:519-520 is never passed) — the upstream mechanism
+            # lives in the out-of-checkout vilbert package, so there is no
+            # pinned behavior to reproduce. Fail loudly rather than silently
+            # ignore a config that would change the math.
+            raise NotImplementedError(
+                "dynamic_attention=True is not supported (dead flag in the "
+                "reference demo path; see SURVEY.md §2.2)"
+            )
         self.config = config
         c = config
         self.embeddings = TextEmbeddings(c)

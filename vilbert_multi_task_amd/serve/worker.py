@@ -45,6 +45,27 @@ from .features import SyntheticFeatureProvider, tensorize_regions
 from .push import PushClient, log_to_terminal
 
 
+def guesswhat_dialog_rewrite(query: str) -> str:
+    """Rewrite a GuessWhat dialog ``q: ..? a: ..`` into the
+    ``start <q> answer <a> stop`` token stream the model expects.
+
+    The reference builds exactly this string (worker.py:391-400) but then
+    overwrites it with the raw query (worker.py:402) — dead code / latent
+    bug flagged in SURVEY.md §2.1. We apply the rewrite for real (a
+    deliberate, documented divergence). Queries without a ``q:`` marker
+    (e.g. the frontend let a plain question through) pass unchanged, which
+    also matches the reference's effective behavior for those.
+    """
+    if "q:" not in query:
+        return query
+    parts = []
+    for dialog in query.split("q:")[1:]:
+        qa = dialog.split("a:")
+        a = qa[1] if len(qa) > 1 else ""
+        parts.append("start " + qa[0].strip() + " answer " + a.strip() + " stop")
+    return " ".join(parts)
+
+
 @dataclass
 class _Request:
     delivery: Delivery
@@ -133,14 +154,27 @@ class ServingWorker:
 
     # ------------------------------------------------------------------
     def build_batch(self, reqs: Sequence[_Request]) -> Dict[str, torch.Tensor]:
+        """Tokenize + tensorize a drained request group into one model batch.
+
+        Text follows the reference serving contract (worker.py:402-414):
+        wordpiece + [CLS]/[SEP], END-padded to 37 (the reference's comment
+        claims front-padding but its code pads at the end — we mirror the
+        code). GuessWhat dialogs get the ``q:/a:`` -> ``start .. answer ..
+        stop`` rewrite the reference computes but accidentally discards
+        (worker.py:391-402 overwrites ``tokens``) — fixed deliberately here;
+        see ``guesswhat_dialog_rewrite``.
+        """
         q_rows, mask_rows, seg_rows, task_rows = [], [], [], []
         infos_all: List[Dict] = []
         for r in reqs:
-            cached = self._tok_cache.get(r.question)
+            text = r.question
+            if r.task_id == 16:
+                text = guesswhat_dialog_rewrite(text)
+            cached = self._tok_cache.get(text)
             if cached is None:
-                cached = self.tokenizer.encode_for_serving(r.question, MAX_SEQ_LENGTH)
+                cached = self.tokenizer.encode_for_serving(text, MAX_SEQ_LENGTH)
                 if len(self._tok_cache) < 8192:
-                    self._tok_cache[r.question] = cached
+                    self._tok_cache[text] = cached
             ids, mask, seg = cached
             r.infos = self.provider.extract(r.image_paths)
             infos_all.extend(r.infos)
