@@ -102,10 +102,12 @@ Plan& get_plan(long M, long N, long K, int kind, void* workspace, size_t ws_byte
   HIPBLASLT_CHECK(hipblasLtMatmulPreferenceCreate(&pref));
   HIPBLASLT_CHECK(hipblasLtMatmulPreferenceSetAttribute(
       pref, HIPBLASLT_MATMUL_PREF_MAX_WORKSPACE_BYTES, &ws_bytes, sizeof(ws_bytes)));
-  hipblasLtMatmulHeuristicResult_t results[8];
+  // 32 heuristic candidates: autotune() times them once at eager warmup
+  // (pre-capture), so a wider search costs warmup-only milliseconds.
+  hipblasLtMatmulHeuristicResult_t results[32];
   int found = 0;
   HIPBLASLT_CHECK(hipblasLtMatmulAlgoGetHeuristic(
-      handle_once(), p.op, p.a, p.b, p.c, p.c, pref, 8, results, &found));
+      handle_once(), p.op, p.a, p.b, p.c, p.c, pref, 32, results, &found));
   hipblasLtMatmulPreferenceDestroy(pref);
   if (found > 0) {
     p.algo = results[0].algo;
