@@ -39,7 +39,7 @@ class SimpleClassifier(nn.Module):
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         h = F_ops.linear_bias_gelu(x, self.dense.weight, self.dense.bias)
         h = self.layer_norm(h)
-        return self.decoder(h)
+        return F_ops.linear_bias(h, self.decoder.weight, self.decoder.bias)
 
 
 class LMHead(nn.Module):
@@ -55,7 +55,8 @@ class LMHead(nn.Module):
     def forward(self, t: torch.Tensor) -> torch.Tensor:
         h = F_ops.linear_bias_gelu(t, self.transform.weight, self.transform.bias)
         h = self.layer_norm(h)
-        return torch.nn.functional.linear(h, self.decoder_weight, self.decoder_bias)
+        # 30522-way vocab GEMM — the single biggest head GEMM; autotuned path
+        return F_ops.linear_bias(h, self.decoder_weight, self.decoder_bias)
 
 
 class VILBertForVLTasks(nn.Module):
@@ -125,7 +126,10 @@ class VILBertForVLTasks(nn.Module):
             linguisic_prediction = t.new_zeros(b, t.shape[1], 0)
             linguisic_logit = t.new_zeros(b, t.shape[1], 0)
         else:
-            vision_prediction = self.vision_prediction(self.dropout(v))
+            vp_in = self.dropout(v)
+            vision_prediction = F_ops.linear_bias(
+                vp_in, self.vision_prediction.weight, self.vision_prediction.bias
+            )
             linguisic_prediction = self.linguistic_prediction(self.dropout(t))
             linguisic_logit = self.linguistic_logit(self.dropout(t))
 
