@@ -37,7 +37,7 @@ def main() -> None:
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--steps", type=int, default=32)
     ap.add_argument("--warmup", type=int, default=8)
-    ap.add_argument("--batch", type=int, default=512, help="per-GPU queries per step")
+    ap.add_argument("--batch", type=int, default=1024, help="per-GPU queries per step")
     ap.add_argument("--no-graphs", action="store_true")
     ap.add_argument("--eager", action="store_true", help="force eager torch ops (debug)")
     ap.add_argument(
