@@ -54,7 +54,7 @@ def main() -> None:
         cfg = ViLBertConfig.tiny()
         args.seq_len, args.regions = 20, 36
     elif args.config:
-        cfg = ViLBertConfig.from_json_file(args.config)
+        cfg = ViLBertConfig.from_file(args.config)
     else:
         cfg = ViLBertConfig.base_12in1()
 

@@ -59,3 +59,27 @@ def test_config_ignores_unknown_keys(tmp_path):
 def test_config_validation():
     with pytest.raises(ValueError):
         ViLBertConfig(hidden_size=100, num_attention_heads=7)
+
+
+def test_tasks_yaml_round_trip():
+    """configs/vilbert_tasks.yml (the reference's worker.py:496 registry
+    file, rebuilt) must load to exactly the built-in TASKS registry."""
+    from vilbert_multi_task_amd.tasks import TASKS, load_tasks_yaml
+
+    reg = load_tasks_yaml("configs/vilbert_tasks.yml")
+    assert reg == TASKS
+
+
+def test_config_from_yaml(tmp_path):
+    import json
+
+    import yaml
+
+    from vilbert_multi_task_amd.config import ViLBertConfig
+
+    raw = json.load(open("configs/bert_base_6layer_6conect.json"))
+    p = tmp_path / "cfg.yaml"
+    p.write_text(yaml.safe_dump(raw))
+    assert ViLBertConfig.from_file(str(p)) == ViLBertConfig.from_json_file(
+        "configs/bert_base_6layer_6conect.json"
+    )

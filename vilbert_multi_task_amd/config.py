@@ -92,6 +92,17 @@ class ViLBertConfig:
         return cls.from_dict(raw)
 
     @classmethod
+    def from_file(cls, path: str) -> "ViLBertConfig":
+        """Load a model config from .json or .yaml/.yml (one
+        typed config format replacing the reference's four ad-hoc layers)."""
+        if str(path).endswith((".yaml", ".yml")):
+            import yaml
+
+            with open(path) as f:
+                return cls.from_dict(yaml.safe_load(f) or {})
+        return cls.from_json_file(path)
+
+    @classmethod
     def from_dict(cls, raw: dict) -> "ViLBertConfig":
         known = {f.name for f in dataclasses.fields(cls)}
         # accept upstream key aliases

@@ -46,7 +46,7 @@ def build_worker(args):
     from .worker import ServingWorker
 
     cfg = (
-        ViLBertConfig.from_json_file(args.config)
+        ViLBertConfig.from_file(args.config)
         if args.config
         else ViLBertConfig.base_12in1()
     )

@@ -109,3 +109,34 @@ def validate_request(task_id: int, num_images: int) -> Optional[str]:
             f"Task {spec.name} expects {want} image(s), got {num_images}"
         )
     return None
+
+def load_tasks_yaml(path: str) -> Dict[int, TaskSpec]:
+    """Load a task registry from a ``vilbert_tasks.yml``-style file
+    (upstream shape: top-level ``TASK<N>`` sections — worker.py:496-497
+    loads the same file into an EasyDict). Unknown per-task keys are
+    ignored so upstream training fields (dataroot, lr, ...) pass through.
+    Returns a registry dict; the built-in ``TASKS`` stays the default."""
+    import yaml
+
+    with open(path) as f:
+        raw = yaml.safe_load(f) or {}
+    fields = {"task_id", "name", "decode", "min_images", "max_images", "reachable", "dataset"}
+    reg: Dict[int, TaskSpec] = {}
+    for key, spec in raw.items():
+        if not str(key).upper().startswith("TASK") or not isinstance(spec, dict):
+            continue
+        tid = int(spec.get("task_id", str(key)[4:]))
+        base = TASKS.get(tid)
+        kw = {k: v for k, v in spec.items() if k in fields}
+        kw["task_id"] = tid
+        if "decode" in kw:
+            kw["decode"] = DecodeFamily(kw["decode"])
+        elif base is not None:
+            kw["decode"] = base.decode
+        else:
+            raise ValueError(f"task {tid}: 'decode' family required for new tasks")
+        if base is not None:  # fill unspecified fields from the built-in spec
+            for k in fields - set(kw):
+                kw[k] = getattr(base, k)
+        reg[tid] = TaskSpec(**{k: kw[k] for k in fields if k in kw})
+    return reg
