@@ -191,3 +191,50 @@ def test_grad_accumulation_matches_large_batch():
     assert ds_a == ds_b
     assert la == la and lb == lb
     assert a.opt.param_groups[0]["lr"] == b.opt.param_groups[0]["lr"]
+
+# ---------------------------------------------------------------------------
+# Failure detection: a dropped rank must surface as a loud
+# error on the surviving rank within the collective timeout — not a silent
+# hang. Recovery path = restart the job and resume from the last checkpoint
+# (test_checkpoint_roundtrip covers the resume half).
+# ---------------------------------------------------------------------------
+
+def _dropped_rank_worker(rank, world, tmpdir):
+    import datetime
+
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = "29573"
+    dist.init_process_group(
+        "gloo", rank=rank, world_size=world,
+        timeout=datetime.timedelta(seconds=10),
+    )
+    if rank == 1:
+        # simulated crash after rendezvous, before the first all-reduce
+        os._exit(0)
+    try:
+        tr = _tiny_trainer(rank=rank, world_size=world, bucket_bytes=1 << 18)
+        tr.train_step()  # all-reduce against a dead peer -> must raise
+        outcome = "no-error"
+    except Exception as e:
+        outcome = f"raised:{type(e).__name__}"
+    with open(os.path.join(tmpdir, "outcome.txt"), "w") as f:
+        f.write(outcome)
+
+
+@pytest.mark.timeout(240)
+def test_dropped_rank_detected(tmp_path):
+    ctx = mp.get_context("spawn")
+    procs = [
+        ctx.Process(target=_dropped_rank_worker, args=(r, 2, str(tmp_path)))
+        for r in range(2)
+    ]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(120)
+        if p.is_alive():  # never hang the suite
+            p.terminate()
+            p.join(10)
+            pytest.fail("surviving rank hung instead of detecting the dead peer")
+    outcome = (tmp_path / "outcome.txt").read_text()
+    assert outcome.startswith("raised:"), outcome
