@@ -15,6 +15,8 @@
 
 #include <hip/hip_runtime.h>
 #include <hipblaslt/hipblaslt.h>
+#include <hipblaslt/hipblaslt-ext.hpp>
+#include <cstdlib>
 
 #include <map>
 #include <vector>
@@ -114,6 +116,34 @@ Plan& get_plan(long M, long N, long K, int kind, void* workspace, size_t ws_byte
     p.has_algo = true;
     p.beta = kind == 1 ? 1.0f : 0.0f;
     p.candidates.assign(results, results + found);
+    // Full-catalogue autotune (default ON; VILBERT_GEMM_TUNE_FULL=0 keeps
+    // just the 32-candidate heuristic shortlist): sweep every library algo
+    // supported for this problem. Warmup-only cost — autotune() times each
+    // candidate pre-capture — measured +1.1% steady-state @B1024 over the
+    // heuristic shortlist.
+    static const bool tune_full = [] {
+      const char* e = std::getenv("VILBERT_GEMM_TUNE_FULL");
+      return !(e && e[0] == '0');
+    }();
+    if (tune_full) {
+      std::vector<hipblasLtMatmulHeuristicResult_t> all;
+      if (hipblaslt_ext::getAllAlgos(
+              handle_once(), hipblaslt_ext::GemmType::HIPBLASLT_GEMM,
+              HIPBLAS_OP_T, HIPBLAS_OP_N, ab_t, ab_t, d_t, d_t,
+              HIPBLAS_COMPUTE_32F, all) == HIPBLAS_STATUS_SUCCESS) {
+        float alpha = 1.0f, beta = p.beta;
+        for (auto& c : all) {
+          size_t need = 0;
+          if (hipblaslt_ext::matmulIsAlgoSupported(
+                  handle_once(), p.op, &alpha, p.a, p.b, &beta, p.c, p.c,
+                  c.algo, need) == HIPBLAS_STATUS_SUCCESS &&
+              need <= ws_bytes) {
+            p.candidates.push_back(c);
+            if (p.candidates.size() >= 400) break;
+          }
+        }
+      }
+    }
   }
   auto r = cache.emplace(key, p);
   return r.first->second;
