@@ -10,6 +10,8 @@ import math
 import pytest
 import torch
 
+from vilbert_multi_task_amd.ops import functional as F_ops
+
 pytestmark = pytest.mark.gpu
 
 
@@ -258,3 +260,86 @@ def test_linear_bias_residual_fused(ext):
     ref = x.float() @ w.float().T + b.float() + res.float()
     err = (y.float() - ref).abs() - ref.abs() / 64
     assert err.max() < 3e-2, err.max().item()
+
+
+@pytest.mark.gpu
+def test_ln_train_bwd_matches_fp32_autograd():
+    """train_bwd.hip LN fwd+bwd vs a full fp32 torch.autograd reference."""
+    torch.manual_seed(0)
+    for rows, dim, has_res in [(300, 768, True), (300, 768, False), (257, 1024, True), (65, 52, True)]:
+        xf = torch.randn(rows, dim, device="cuda")
+        rf = torch.randn(rows, dim, device="cuda") if has_res else None
+        wf = torch.randn(dim, device="cuda")
+        bf = torch.randn(dim, device="cuda")
+        gy = torch.randn(rows, dim, device="cuda")
+
+        x32 = xf.clone().requires_grad_()
+        r32 = rf.clone().requires_grad_() if has_res else None
+        w32 = wf.clone().requires_grad_()
+        b32 = bf.clone().requires_grad_()
+        h = x32 + r32 if has_res else x32
+        torch.nn.functional.layer_norm(h, (dim,), w32, b32, 1e-12).backward(gy)
+
+        x = xf.bfloat16().requires_grad_()
+        r = rf.bfloat16().requires_grad_() if has_res else None
+        w = wf.bfloat16().requires_grad_()
+        b = bf.bfloat16().requires_grad_()
+        y = F_ops.layer_norm(x, w, b, 1e-12, residual=r)
+        assert "LayerNormTrain" in type(y.grad_fn).__name__, type(y.grad_fn).__name__
+        y.backward(gy.bfloat16())
+
+        for got, ref, name in [
+            (x.grad, x32.grad, "gx"),
+            (w.grad, w32.grad, "gw"),
+            (b.grad, b32.grad, "gb"),
+        ] + ([(r.grad, r32.grad, "gres")] if has_res else []):
+            got = got.float()
+            tol = 0.05 * ref.abs().max().clamp(min=1.0)
+            assert (got - ref).abs().max() <= tol, (rows, dim, has_res, name, (got - ref).abs().max())
+
+
+@pytest.mark.gpu
+def test_linear_bias_gelu_train_bwd_matches_fp32_autograd():
+    torch.manual_seed(1)
+    M, K, N = 300, 256, 512
+    xf = torch.randn(M, K, device="cuda")
+    wf = torch.randn(N, K, device="cuda") * 0.05
+    bf = torch.randn(N, device="cuda") * 0.1
+    gy = torch.randn(M, N, device="cuda")
+
+    x32 = xf.clone().requires_grad_()
+    w32 = wf.clone().requires_grad_()
+    b32 = bf.clone().requires_grad_()
+    torch.nn.functional.gelu(
+        torch.nn.functional.linear(x32, w32, b32)
+    ).backward(gy)
+
+    x = xf.bfloat16().requires_grad_()
+    w = wf.bfloat16().requires_grad_()
+    b = bf.bfloat16().requires_grad_()
+    y = F_ops.linear_bias_gelu(x, w, b)
+    assert "LinearBiasGelu" in type(y.grad_fn).__name__, type(y.grad_fn).__name__
+    y.backward(gy.bfloat16())
+
+    for got, ref, name in [(x.grad, x32.grad, "gx"), (w.grad, w32.grad, "gw"), (b.grad, b32.grad, "gb")]:
+        got = got.float()
+        tol = 0.06 * ref.abs().max().clamp(min=1.0)
+        assert (got - ref).abs().max() <= tol, (name, (got - ref).abs().max(), ref.abs().max())
+
+
+@pytest.mark.gpu
+def test_ln_train_bwd_deterministic():
+    """fixed-chunk partials, no atomics: param grads bitwise-stable."""
+    torch.manual_seed(2)
+    x = torch.randn(5000, 768, device="cuda").bfloat16()
+    w = torch.randn(768, device="cuda").bfloat16().requires_grad_()
+    b = torch.randn(768, device="cuda").bfloat16().requires_grad_()
+    gy = torch.randn(5000, 768, device="cuda").bfloat16()
+    grads = []
+    for _ in range(2):
+        wg = w.detach().clone().requires_grad_()
+        bg = b.detach().clone().requires_grad_()
+        F_ops.layer_norm(x, wg, bg, 1e-12).backward(gy)
+        grads.append((wg.grad.clone(), bg.grad.clone()))
+    assert torch.equal(grads[0][0], grads[1][0])
+    assert torch.equal(grads[0][1], grads[1][1])

@@ -60,6 +60,15 @@ int hipblaslt_fp8_linear_gelu_fp8out(const void*, const void*, const void*,
 template <typename T>
 void launch_roi_align(const T*, const float*, T*, int, int, int, int, int, int,
                       int, float, int, hipStream_t);
+void launch_ln_fwd_train(const bf16*, const bf16*, const bf16*, const bf16*,
+                         bf16*, bf16*, float*, float*, long, int, float,
+                         hipStream_t);
+void launch_ln_bwd_input(const bf16*, const bf16*, const float*, const float*,
+                         const bf16*, bf16*, long, int, hipStream_t);
+void launch_ln_bwd_param(const bf16*, const bf16*, const float*, const float*,
+                         float*, float*, long, int, int, int, hipStream_t);
+void launch_bias_gelu_bwd(const bf16*, const bf16*, bf16*, float*, long, int,
+                          int, int, hipStream_t);
 
 namespace {
 
@@ -459,6 +468,79 @@ at::Tensor nms_multiclass(const at::Tensor& boxes, const at::Tensor& scores,
   return out;
 }
 
+// ---- training-path fused LN / GELU backward (train_bwd.hip) ---------------
+
+constexpr int kBwdChunks = 128;  // fixed partial-chunk count => deterministic
+
+std::tuple<at::Tensor, at::Tensor, at::Tensor, at::Tensor> ln_fwd_train(
+    const at::Tensor& x, const c10::optional<at::Tensor>& res,
+    const at::Tensor& w, const at::Tensor& b, double eps) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous() &&
+                  x.scalar_type() == at::kBFloat16,
+              "ln_fwd_train: x bf16 contiguous");
+  const int dim = (int)x.size(-1);
+  const long rows = x.numel() / dim;
+  auto y = at::empty_like(x);
+  const bool hr = res.has_value();
+  if (hr) TORCH_CHECK(res->is_contiguous() && res->sizes() == x.sizes());
+  // xs = x + res (saved for backward); without a residual x itself is xs
+  auto xs = hr ? at::empty_like(x) : x;
+  auto f32 = x.options().dtype(at::kFloat);
+  auto mean = at::empty({rows}, f32);
+  auto rstd = at::empty({rows}, f32);
+  launch_ln_fwd_train((const bf16*)x.data_ptr(),
+                      hr ? (const bf16*)res->data_ptr() : nullptr,
+                      (const bf16*)w.data_ptr(), (const bf16*)b.data_ptr(),
+                      (bf16*)y.data_ptr(), (bf16*)xs.data_ptr(),
+                      mean.data_ptr<float>(), rstd.data_ptr<float>(), rows,
+                      dim, (float)eps, cur_stream());
+  return {y, xs, mean, rstd};
+}
+
+std::tuple<at::Tensor, at::Tensor, at::Tensor> ln_bwd(
+    const at::Tensor& gy, const at::Tensor& xs, const at::Tensor& mean,
+    const at::Tensor& rstd, const at::Tensor& w) {
+  TORCH_CHECK(gy.is_cuda() && gy.is_contiguous() &&
+                  gy.scalar_type() == at::kBFloat16,
+              "ln_bwd: grad bf16 contiguous");
+  const int dim = (int)gy.size(-1);
+  const long rows = gy.numel() / dim;
+  auto gxs = at::empty_like(gy);
+  launch_ln_bwd_input((const bf16*)gy.data_ptr(), (const bf16*)xs.data_ptr(),
+                      mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                      (const bf16*)w.data_ptr(), (bf16*)gxs.data_ptr(), rows,
+                      dim, cur_stream());
+  const int nchunks = (int)std::min<long>(kBwdChunks, rows);
+  const int rpc = (int)((rows + nchunks - 1) / nchunks);
+  auto f32 = gy.options().dtype(at::kFloat);
+  auto gw_part = at::empty({nchunks, dim}, f32);
+  auto gb_part = at::empty({nchunks, dim}, f32);
+  launch_ln_bwd_param((const bf16*)gy.data_ptr(), (const bf16*)xs.data_ptr(),
+                      mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                      gw_part.data_ptr<float>(), gb_part.data_ptr<float>(),
+                      rows, dim, nchunks, rpc, cur_stream());
+  return {gxs, gw_part.sum(0).to(w.scalar_type()),
+          gb_part.sum(0).to(w.scalar_type())};
+}
+
+std::tuple<at::Tensor, at::Tensor> bias_gelu_bwd(const at::Tensor& gy,
+                                                 const at::Tensor& pre) {
+  TORCH_CHECK(gy.is_cuda() && gy.is_contiguous() &&
+                  gy.scalar_type() == at::kBFloat16 &&
+                  pre.is_contiguous() && pre.sizes() == gy.sizes(),
+              "bias_gelu_bwd: bf16 contiguous");
+  const int dim = (int)gy.size(-1);
+  const long rows = gy.numel() / dim;
+  auto gpre = at::empty_like(gy);
+  const int nchunks = (int)std::min<long>(kBwdChunks, rows);
+  const int rpc = (int)((rows + nchunks - 1) / nchunks);
+  auto gb_part = at::empty({nchunks, dim}, gy.options().dtype(at::kFloat));
+  launch_bias_gelu_bwd((const bf16*)gy.data_ptr(), (const bf16*)pre.data_ptr(),
+                       (bf16*)gpre.data_ptr(), gb_part.data_ptr<float>(), rows,
+                       dim, nchunks, rpc, cur_stream());
+  return {gpre, gb_part.sum(0).to(gy.scalar_type())};
+}
+
 }  // namespace
 
 TORCH_LIBRARY(vilbert_amd, m) {
@@ -480,6 +562,9 @@ TORCH_LIBRARY(vilbert_amd, m) {
   m.def("update_fp8_scales(Tensor(a!) scales, Tensor(b!) inv_scales, Tensor(c!) amaxes) -> ()");
   m.def("fp8_linear(Tensor x8, Tensor w8, Tensor bias, Tensor w_scale, Tensor x_scale) -> Tensor");
   m.def("fp8_linear_gelu_fp8out(Tensor x8, Tensor w8, Tensor bias, Tensor w_scale, Tensor x_scale, Tensor d_inv_scale, Tensor(a!) amax_out) -> Tensor");
+  m.def("ln_fwd_train(Tensor x, Tensor? res, Tensor w, Tensor b, float eps) -> (Tensor, Tensor, Tensor, Tensor)");
+  m.def("ln_bwd(Tensor gy, Tensor xs, Tensor mean, Tensor rstd, Tensor w) -> (Tensor, Tensor, Tensor)");
+  m.def("bias_gelu_bwd(Tensor gy, Tensor pre) -> (Tensor, Tensor)");
 }
 
 TORCH_LIBRARY_IMPL(vilbert_amd, CUDA, m) {
@@ -501,4 +586,7 @@ TORCH_LIBRARY_IMPL(vilbert_amd, CUDA, m) {
   m.impl("update_fp8_scales", update_fp8_scales);
   m.impl("fp8_linear", fp8_linear);
   m.impl("fp8_linear_gelu_fp8out", fp8_linear_gelu_fp8out);
+  m.impl("ln_fwd_train", ln_fwd_train);
+  m.impl("ln_bwd", ln_bwd);
+  m.impl("bias_gelu_bwd", bias_gelu_bwd);
 }

@@ -8,8 +8,10 @@ Rules of the build (BASELINE.json north star):
   .so the GPU tests actually loaded).
 - On CPU the plain fp32 PyTorch implementations below are the numerics
   oracle each HIP kernel is tested against (SURVEY.md §4 consequence (1)).
-- Autograd paths (training) use the PyTorch ops; the HIP kernels are
-  forward/serving kernels (hipGraph-capturable, no sync, no alloc churn).
+- Training: LayerNorm and bias+GELU run fused HIP forward/backward kernels
+  through torch.autograd.Function wrappers (train_bwd.hip — deterministic
+  param grads, no atomics); everything else uses autograd-capable torch
+  ops. VILBERT_AMD_EAGER_BWD=1 opts training back to plain autograd.
 
 Set VILBERT_AMD_FORCE_EAGER=1 to force the PyTorch path on GPU (debug only).
 """
@@ -64,6 +66,67 @@ def _want_hip(*tensors: torch.Tensor) -> bool:
     return True
 
 
+def _want_hip_train(*tensors: Optional[torch.Tensor]) -> bool:
+    """True iff the fused-backward HIP training path should run: bf16 on
+    GPU, grad enabled, extension present. VILBERT_AMD_EAGER_BWD=1 opts out
+    (falls back to plain autograd torch ops)."""
+    x = tensors[0]
+    if not x.is_cuda or x.dtype != torch.bfloat16:
+        return False
+    if os.environ.get("VILBERT_AMD_FORCE_EAGER") == "1":
+        return False
+    if os.environ.get("VILBERT_AMD_EAGER_BWD") == "1":
+        return False
+    if not (torch.is_grad_enabled() and any(t is not None and t.requires_grad for t in tensors)):
+        return False
+    return _load_extension() is not None
+
+
+class _LayerNormTrainFn(torch.autograd.Function):
+    """Fused LN(x (+res)) with HIP forward-with-stats and 2-kernel backward
+    (train_bwd.hip). Deterministic param grads (fixed-chunk partials, no
+    atomics). The residual grad is the input grad (d(x+res)/dx = d/dres)."""
+
+    @staticmethod
+    def forward(ctx, x, weight, bias, residual, eps):
+        ext = _load_extension()
+        y, xs, mean, rstd = ext.ln_fwd_train(x, residual, weight, bias, eps)
+        ctx.save_for_backward(xs, mean, rstd, weight)
+        ctx.has_res = residual is not None
+        return y
+
+    @staticmethod
+    def backward(ctx, gy):
+        xs, mean, rstd, weight = ctx.saved_tensors
+        ext = _load_extension()
+        gxs, gw, gb = ext.ln_bwd(gy.contiguous(), xs, mean, rstd, weight)
+        return gxs, gw, gb, (gxs if ctx.has_res else None), None
+
+
+class _LinearBiasGeluFn(torch.autograd.Function):
+    """gelu(x @ w.T + bias) with a fused single-pass backward: grad_pre and
+    grad_bias come from one HIP kernel (vs torch's gelu_backward + separate
+    sum-over-rows reduce); the GEMM grads stay on hipBLASLt."""
+
+    @staticmethod
+    def forward(ctx, x, w, bias):
+        pre = torch.nn.functional.linear(x, w, bias).contiguous()
+        ext = _load_extension()
+        y = ext.bias_gelu(pre, None)
+        ctx.save_for_backward(x, w, pre)
+        return y
+
+    @staticmethod
+    def backward(ctx, gy):
+        x, w, pre = ctx.saved_tensors
+        ext = _load_extension()
+        gpre, gb = ext.bias_gelu_bwd(gy.contiguous(), pre)
+        g2 = gpre.reshape(-1, gpre.shape[-1])
+        gx = (g2 @ w).reshape(x.shape)
+        gw = g2.t() @ x.reshape(-1, x.shape[-1])
+        return gx, gw, gb
+
+
 # --------------------------------------------------------------------------
 # LayerNorm (+ optional fused residual add)
 # --------------------------------------------------------------------------
@@ -79,6 +142,10 @@ def layer_norm(
     if _want_hip(x, residual):
         ext = _load_extension()
         return ext.residual_layer_norm(x.contiguous(), _opt(residual), weight, bias, eps)
+    if _want_hip_train(x, weight, bias, residual):
+        return _LayerNormTrainFn.apply(
+            x.contiguous(), weight, bias, _opt(residual), eps
+        )
     if residual is not None:
         x = x + residual
     return torch.nn.functional.layer_norm(x, (x.shape[-1],), weight, bias, eps)
@@ -115,6 +182,8 @@ def linear_bias_gelu(x: torch.Tensor, w: torch.Tensor, bias: torch.Tensor) -> to
             return ext.linear_bias_gelu(x, w, bias)
         except RuntimeError:
             _LINEAR_GELU_OK = False  # no algo for this arch/shape: fall back
+    if _want_hip_train(x, w, bias):
+        return _LinearBiasGeluFn.apply(x, w, bias)
     h = torch.nn.functional.linear(x, w)
     return bias_gelu(h, bias)
 
