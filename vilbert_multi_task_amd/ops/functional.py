@@ -139,13 +139,16 @@ def layer_norm(
     residual: Optional[torch.Tensor] = None,
 ) -> torch.Tensor:
     """y = LayerNorm(x + residual) over the last dim."""
-    if _want_hip(x, residual):
-        ext = _load_extension()
-        return ext.residual_layer_norm(x.contiguous(), _opt(residual), weight, bias, eps)
+    # train check first: _want_hip only inspects (x, residual), so a call
+    # where just weight/bias require grad would otherwise take the grad-less
+    # inference kernel.
     if _want_hip_train(x, weight, bias, residual):
         return _LayerNormTrainFn.apply(
             x.contiguous(), weight, bias, _opt(residual), eps
         )
+    if _want_hip(x, residual):
+        ext = _load_extension()
+        return ext.residual_layer_norm(x.contiguous(), _opt(residual), weight, bias, eps)
     if residual is not None:
         x = x + residual
     return torch.nn.functional.layer_norm(x, (x.shape[-1],), weight, bias, eps)
