@@ -119,7 +119,9 @@ def decode_grounding(
     pixel coords (worker.py:369-386). Returns boxes; the worker renders them
     into result images (worker.py:591-600 equivalent)."""
     scores = outputs[6][row, :, 0].float()
-    probs = torch.sigmoid(scores)
+    # softmax across the 101 regions (worker.py:374) — NOT sigmoid: the
+    # reference normalizes region confidence over the whole image.
+    probs = torch.softmax(scores, dim=0)
     conf, idx = probs.topk(min(k, scores.shape[0]))
     boxes = []
     for i in idx.tolist():
