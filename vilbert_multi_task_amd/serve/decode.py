@@ -44,7 +44,11 @@ class AnswerVocab:
 
 
 _BINARY_ANSWERS = ["False", "True"]  # worker.py:325-338 (index 1 = True)
-_TRI_ANSWERS = ["contradiction", "neutral", "entailment"]  # worker.py:340-354
+_TRI_ANSWERS = [  # exact reference label strings (worker.py:342)
+    "contradiction (false)",
+    "neutral",
+    "entailment (true)",
+]
 
 
 def _topk_answers(logits: torch.Tensor, vocab: AnswerVocab, k: int = 3) -> List[Dict[str, Any]]:
