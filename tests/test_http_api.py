@@ -98,3 +98,11 @@ def test_admin_surface(client):
     assert len(body["tasks"]) == 9  # the 9 registry task ids
     assert body["queue"]["ready"] >= 1
     assert body["recent_questions"] == []  # worker inserts the row, not the app
+
+
+def test_media_path_traversal_blocked(client):
+    c, _ = client
+    r = c.get("/media/../../etc/passwd")
+    assert r.status_code == 404
+    r = c.get("/media/%2e%2e/%2e%2e/etc/passwd")
+    assert r.status_code == 404

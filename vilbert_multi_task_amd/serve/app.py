@@ -141,7 +141,9 @@ def create_app(
         from fastapi.responses import FileResponse
 
         full = os.path.normpath(os.path.join(media_root, path))
-        if not full.startswith(os.path.normpath(media_root)) or not os.path.isfile(full):
+        root = os.path.normpath(media_root)
+        # commonpath (not startswith): "/mediaX" must not pass for "/media"
+        if os.path.commonpath([full, root]) != root or not os.path.isfile(full):
             return JSONResponse({"error": "not found"}, status_code=404)
         return FileResponse(full)
 
