@@ -230,3 +230,26 @@ def test_guesswhat_dialog_rewrite():
     assert guesswhat_dialog_rewrite("what is this") == "what is this"
     # unanswered trailing question
     assert guesswhat_dialog_rewrite("q: is it red?") == "start is it red? answer  stop"
+
+
+def test_tensorize_regions_mixed_box_counts():
+    """Slow path: images with fewer boxes are zero-padded and masked, and
+    assembly stays on the features' device (detector output case)."""
+    from vilbert_multi_task_amd.serve.features import tensorize_regions
+
+    def info(nb):
+        return {
+            "features": torch.randn(nb, 64).abs(),
+            "bbox": torch.tensor([[0.0, 0.0, 320.0, 240.0]] * nb),
+            "image_width": 640.0,
+            "image_height": 480.0,
+            "num_boxes": nb,
+        }
+
+    out = tensorize_regions([info(10), info(5)], num_regions=12)
+    assert out["features"].shape == (2, 12, 64)
+    assert out["image_mask"][0].sum() == 11 and out["image_mask"][1].sum() == 6
+    assert torch.all(out["features"][1, 7:] == 0)
+    # global box row
+    assert torch.allclose(out["spatials"][0, 0], torch.tensor([0.0, 0, 1, 1, 1]))
+    assert torch.allclose(out["spatials"][1, 1, :4], torch.tensor([0.0, 0.0, 0.5, 0.5]))

@@ -126,20 +126,21 @@ def tensorize_regions(
             "spatials": spatials,
             "image_mask": torch.ones(n, num_regions, dtype=torch.long),
         }
-    features = torch.zeros(n, num_regions, feat_dim)
-    spatials = torch.zeros(n, num_regions, 5)
+    dev = infos[0]["features"].device  # keep assembly on the features' device
+    features = torch.zeros(n, num_regions, feat_dim, device=dev)
+    spatials = torch.zeros(n, num_regions, 5, device=dev)
     image_mask = torch.zeros(n, num_regions, dtype=torch.long)
     for i, info in enumerate(infos):
-        f = info["features"].float()
+        f = info["features"].to(dev).float()
         num = min(int(info["num_boxes"]), num_regions - 1)
         w, h = float(info["image_width"]), float(info["image_height"])
         features[i, 0] = f[:num].mean(dim=0)  # global mean-pooled feature
         features[i, 1 : num + 1] = f[:num]
-        bbox = info["bbox"].float()[:num]
+        bbox = info["bbox"].to(dev).float()[:num]
         x1, y1 = bbox[:, 0] / w, bbox[:, 1] / h
         x2, y2 = bbox[:, 2] / w, bbox[:, 3] / h
         area = (x2 - x1) * (y2 - y1)
-        spatials[i, 0] = torch.tensor([0.0, 0.0, 1.0, 1.0, 1.0])  # worker.py:443
+        spatials[i, 0] = torch.tensor([0.0, 0.0, 1.0, 1.0, 1.0], device=dev)  # worker.py:443
         spatials[i, 1 : num + 1] = torch.stack([x1, y1, x2, y2, area], dim=1)
         image_mask[i, : num + 1] = 1
     return {"features": features, "spatials": spatials, "image_mask": image_mask}
