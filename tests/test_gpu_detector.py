@@ -76,3 +76,33 @@ def test_serving_worker_gpu_end_to_end(tmp_path):
     vqa = next(r for r in results if r["task_id"] == 1)
     assert len(vqa["result"]) == 3
     assert all(0 <= e["confidence"] <= 1 for e in vqa["result"])
+
+
+def test_x152_detector_bf16_matches_fp32():
+    """bf16 serving mode of the full X-152 stack: same kept boxes (box math
+    stays fp32), fc6 features cosine-close to the fp32 run."""
+    from vilbert_multi_task_amd.detector import DetectionModel, DetectorConfig
+    from vilbert_multi_task_amd.detector.extractor import DetectorFeatureProvider
+
+    torch.manual_seed(0)
+    m32 = DetectionModel(DetectorConfig.x152()).eval()
+    p32 = DetectorFeatureProvider(m32, device="cuda")
+    import copy
+
+    mbf = copy.deepcopy(m32)
+    pbf = DetectorFeatureProvider(mbf, device="cuda", dtype="bfloat16")
+
+    img = (torch.rand(3, 480, 640) * 255).to(torch.uint8)
+    import tempfile
+
+    from PIL import Image
+
+    with tempfile.NamedTemporaryFile(suffix=".jpg") as f:
+        Image.fromarray(img.permute(1, 2, 0).numpy()).save(f.name)
+        r32 = p32.extract([f.name])[0]
+        rbf = pbf.extract([f.name])[0]
+    assert rbf["features"].shape == r32["features"].shape
+    a = r32["features"].flatten().float()
+    b = rbf["features"].flatten().float()
+    cos = torch.nn.functional.cosine_similarity(a, b, dim=0)
+    assert cos > 0.98, float(cos)

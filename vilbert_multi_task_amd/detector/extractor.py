@@ -32,8 +32,11 @@ class DetectorFeatureProvider:
         conf_thresh: float = 0.0,
         min_size: int = 800,
         max_size: int = 1333,
+        dtype: str = "float32",
     ):
         self.model = (model or DetectionModel(DetectorConfig.x152())).eval().to(device)
+        if dtype == "bfloat16":
+            self.model.to_bf16()
         self.device = device
         self.num_features = num_features
         self.nms_iou = nms_iou

@@ -75,10 +75,26 @@ class DetectionModel(nn.Module):
             cfg.fpn_channels, cfg.pool_size, cfg.rep_dim, cfg.num_classes
         )
 
+    def to_bf16(self) -> "DetectionModel":
+        """Serving-precision mode: convs (MIOpen) + fc6/cls GEMMs in bf16;
+        box arithmetic stays fp32 (anchor coords up to 1333 px exceed bf16's
+        8 mantissa bits — a bf16 anchor is off by multiple pixels, and the
+        RPN decode/NMS math is latency-trivial anyway). The RPN conv head
+        emits bf16 which decode upcasts (rpn.py .float() calls)."""
+        self.backbone.to(torch.bfloat16)
+        self.box_head.to(torch.bfloat16)
+        self.rpn.to(torch.bfloat16)
+        for m in self.rpn.modules():  # anchors and friends back to fp32
+            for k, b in m._buffers.items():
+                if b is not None and b.dtype == torch.bfloat16:
+                    m._buffers[k] = b.float()
+        return self
+
     @torch.no_grad()
     def forward(
         self, images: torch.Tensor, image_sizes: List[Tuple[int, int]]
     ) -> List[Dict[str, torch.Tensor]]:
+        images = images.to(next(self.backbone.parameters()).dtype)
         feats = self.backbone(images)
         proposals = self.rpn(feats, image_sizes)
         results = []
