@@ -79,48 +79,29 @@ def test_serving_worker_gpu_end_to_end(tmp_path):
 
 
 def test_x152_detector_bf16_matches_fp32():
-    """bf16 serving mode of the full X-152 stack: same kept boxes (box math
-    stays fp32), fc6 features cosine-close to the fp32 run."""
-    from vilbert_multi_task_amd.detector import DetectionModel, DetectorConfig
-    from vilbert_multi_task_amd.detector.extractor import DetectorFeatureProvider
-
-    torch.manual_seed(0)
-    m32 = DetectionModel(DetectorConfig.x152()).eval()
-    p32 = DetectorFeatureProvider(m32, device="cuda")
+    """bf16 serving mode vs fp32 on FIXED proposals. The full pipeline
+    cannot be compared end-to-end with random-init weights: RPN objectness
+    of a random network is near-uniform, so any numeric perturbation
+    reshuffles the kept top-100 (measured: only 11% box overlap). Instead
+    feed both precisions the same rois and compare the fc6 features —
+    which is what ViLBERT actually consumes."""
     import copy
 
-    mbf = copy.deepcopy(m32)
-    pbf = DetectorFeatureProvider(mbf, device="cuda", dtype="bfloat16")
+    from vilbert_multi_task_amd.detector import DetectionModel, DetectorConfig
 
-    img = (torch.rand(3, 480, 640) * 255).to(torch.uint8)
-    import tempfile
-
-    from PIL import Image
-
-    with tempfile.NamedTemporaryFile(suffix=".jpg") as f:
-        Image.fromarray(img.permute(1, 2, 0).numpy()).save(f.name)
-        r32 = p32.extract([f.name])[0]
-        rbf = pbf.extract([f.name])[0]
-    assert rbf["features"].shape == r32["features"].shape
-    # Match boxes across the two runs (bf16 logits can reorder the top-100
-    # tail) and compare fc6 per matched pair. Random-init X-152 is the
-    # worst case for bf16 accumulation — 152 layers of random weights have
-    # no trained scale structure — so the gate is deliberately loose; with
-    # real checkpoints agreement is far tighter.
-    def box_iou(a, b):
-        area_a = (a[:, 2] - a[:, 0]).clamp(min=0) * (a[:, 3] - a[:, 1]).clamp(min=0)
-        area_b = (b[:, 2] - b[:, 0]).clamp(min=0) * (b[:, 3] - b[:, 1]).clamp(min=0)
-        lt = torch.maximum(a[:, None, :2], b[None, :, :2])
-        rb = torch.minimum(a[:, None, 2:], b[None, :, 2:])
-        wh = (rb - lt).clamp(min=0)
-        inter = wh[..., 0] * wh[..., 1]
-        return inter / (area_a[:, None] + area_b[None, :] - inter + 1e-9)
-
-    iou = box_iou(rbf["bbox"].float(), r32["bbox"].float())
-    best, idx = iou.max(dim=1)
-    matched = best > 0.95
-    assert matched.float().mean() > 0.6, float(matched.float().mean())
-    a = rbf["features"][matched].float()
-    b = r32["features"][idx[matched]].float()
-    cos = torch.nn.functional.cosine_similarity(a, b, dim=1).mean()
-    assert cos > 0.9, float(cos)
+    torch.manual_seed(0)
+    m32 = DetectionModel(DetectorConfig.x152()).eval().cuda()
+    mbf = copy.deepcopy(m32).to_bf16()
+    images = torch.randn(1, 3, 480, 640, device="cuda")
+    with torch.no_grad():
+        f32 = m32.backbone(images)
+        fbf = mbf.backbone(images.bfloat16())
+        boxes = torch.rand(100, 4, device="cuda") * 300
+        boxes[:, 2:] += boxes[:, :2] + 32
+        rois = torch.cat([torch.zeros(100, 1, device="cuda"), boxes], dim=1)
+        fc32 = m32.box_head(m32.pooler(f32[:4], rois))[0].float()
+        fcbf = mbf.box_head(mbf.pooler(fbf[:4], rois))[0].float()
+    cos = torch.nn.functional.cosine_similarity(fc32, fcbf, dim=1)
+    # random-init X-152 is the bf16 worst case (no trained scale structure)
+    assert cos.mean() > 0.9, float(cos.mean())
+    assert cos.min() > 0.7, float(cos.min())
