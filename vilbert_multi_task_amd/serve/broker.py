@@ -102,11 +102,16 @@ class Broker:
                         "UPDATE messages SET state='dead' WHERE id=?", (r["id"],)
                     )
                     continue
-                self._conn.execute(
+                # atomic claim: the state='ready' guard makes competing
+                # consumers safe — SQLite serializes the writes, the loser's
+                # UPDATE matches 0 rows and the message is delivered once
+                cur = self._conn.execute(
                     "UPDATE messages SET state='unacked', consumer=?, leased_at=?,"
-                    " attempts=attempts+1 WHERE id=?",
+                    " attempts=attempts+1 WHERE id=? AND state='ready'",
                     (self.consumer_tag, time.time(), r["id"]),
                 )
+                if cur.rowcount == 0:
+                    continue  # another consumer won this message
                 out.append(Delivery(int(r["id"]), json.loads(r["body"]), int(r["attempts"]) + 1))
         return out
 
