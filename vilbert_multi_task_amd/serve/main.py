@@ -56,9 +56,17 @@ def build_worker(args):
         torch.manual_seed(0)
         model = VILBertForVLTasks(cfg)
     device = "cuda" if torch.cuda.is_available() else "cpu"
-    if device == "cuda":
+    if device == "cuda" and args.device >= 0:
+        # multi-GPU serving = one worker process per GPU, all competing on
+        # the same durable queue (SURVEY.md §2.4 scale-out model):
+        #   for i in $(seq 0 7); do
+        #     python -m vilbert_multi_task_amd.serve.main worker --device $i &
+        #   done
+        torch.cuda.set_device(args.device)
+        device = f"cuda:{args.device}"
+    if device.startswith("cuda"):
         model = model.to(device=device, dtype=torch.bfloat16)
-    runner = GraphRunner(model, device=device, use_graphs=device == "cuda", serving_fast=True, fp8=args.fp8)
+    runner = GraphRunner(model, device=device, use_graphs=device.startswith("cuda"), serving_fast=True, fp8=args.fp8)
 
     provider = None
     if args.detector:
@@ -98,6 +106,9 @@ def main() -> None:
                     help="fp8 (e4m3) encoder GEMMs: +9-14%% throughput, "
                     "logit cosine > 0.97 vs bf16 (docs/PERFORMANCE.md)")
     ap.add_argument("--max-batch", type=int, default=64)
+    ap.add_argument("--device", type=int, default=-1,
+                    help="GPU ordinal for this worker (-1: default device); "
+                    "run one worker per GPU against the same --queue")
     ap.add_argument("--metrics-port", type=int, default=0,
                     help="Prometheus /metrics port for the worker (0 = off)")
     args = ap.parse_args()
