@@ -66,8 +66,8 @@ DEV bf16x8 lds_b128(const char* p) {
 
 DEV void lds_store_b128(char* p, uint4 v) { *reinterpret_cast<uint4*>(p) = v; }
 
-template <int D, bool KGLOBAL, int NTMAX, bool FP8OUT = false>
-__global__ __launch_bounds__(256) void attn_kernel(
+template <int D, bool KGLOBAL, int NTMAX, bool FP8OUT = false, int THREADS = 256>
+__global__ __launch_bounds__(THREADS) void attn_kernel(
     const bf16* __restrict__ q, const bf16* __restrict__ kg,
     const bf16* __restrict__ vg, const bf16* __restrict__ mask,
     bf16* __restrict__ out, int B, int H, int Lq, int Lk, int mask_mode,
@@ -362,13 +362,28 @@ void launch_attention_impl(const bf16* q, const bf16* k, const bf16* v,
   }();
   // measured on MI355X (B=256): nsplit>1 duplicates staging and loses
   int nsplit = nsplit_env > 0 ? nsplit_env : 1;
-  // (K if staged) + V + 4x per-wave P (all bf16)
+  static const int waves_env = [] {
+    const char* e = getenv("VILBERT_ATTN_WAVES");
+    return e ? atoi(e) : 0;  // 0 = default (4); 8 = double-occupancy D=128
+  }();
+  const int nwaves = (waves_env == 8 && D == 128 && !kglobal_env) ? 8 : 4;
+  // (K if staged) + V + per-wave P (all bf16)
   const size_t lds = sizeof(bf16) *
-      (size_t)((kglobal_env ? 1 : 2) * LK_PAD * D + 4 * 16 * LK_PAD);
+      (size_t)((kglobal_env ? 1 : 2) * LK_PAD * D + nwaves * 16 * LK_PAD);
   const dim3 grid(B * H * nsplit);
 #define LAUNCH_ATTN(DD, KG, NTM)                                              \
   do {                                                                        \
-    if (out8)                                                                 \
+    if (nwaves == 8) {                                                        \
+      if (out8)                                                               \
+        hipLaunchKernelGGL((attn_kernel<DD, KG, NTM, true, 512>), grid,       \
+                           dim3(512), lds, stream, q, k, v, mask, out, B, H,  \
+                           Lq, Lk, mask_mode, scale, qs, ks, vs, nsplit,      \
+                           out8, fp8_scales, fp8_amaxes, fp8_site);           \
+      else                                                                    \
+        hipLaunchKernelGGL((attn_kernel<DD, KG, NTM, false, 512>), grid,      \
+                           dim3(512), lds, stream, q, k, v, mask, out, B, H,  \
+                           Lq, Lk, mask_mode, scale, qs, ks, vs, nsplit);     \
+    } else if (out8)                                                          \
       hipLaunchKernelGGL((attn_kernel<DD, KG, NTM, true>), grid, dim3(256),   \
                          lds, stream, q, k, v, mask, out, B, H, Lq, Lk,       \
                          mask_mode, scale, qs, ks, vs, nsplit, out8,          \
