@@ -61,3 +61,31 @@ def test_base_bert_single_stream_ten_outputs(tiny_config):
     assert out[0].shape == (2, tiny_config.num_labels_vqa)
     assert out[5].shape == (2, 12, tiny_config.v_target_size)
     assert out[7].shape[1] == 21  # task token inserted
+
+
+def test_pretrain_script_smoke(tmp_path):
+    import subprocess
+    import sys
+
+    r = subprocess.run(
+        [sys.executable, "scripts/pretrain.py", "--tiny", "--steps", "2",
+         "--batch", "2", "--log-every", "1", "--save-every", "2",
+         "--checkpoint", str(tmp_path / "ck.bin")],
+        capture_output=True, text=True, timeout=240,
+    )
+    assert r.returncode == 0, r.stderr[-800:]
+    assert '"event": "pretrain"' in r.stdout
+    assert (tmp_path / "ck.bin").exists()
+
+
+def test_evaluate_script_smoke():
+    import subprocess
+    import sys
+
+    r = subprocess.run(
+        [sys.executable, "scripts/evaluate.py", "--tiny", "--datasets",
+         "vqa_v2", "snli_ve", "--batches", "1", "--batch", "4"],
+        capture_output=True, text=True, timeout=240,
+    )
+    assert r.returncode == 0, r.stderr[-800:]
+    assert '"event": "eval_summary"' in r.stdout
