@@ -304,6 +304,234 @@ __global__ __launch_bounds__(THREADS) void attn_kernel(
 }
 
 // ---------------------------------------------------------------------------
+// Pipelined variant (VILBERT_ATTN_PIPE=1): hide the V staging latency behind
+// the first stripe's S/softmax compute. PMC (profiles/r06) shows the default
+// kernel at 13% MFMA-busy — staging-latency bound — because ALL K+V HBM
+// loads must land (and store to LDS) before the first MFMA issues. Here:
+//   1. issue K loads, then V loads; store ONLY K; barrier   (V still flying)
+//   2. each wave runs its FIRST stripe's S = QK^T + softmax + P->LDS
+//      (needs only K_lds) while the V loads complete in the background
+//   3. all threads store V transposed from registers; barrier
+//   4. first stripe's O = P·V, then the remaining stripes run the full
+//      unpipelined body (V resident by then).
+// The phase barriers sit OUTSIDE the per-stripe loops (every thread reaches
+// them exactly once), so unequal stripe counts across waves cannot deadlock.
+// ---------------------------------------------------------------------------
+template <int D, int NTMAX, bool FP8OUT = false>
+__global__ __launch_bounds__(256) void attn_kernel_pipe(
+    const bf16* __restrict__ q, const bf16* __restrict__ kg,
+    const bf16* __restrict__ vg, const bf16* __restrict__ mask,
+    bf16* __restrict__ out, int B, int H, int Lq, int Lk, int mask_mode,
+    float scale, int qs, int ks, int vs, int nsplit,
+    unsigned char* __restrict__ out8 = nullptr,
+    const float* __restrict__ fp8_scales = nullptr,
+    float* __restrict__ fp8_amaxes = nullptr, int fp8_site = 0) {
+  const float fp8_inv = FP8OUT ? 1.0f / fp8_scales[fp8_site] : 0.f;
+  float fp8_amax = 0.f;
+  constexpr int KCH = D / 8;
+  const int HD = H * D;
+  const int bh = blockIdx.x;  // nsplit unused in the pipe variant
+  const int b = bh / H;
+  const int h = bh % H;
+  const int tid = threadIdx.x;
+  const int lane = lane_id();
+  const int wid = wave_id();
+
+  const int LK_PAD = (Lk + 31) & ~31;
+  const int NT = LK_PAD / 16;
+
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  char* K_lds = smem;
+  char* V_lds = smem + LK_PAD * D * 2;
+  char* P_lds = V_lds + LK_PAD * D * 2 + wid * 16 * LK_PAD * 2;
+
+  // ---- phase 1: K loads first, V loads after, store K only --------------
+  const int rows_per_pass = blockDim.x / KCH;
+  const int r0 = tid / KCH;
+  const int c = tid % KCH;
+  const long kbase0 = (long)b * Lk * ks + (long)h * D;
+  const long vbase0 = (long)b * Lk * vs + (long)h * D;
+  const int npass = (LK_PAD + rows_per_pass - 1) / rows_per_pass;
+  uint4 kr[8], vr[8];
+#pragma unroll
+  for (int pi = 0; pi < 8; ++pi) {
+    if (pi >= npass) break;
+    const int r = r0 + pi * rows_per_pass;
+    kr[pi] = make_uint4(0, 0, 0, 0);
+    if (r < Lk)
+      kr[pi] = *reinterpret_cast<const uint4*>(kg + kbase0 + (long)r * ks + c * 8);
+  }
+#pragma unroll
+  for (int pi = 0; pi < 8; ++pi) {
+    if (pi >= npass) break;
+    const int r = r0 + pi * rows_per_pass;
+    vr[pi] = make_uint4(0, 0, 0, 0);
+    if (r < Lk)
+      vr[pi] = *reinterpret_cast<const uint4*>(vg + vbase0 + (long)r * vs + c * 8);
+  }
+#pragma unroll
+  for (int pi = 0; pi < 8; ++pi) {
+    if (pi >= npass) break;
+    const int r = r0 + pi * rows_per_pass;
+    if (r >= LK_PAD) break;
+    lds_store_b128(K_lds + r * (D * 2) + ((c * 16) ^ SWZ(r)), kr[pi]);
+  }
+  __syncthreads();  // K_lds ready; vr[] still in flight / in regs
+
+  const int nstripes = (Lq + 15) / 16;
+  const int nw = blockDim.x / WAVE;
+  const int col0 = lane & 15;
+
+  float inv_l[4];
+  // ---- S phase for one stripe: S = QK^T, mask+softmax, P -> P_lds -------
+  auto s_phase = [&](int qrow0) {
+    bf16x8 aq[D / 32];
+    {
+      const int row = min(qrow0 + (lane & 15), Lq - 1);
+      const long qoff = ((long)b * Lq + row) * qs + (long)h * D + (lane >> 4) * 8;
+#pragma unroll
+      for (int kk = 0; kk < D / 32; ++kk) aq[kk] = load_bf16x8(q + qoff + kk * 32);
+    }
+    f32x4 acc_s[NTMAX];
+#pragma unroll
+    for (int nt = 0; nt < NTMAX; ++nt) acc_s[nt] = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+    for (int nt = 0; nt < NTMAX; ++nt) {
+      if (nt >= NT) break;
+      const int key = nt * 16 + (lane & 15);
+      const char* kbase = K_lds + key * (D * 2);
+      const int ksw = SWZ(key);
+#pragma unroll
+      for (int kk = 0; kk < D / 32; ++kk) {
+        const bf16x8 bk = lds_b128(kbase + (((kk * 64) + ((lane >> 4) * 16)) ^ ksw));
+        acc_s[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            (bf16x8)aq[kk], bk, acc_s[nt], 0, 0, 0);
+      }
+    }
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int row = qrow0 + (lane >> 4) * 4 + r;
+      float mx = -3.0e38f;
+#pragma unroll
+      for (int nt = 0; nt < NTMAX; ++nt) {
+        if (nt >= NT) break;
+        const int col = nt * 16 + col0;
+        float sv = acc_s[nt][r] * scale;
+        if (col < Lk) {
+          if (mask_mode == 1)
+            sv += bf2f(mask[(long)b * Lk + col]);
+          else if (mask_mode == 2)
+            sv += bf2f(mask[((long)b * Lq + min(row, Lq - 1)) * Lk + col]);
+          acc_s[nt][r] = sv;
+          mx = fmaxf(mx, sv);
+        } else {
+          acc_s[nt][r] = -3.0e38f;
+        }
+      }
+      const float mrow = group16_max(mx);
+      float sum = 0.f;
+#pragma unroll
+      for (int nt = 0; nt < NTMAX; ++nt) {
+        if (nt >= NT) break;
+        const int col = nt * 16 + col0;
+        const float p = (col < Lk) ? __expf(acc_s[nt][r] - mrow) : 0.f;
+        acc_s[nt][r] = p;
+        sum += p;
+      }
+      inv_l[r] = 1.0f / group16_sum(sum);
+    }
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int prow = (lane >> 4) * 4 + r;
+      char* prow_base = P_lds + prow * (LK_PAD * 2);
+      const int psw = SWZ(prow);
+#pragma unroll
+      for (int nt = 0; nt < NTMAX; ++nt) {
+        if (nt >= NT) break;
+        const int col = nt * 16 + col0;
+        *reinterpret_cast<short*>(prow_base + ((col * 2) ^ psw)) =
+            (short)f2us(acc_s[nt][r]);
+      }
+    }
+  };
+
+  // ---- PV phase for one stripe: O = P·V, normalize, store ---------------
+  auto pv_phase = [&](int qrow0) {
+    f32x4 acc_o[D / 16];
+#pragma unroll
+    for (int nt = 0; nt < D / 16; ++nt) acc_o[nt] = {0.f, 0.f, 0.f, 0.f};
+    const int parow = lane & 15;
+    const char* pa_base = P_lds + parow * (LK_PAD * 2);
+    const int pasw = SWZ(parow);
+#pragma unroll
+    for (int kk = 0; kk < NTMAX / 2; ++kk) {
+      if (kk * 32 >= LK_PAD) break;
+      const bf16x8 ap =
+          lds_b128(pa_base + (((kk * 64) + ((lane >> 4) * 16)) ^ pasw));
+      const int keyoff = (kk * 64) + ((lane >> 4) * 16);
+#pragma unroll
+      for (int nt = 0; nt < D / 16; ++nt) {
+        const int d = nt * 16 + (lane & 15);
+        const bf16x8 bv =
+            lds_b128(V_lds + d * (LK_PAD * 2) + (keyoff ^ SWZ(d)));
+        acc_o[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ap, bv, acc_o[nt], 0, 0, 0);
+      }
+    }
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int row = qrow0 + (lane >> 4) * 4 + r;
+      if (row < Lq) {
+        const long obase = ((long)b * Lq + row) * HD + (long)h * D;
+#pragma unroll
+        for (int nt = 0; nt < D / 16; ++nt) {
+          const float o = acc_o[nt][r] * inv_l[r];
+          out[obase + nt * 16 + col0] = f2bf(o);
+          if (FP8OUT) {
+            fp8_amax = fmaxf(fp8_amax, fabsf(o));
+            const float cc = fminf(fmaxf(o * fp8_inv, -448.f), 448.f);
+            int w = __builtin_amdgcn_cvt_pk_fp8_f32(cc, cc, 0, false);
+            out8[obase + nt * 16 + col0] = (unsigned char)(w & 0xff);
+          }
+        }
+      }
+    }
+  };
+
+  // ---- phase 2: first stripe's S while V loads land ----------------------
+  const bool has0 = wid < nstripes;
+  if (has0) s_phase(wid * 16);
+
+  // ---- phase 3: V transposed stores from regs, then block-wide barrier ---
+#pragma unroll
+  for (int pi = 0; pi < 8; ++pi) {
+    if (pi >= npass) break;
+    const int r = r0 + pi * rows_per_pass;
+    if (r >= LK_PAD) break;
+    union { uint4 u; short s[8]; } vv;
+    vv.u = vr[pi];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const int d = c * 8 + j;
+      *reinterpret_cast<short*>(
+          V_lds + d * (LK_PAD * 2) + ((r * 2) ^ SWZ(d))) = vv.s[j];
+    }
+  }
+  __syncthreads();  // V_lds ready for every wave
+
+  // ---- phase 4: first stripe's PV, then the remaining stripes ------------
+  if (has0) pv_phase(wid * 16);
+  for (int s = wid + nw; s < nstripes; s += nw) {
+    s_phase(s * 16);
+    pv_phase(s * 16);
+  }
+  if (FP8OUT) {
+    fp8_amax = wave_max(fp8_amax);
+    if (lane_id() == 0 && fp8_amax > 0.f)
+      attn_atomic_max_f32(&fp8_amaxes[fp8_site], fp8_amax);
+  }
+}
+
+// ---------------------------------------------------------------------------
 // mfma layout probe: one-wave 16x16x32 product for on-device layout checks
 // ---------------------------------------------------------------------------
 __global__ void mfma_probe_kernel(const bf16* __restrict__ a,
@@ -366,6 +594,10 @@ void launch_attention_impl(const bf16* q, const bf16* k, const bf16* v,
     const char* e = getenv("VILBERT_ATTN_WAVES");
     return e ? atoi(e) : 0;  // 0 = default (4); 8 = double-occupancy D=128
   }();
+  static const int pipe_env = [] {
+    const char* e = getenv("VILBERT_ATTN_PIPE");
+    return e ? atoi(e) : 0;  // 1 = V-staging pipelined behind stripe-0 S
+  }();
   const int nwaves = (waves_env == 8 && D == 128 && !kglobal_env) ? 8 : 4;
   // (K if staged) + V + per-wave P (all bf16)
   const size_t lds = sizeof(bf16) *
@@ -373,7 +605,17 @@ void launch_attention_impl(const bf16* q, const bf16* k, const bf16* v,
   const dim3 grid(B * H * nsplit);
 #define LAUNCH_ATTN(DD, KG, NTM)                                              \
   do {                                                                        \
-    if (nwaves == 8) {                                                        \
+    if (pipe_env && !KG && nsplit == 1 && nwaves == 4) {                      \
+      if (out8)                                                               \
+        hipLaunchKernelGGL((attn_kernel_pipe<DD, NTM, true>), grid, dim3(256),\
+                           lds, stream, q, k, v, mask, out, B, H, Lq, Lk,     \
+                           mask_mode, scale, qs, ks, vs, nsplit, out8,        \
+                           fp8_scales, fp8_amaxes, fp8_site);                 \
+      else                                                                    \
+        hipLaunchKernelGGL((attn_kernel_pipe<DD, NTM, false>), grid,          \
+                           dim3(256), lds, stream, q, k, v, mask, out, B, H,  \
+                           Lq, Lk, mask_mode, scale, qs, ks, vs, nsplit);     \
+    } else if (nwaves == 8) {                                                        \
       if (out8)                                                               \
         hipLaunchKernelGGL((attn_kernel<DD, KG, NTM, true, 512>), grid,       \
                            dim3(512), lds, stream, q, k, v, mask, out, B, H,  \
