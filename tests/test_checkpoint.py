@@ -57,3 +57,33 @@ def test_from_pretrained_file(tmp_path, tiny_model, tiny_config):
         tiny_model.state_dict().items(), m2.state_dict().items()
     ):
         assert torch.equal(v1, v2), k1
+
+
+def test_convert_checkpoint_cli_roundtrip(tmp_path, tiny_config):
+    """CLI to-upstream then to-native reproduces the weights exactly."""
+    import json as _json
+    import subprocess
+    import sys
+
+    from vilbert_multi_task_amd.models.heads import VILBertForVLTasks
+
+    cfg_path = tmp_path / "cfg.json"
+    cfg_path.write_text(_json.dumps(tiny_config.to_dict()))
+    torch.manual_seed(3)
+    m = VILBertForVLTasks(tiny_config)
+    native = tmp_path / "native.pt"
+    torch.save(m.state_dict(), native)
+
+    up = tmp_path / "pytorch_model_9.bin"
+    back = tmp_path / "back.pt"
+    for mode, src, dst in [("to-upstream", native, up), ("to-native", up, back)]:
+        r = subprocess.run(
+            [sys.executable, "scripts/convert_checkpoint.py", mode, str(src),
+             str(dst), "--config", str(cfg_path)],
+            capture_output=True, text=True, timeout=240,
+        )
+        assert r.returncode == 0, r.stderr[-500:]
+    sd0 = m.state_dict()
+    sd1 = torch.load(back, weights_only=True)
+    for k in sd0:
+        assert torch.equal(sd0[k], sd1[k]), k
