@@ -253,3 +253,31 @@ def test_tensorize_regions_mixed_box_counts():
     # global box row
     assert torch.allclose(out["spatials"][0, 0], torch.tensor([0.0, 0, 1, 1, 1]))
     assert torch.allclose(out["spatials"][1, 1, :4], torch.tensor([0.0, 0.0, 0.5, 0.5]))
+
+
+def test_mixed_batch_nlvr2_first_even_alignment(serving):
+    """A drained batch mixing NLVR2 (2-row) and single-row tasks places the
+    pair requests first at even row offsets (worker gather contract — the
+    pair head reads consecutive even/odd rows)."""
+    broker, db, push, worker = serving
+    from vilbert_multi_task_amd.serve.broker import vilbert_task
+
+    vilbert_task(broker, ["/x.jpg"], "what is this", 1, "s1")
+    vilbert_task(broker, ["/a.jpg", "/b.jpg"], "true or false", 12, "s2")
+    vilbert_task(broker, ["/y.jpg"], "entailment?", 13, "s3")
+    vilbert_task(broker, ["/c.jpg", "/d.jpg"], "left image has", 12, "s4")
+    reqs = worker.gather_batch(0.0)
+    assert len(reqs) == 4
+    # NLVR2 requests first, each starting on an even row
+    kinds = [r.task_id for r in reqs]
+    assert kinds[:2] == [12, 12]
+    for r in reqs:
+        if r.task_id == 12:
+            assert r.row_start % 2 == 0 and r.num_rows == 2
+    # the gathered requests were leased (unacked); nack them back so the
+    # full-path check below serves a fresh equivalent mix
+    for r in reqs:
+        broker.nack(r.delivery.msg_id)
+    n = 0
+    while n < 4:
+        n += worker.process_once()
