@@ -238,3 +238,26 @@ def test_dropped_rank_detected(tmp_path):
             pytest.fail("surviving rank hung instead of detecting the dead peer")
     outcome = (tmp_path / "outcome.txt").read_text()
     assert outcome.startswith("raised:"), outcome
+
+
+def test_resume_continuation_bitwise():
+    """Crash-recovery contract: a trainer resumed from a
+    checkpoint continues BITWISE identically to one that never stopped
+    (model + AdamW moments + sampler state all round-trip; dropout=0)."""
+    import tempfile
+
+    with tempfile.TemporaryDirectory() as td:
+        path = os.path.join(td, "ck.bin")
+        ref = _tiny_trainer()
+        for _ in range(3):
+            ref.train_step()
+        ref.save_checkpoint(path)
+        resumed = _tiny_trainer()
+        resumed.load_checkpoint(path)
+        for _ in range(2):  # continue both for two more steps
+            ref.train_step()
+            resumed.train_step()
+        for (k1, v1), (k2, v2) in zip(
+            ref.model.state_dict().items(), resumed.model.state_dict().items()
+        ):
+            assert k1 == k2 and torch.equal(v1, v2), k1
