@@ -111,11 +111,16 @@ def get_metrics():
     return _METRICS
 
 
-def start_metrics_server(port: int) -> bool:
+def start_metrics_server(port: int) -> Optional[int]:
+    """Start the Prometheus exposition endpoint. Returns the bound port
+    (port=0 binds an ephemeral one when the client lib reports it back),
+    or None when prometheus_client is unavailable."""
     try:
         from prometheus_client import start_http_server
 
-        start_http_server(port)
-        return True
+        res = start_http_server(port)
+        if isinstance(res, tuple) and res and hasattr(res[0], "server_port"):
+            return int(res[0].server_port)
+        return port if port else None
     except Exception:
-        return False
+        return None
