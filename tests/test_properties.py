@@ -129,3 +129,50 @@ def _run_broker_ops(path, ops):
             broker.ack(d.msg_id)
     assert seen | acked_ids == ids  # nothing lost
     assert seen <= ids and acked_ids <= ids  # nothing invented
+
+
+@given(
+    rows=st.integers(min_value=1, max_value=5),
+    dim=st.sampled_from([8, 24, 52, 64]),
+    has_res=st.booleans(),
+    seed=st.integers(min_value=0, max_value=2**16),
+)
+@settings(max_examples=30, deadline=None)
+def test_layer_norm_oracle_matches_torch(rows, dim, has_res, seed):
+    """The CPU dispatch path of F_ops.layer_norm (the numerics oracle every
+    HIP kernel is tested against) must equal torch layer_norm exactly."""
+    import torch
+
+    from vilbert_multi_task_amd.ops import functional as F_ops
+
+    g = torch.Generator().manual_seed(seed)
+    x = torch.randn(rows, dim, generator=g)
+    res = torch.randn(rows, dim, generator=g) if has_res else None
+    w = torch.randn(dim, generator=g)
+    b = torch.randn(dim, generator=g)
+    got = F_ops.layer_norm(x, w, b, 1e-12, residual=res)
+    ref = torch.nn.functional.layer_norm(
+        x + res if has_res else x, (dim,), w, b, 1e-12
+    )
+    assert torch.allclose(got, ref, atol=1e-6), (got - ref).abs().max()
+
+
+@given(
+    m=st.integers(min_value=1, max_value=6),
+    k=st.sampled_from([4, 16, 32]),
+    n=st.sampled_from([3, 8, 17]),
+    seed=st.integers(min_value=0, max_value=2**16),
+)
+@settings(max_examples=30, deadline=None)
+def test_linear_bias_gelu_oracle_matches_torch(m, k, n, seed):
+    import torch
+
+    from vilbert_multi_task_amd.ops import functional as F_ops
+
+    g = torch.Generator().manual_seed(seed)
+    x = torch.randn(m, k, generator=g)
+    w = torch.randn(n, k, generator=g)
+    b = torch.randn(n, generator=g)
+    got = F_ops.linear_bias_gelu(x, w, b)
+    ref = torch.nn.functional.gelu(torch.nn.functional.linear(x, w, b))
+    assert torch.allclose(got, ref, atol=1e-5), (got - ref).abs().max()
