@@ -304,3 +304,47 @@ def test_prometheus_metrics_exposition(serving):
     ).read().decode()
     assert "vilbert_requests_total" in body
     assert "vilbert_batch_rows" in body
+
+
+def test_runner_chunked_replay_concatenation(tiny_model, tiny_config):
+    """Batches above max_bucket split into even chunks whose outputs are
+    torch.cat-ed, with the pair head (index 3) carrying n_chunk/2 rows
+    (the fp8 bucket-cap path in engine/runner.py)."""
+    from vilbert_multi_task_amd.data.synthetic import synthetic_batch
+    from vilbert_multi_task_amd.engine.runner import GraphRunner
+
+    runner = GraphRunner(
+        tiny_model, device="cpu", use_graphs=False,
+        feat_dim=tiny_config.v_feature_size, seq_len=20, regions=12,
+    )
+    runner.use_graphs = True  # force the chunk branch of run()
+    runner.max_bucket = 4
+    seen = []
+    orig_run = GraphRunner.run
+
+    def spy(self, batch):
+        n = batch["question"].shape[0]
+        if n <= self.max_bucket:  # leaf chunk: serve eagerly, record size
+            seen.append(n)
+            self.use_graphs = False
+            try:
+                return orig_run(self, batch)
+            finally:
+                self.use_graphs = True
+        return orig_run(self, batch)
+
+    b = synthetic_batch(
+        10, seq_len=20, regions=12, feat_dim=tiny_config.v_feature_size,
+        vocab_size=tiny_config.vocab_size,
+    )
+    import unittest.mock as mock
+
+    with mock.patch.object(GraphRunner, "run", spy):
+        out = GraphRunner.run(runner, b)
+    assert seen == [4, 4, 2]
+    assert out[0].shape[0] == 10          # row-wise heads concatenated
+    assert out[3].shape[0] == 5           # pair head: 2+2+1 rows
+    # chunked result equals the unchunked eager forward
+    runner.use_graphs = False
+    ref = runner.run(b)
+    assert torch.allclose(out[0], ref[0], atol=1e-5)
