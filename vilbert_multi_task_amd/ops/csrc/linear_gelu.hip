@@ -157,6 +157,19 @@ Plan& get_plan(long M, long N, long K, int kind, void* workspace, size_t ws_byte
 static void autotune(Plan& p, const void* x, const void* w, const void* bias,
                      const void* cmat, void* y, void* workspace,
                      size_t ws_bytes, hipStream_t stream) {
+  if (p.candidates.empty()) return;
+  // debug hook (scripts/debug_hipblaslt_algos.py): pin one candidate by
+  // heuristic position instead of timing — lets a subprocess-per-algo
+  // harness isolate which algo index faults at a given shape.
+  static const int pin_pos = [] {
+    const char* e = std::getenv("VILBERT_GEMM_TUNE_POS");
+    return e ? atoi(e) : -1;
+  }();
+  if (pin_pos >= 0) {
+    p.algo = p.candidates[std::min<size_t>(pin_pos, p.candidates.size() - 1)].algo;
+    p.candidates.clear();
+    return;
+  }
   if (p.candidates.size() <= 1) return;
   float alpha = 1.0f, beta = p.beta;
   hipEvent_t ev0, ev1;
