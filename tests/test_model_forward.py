@@ -1,7 +1,6 @@
 """CPU tests of the 10-output forward contract (BASELINE.json config 1:
 tiny 2-layer ViLBERT VQA forward on CPU)."""
 
-import pytest
 import torch
 
 from vilbert_multi_task_amd.config import ViLBertConfig

@@ -1,6 +1,5 @@
 """Property-based tests (hypothesis) for the serving-side invariants."""
 
-import json
 
 import pytest
 import torch

@@ -7,9 +7,7 @@ import json
 import pytest
 import torch
 
-from vilbert_multi_task_amd.config import ViLBertConfig
 from vilbert_multi_task_amd.engine.runner import GraphRunner
-from vilbert_multi_task_amd.models import VILBertForVLTasks
 from vilbert_multi_task_amd.serve.broker import Broker, vilbert_task
 from vilbert_multi_task_amd.serve.db import Database
 from vilbert_multi_task_amd.serve.features import SyntheticFeatureProvider
