@@ -135,7 +135,9 @@ def test_http_worker_ws_across_processes(tmp_path):
 
         ws.settimeout(60)
         result = None
-        for _ in range(8):
+        # submit pushes 3 messages (publish before/after + info) and the
+        # worker 3 more per request — scan a wide window for the result
+        for _ in range(14):
             msg = json.loads(_ws_recv(ws))
             if "result" in msg:
                 result = json.loads(msg["result"])
