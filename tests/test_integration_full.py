@@ -143,7 +143,7 @@ def test_http_worker_ws_across_processes(tmp_path):
                 result = json.loads(msg["result"])
                 break
         assert result is not None
-        assert result["task_id"] == 1 and len(result["result"]) == 3
+        assert result["task_id"] == "1" and len(result["result"]) == 3
         proc.join(30)
         assert proc.exitcode == 0
         ws.close()

@@ -51,18 +51,22 @@ def test_vqa_roundtrip(serving):
     res = _result_messages(push, "sock1")
     assert len(res) == 1
     payload = json.loads(res[0]["result"])
-    assert payload["task_id"] == 1
+    # reference wire contract (worker.py:564-579): STRING task_id,
+    # confidence scaled 0-100 and rounded to 2 decimals
+    assert payload["task_id"] == "1"
     assert len(payload["result"]) == 3
     for entry in payload["result"]:
         assert set(entry) == {"answer", "confidence"}
-        assert 0.0 <= entry["confidence"] <= 1.0
+        assert 0.0 <= entry["confidence"] <= 100.0
+        assert entry["confidence"] == round(entry["confidence"], 2)
     # DB row saved (worker.py:548-552,579-645 contract)
     qa = db.get_question(1)
     assert qa["input_text"] == "what animal is this?"
-    assert json.loads(qa["answer_text"])["task_id"] == 1
+    assert json.loads(qa["answer_text"])["task_id"] == "1"
     # terminal messages bracket the result (worker.py:647-649)
     terms = [p for s, p in push.messages if s == "sock1" and "terminal" in p]
-    assert any("Completed" in str(p["terminal"]) for p in terms)
+    # exact reference completion push (worker.py:649)
+    assert terms[-1]["terminal"] == "Completed Task"
 
 
 def test_nlvr2_pair(serving):
@@ -72,7 +76,7 @@ def test_nlvr2_pair(serving):
     payload = json.loads(_result_messages(push, "s12")[0]["result"])
     answers = {e["answer"] for e in payload["result"]}
     assert answers == {"True", "False"}
-    assert abs(sum(e["confidence"] for e in payload["result"]) - 1.0) < 1e-5
+    assert abs(sum(e["confidence"] for e in payload["result"]) - 100.0) < 1e-2
 
 
 def test_retrieval_orders_all_images(serving):
@@ -81,11 +85,14 @@ def test_retrieval_orders_all_images(serving):
     vilbert_task(broker, imgs, "a dog on a beach", 7, "s7")
     assert worker.process_once() == 1
     payload = json.loads(_result_messages(push, "s7")[0]["result"])
-    assert payload["task_id"] == 7
-    assert sorted(payload["image_name_list"]) == sorted(f"{i}.jpg" for i in range(4))
+    assert payload["task_id"] == "7"
+    # worker.py:631-635 name format: test2014/<basename>.<ext of first image>
+    assert sorted(payload["image_name_list"]) == sorted(
+        f"test2014/{i}.jpg" for i in range(4)
+    )
     confs = payload["confidence_list"]
     assert confs == sorted(confs, reverse=True)
-    assert abs(sum(confs) - 1.0) < 1e-5
+    assert abs(sum(confs) - 100.0) < 1e-2
 
 
 def test_grounding_boxes(serving):
@@ -93,11 +100,14 @@ def test_grounding_boxes(serving):
     vilbert_task(broker, ["/img/street.jpg"], "the red car", 11, "s11")
     assert worker.process_once() == 1
     payload = json.loads(_result_messages(push, "s11")[0]["result"])
-    assert payload["task_id"] == 11
-    assert len(payload["boxes"]) == 3
-    for box in payload["boxes"]:
-        x1, y1, x2, y2 = box
-        assert x2 >= x1 and y2 >= y1  # pixel-space, denormalized
+    # reference grounding wire schema (worker.py:599-604): task_id +
+    # image_name_list (bare uuids, no path/extension) + confidence_list
+    assert payload["task_id"] == "11"
+    assert set(payload) == {"task_id", "image_name_list", "confidence_list"}
+    assert len(payload["image_name_list"]) == 3
+    for name in payload["image_name_list"]:
+        assert "/" not in name and not name.endswith(".jpg")
+    assert all(0.0 <= c <= 100.0 for c in payload["confidence_list"])
 
 
 def test_mixed_task_batch(serving):
@@ -111,7 +121,7 @@ def test_mixed_task_batch(serving):
     assert broker.depth() == 0
     for sid, tid in [("m1", 1), ("m2", 12), ("m3", 11), ("m4", 15)]:
         payload = json.loads(_result_messages(push, sid)[0]["result"])
-        assert payload["task_id"] == tid
+        assert payload["task_id"] == str(tid)
 
 
 def test_invalid_image_count_rejected(serving):
@@ -346,3 +356,33 @@ def test_runner_chunked_replay_concatenation(tiny_model, tiny_config):
     runner.use_graphs = False
     ref = runner.run(b)
     assert torch.allclose(out[0], ref[0], atol=1e-5)
+
+
+def test_guesswhat_strict_parity_default(serving, monkeypatch):
+    """Default behavior matches the reference's OBSERVABLE semantics: the
+    q:/a: rewrite is computed-then-discarded there (worker.py:391-402), so
+    the RAW query reaches the tokenizer; VILBERT_GUESSWHAT_REWRITE=1 opts
+    into the intended rewrite."""
+    import vilbert_multi_task_amd.serve.worker as worker_mod
+
+    broker, db, push, worker = serving
+    seen = []
+    orig = worker.tokenizer.encode_for_serving
+
+    def spy(text, max_len):
+        seen.append(text)
+        return orig(text, max_len)
+
+    worker.tokenizer.encode_for_serving = spy
+    q = "q: is it a person? a: yes"
+
+    monkeypatch.delenv("VILBERT_GUESSWHAT_REWRITE", raising=False)
+    vilbert_task(broker, ["/g.jpg"], q, 16, "gw1")
+    assert worker.process_once() == 1
+    assert seen[-1] == q  # raw query (reference-exact)
+
+    monkeypatch.setenv("VILBERT_GUESSWHAT_REWRITE", "1")
+    worker._tok_cache.clear()
+    vilbert_task(broker, ["/g.jpg"], q, 16, "gw2")
+    assert worker.process_once() == 1
+    assert seen[-1] == "start is it a person? answer yes stop"

@@ -162,15 +162,16 @@ def create_app(
         from ..utils.trace import log_json, new_trace_id
 
         trace_id = new_trace_id()
-        # same submit-time terminal sequence as the reference producer
-        # (sender.py:25 before publish, sender.py:34 after)
+        # exact submit-time terminal sequence of the reference:
+        # views.py:35 before dispatch, then sender.py:25 / sender.py:34
+        # around the publish
+        _push_local(app, socket_id, {"terminal": "Starting Vilbert Multitask Job..."})
         _push_local(app, socket_id, {"terminal": "Publishing job to ViLBERT Queue"})
         vilbert_task(
             app.state.broker, paths, question, task_id or "1", socket_id,
             trace_id=trace_id,
         )
         _push_local(app, socket_id, {"terminal": "Job published successfully"})
-        _push_local(app, socket_id, {"info": "Task submitted"})
         log_json("submit", trace_id=trace_id, task_id=task_id, socket_id=socket_id)
         return HTMLResponse(_INDEX_HTML)
 

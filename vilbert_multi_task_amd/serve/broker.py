@@ -87,10 +87,22 @@ class Broker:
             )
 
     def get(self, queue: str = QUEUE_NAME, max_n: int = 1) -> List[Delivery]:
-        """Lease up to max_n ready messages (at-least-once semantics)."""
+        """Lease up to max_n ready messages (at-least-once semantics).
+
+        The claim runs inside BEGIN IMMEDIATE: a deferred transaction that
+        reads then upgrades to a write can hit SQLITE_BUSY_SNAPSHOT
+        ("database is locked", NOT retried by the busy timeout) when a
+        competing consumer commits between the SELECT and the UPDATE —
+        taking the reserved lock up front makes contention wait on the
+        30 s busy timeout instead of raising.
+        """
         self._requeue_expired(queue)
         out: List[Delivery] = []
-        with self._conn:
+        try:
+            self._conn.execute("BEGIN IMMEDIATE")
+        except sqlite3.OperationalError:
+            return out  # writer contention past the busy timeout: back off
+        try:
             rows = self._conn.execute(
                 "SELECT id, body, attempts FROM messages WHERE queue=? AND"
                 " state='ready' ORDER BY id LIMIT ?",
@@ -113,6 +125,10 @@ class Broker:
                 if cur.rowcount == 0:
                     continue  # another consumer won this message
                 out.append(Delivery(int(r["id"]), json.loads(r["body"]), int(r["attempts"]) + 1))
+            self._conn.execute("COMMIT")
+        except BaseException:
+            self._conn.execute("ROLLBACK")
+            raise
         return out
 
     def ack(self, msg_id: int) -> None:

@@ -45,16 +45,27 @@ from .features import SyntheticFeatureProvider, tensorize_regions
 from .push import PushClient, log_to_terminal
 
 
+def guesswhat_rewrite_enabled() -> bool:
+    """The reference computes the GuessWhat ``q:/a:`` rewrite and then
+    discards it (worker.py:391-402 overwrites ``tokens`` with the raw
+    query), so its OBSERVABLE behavior feeds the raw query to the model.
+    Default here is reference-exact (raw query); set
+    ``VILBERT_GUESSWHAT_REWRITE=1`` to apply the rewrite the reference
+    clearly intended (opt-in fix, both behaviors tested)."""
+    import os
+
+    return os.environ.get("VILBERT_GUESSWHAT_REWRITE", "0") == "1"
+
+
 def guesswhat_dialog_rewrite(query: str) -> str:
     """Rewrite a GuessWhat dialog ``q: ..? a: ..`` into the
     ``start <q> answer <a> stop`` token stream the model expects.
 
-    The reference builds exactly this string (worker.py:391-400) but then
-    overwrites it with the raw query (worker.py:402) — dead code / latent
-    bug flagged in SURVEY.md §2.1. We apply the rewrite for real (a
-    deliberate, documented divergence). Queries without a ``q:`` marker
-    (e.g. the frontend let a plain question through) pass unchanged, which
-    also matches the reference's effective behavior for those.
+    This is the string the reference builds (worker.py:391-400) and then
+    accidentally overwrites with the raw query (worker.py:402) — dead code /
+    latent bug flagged in SURVEY.md §2.1. Applied only when
+    ``guesswhat_rewrite_enabled()`` (default: reference-exact, i.e. NOT
+    applied). Queries without a ``q:`` marker pass unchanged either way.
     """
     if "q:" not in query:
         return query
@@ -159,16 +170,16 @@ class ServingWorker:
         Text follows the reference serving contract (worker.py:402-414):
         wordpiece + [CLS]/[SEP], END-padded to 37 (the reference's comment
         claims front-padding but its code pads at the end — we mirror the
-        code). GuessWhat dialogs get the ``q:/a:`` -> ``start .. answer ..
-        stop`` rewrite the reference computes but accidentally discards
-        (worker.py:391-402 overwrites ``tokens``) — fixed deliberately here;
-        see ``guesswhat_dialog_rewrite``.
+        code). GuessWhat dialogs pass through RAW by default (the reference
+        computes the ``q:/a:`` rewrite but discards it — worker.py:391-402);
+        VILBERT_GUESSWHAT_REWRITE=1 opts into the intended rewrite.
         """
         q_rows, mask_rows, seg_rows, task_rows = [], [], [], []
         infos_all: List[Dict] = []
+        rewrite_gw = guesswhat_rewrite_enabled()
         for r in reqs:
             text = r.question
-            if r.task_id == 16:
+            if r.task_id == 16 and rewrite_gw:
                 text = guesswhat_dialog_rewrite(text)
             cached = self._tok_cache.get(text)
             if cached is None:
@@ -201,9 +212,8 @@ class ServingWorker:
         spec = get_task(r.task_id)
         row = r.row_start
         if spec.decode == DecodeFamily.RETRIEVAL:
-            names = [p.split("/")[-1] for p in r.image_paths]
             return decode_retrieval(
-                r.task_id, outputs, range(row, row + r.num_rows), names
+                r.task_id, outputs, range(row, row + r.num_rows), r.image_paths
             )
         if spec.decode == DecodeFamily.GROUNDING:
             info = r.infos[0]
@@ -211,9 +221,11 @@ class ServingWorker:
                 r.task_id, outputs, row, batch["spatials"],
                 info["image_width"], info["image_height"],
             )
-            imgs = self.render_grounding_images(r.image_paths[0], result["boxes"])
-            if imgs:
-                result["image_name_list"] = imgs
+            # worker.py:591-600 contract: image_name_list carries BARE uuid
+            # strings; files land at media/refer_expressions_task/<uuid>.jpg.
+            result["image_name_list"] = self.render_grounding_images(
+                r.image_paths[0], result.pop("boxes")
+            )
             return result
         if spec.decode == DecodeFamily.BINARY:
             return decode_answer_task(
@@ -241,9 +253,6 @@ class ServingWorker:
             )
         for r, qa in zip(reqs, qa_ids):
             r.qa_id = qa
-            log_to_terminal(
-                self.push, r.socket_id, {"terminal": "Processing request..."}
-            )
         try:
             with trace.stage("build_batch"):
                 batch = self.build_batch(reqs)
@@ -274,8 +283,9 @@ class ServingWorker:
                     answers.append((r.qa_id, payload))
                     log_to_terminal(self.push, r.socket_id, {"terminal": payload})
                     log_to_terminal(self.push, r.socket_id, {"result": payload})
+                    # exact reference completion string (worker.py:649)
                     log_to_terminal(
-                        self.push, r.socket_id, {"terminal": "Completed VilBERT task"}
+                        self.push, r.socket_id, {"terminal": "Completed Task"}
                     )
                     acks.append(r.delivery.msg_id)
                     m.requests_total.labels(str(r.task_id), "ok").inc()
@@ -298,37 +308,45 @@ class ServingWorker:
         )
         return served
 
+    # per-box outline colors: the reference's cv2 BGR list
+    # [(0,0,255),(0,255,0),(255,0,0)] (worker.py:589) = red/green/blue in RGB
+    _BOX_COLORS = [(255, 0, 0), (0, 255, 0), (0, 0, 255)]
+
     def render_grounding_images(self, image_path: str, boxes) -> List[str]:
         """Draw top-k grounding boxes into result JPEGs
-        (worker.py:591-600 contract: one image per box under
-        media/refer_expressions_task/, PIL instead of cv2)."""
+        (worker.py:591-600 contract: one image per box at
+        media/refer_expressions_task/<uuid>.jpg; the returned names are the
+        BARE uuid strings the reference puts in image_name_list; PIL instead
+        of cv2, line width 4 like the reference). A missing source image
+        still yields the uuid name list (the file write is skipped) so the
+        wire contract shape is independent of disk state."""
         import os
 
+        names = [str(uuid.uuid4()) for _ in boxes]
         if not os.path.exists(image_path):
-            return []
+            return names
         try:
             from PIL import Image, ImageDraw
 
             out_dir = os.path.join("media", "refer_expressions_task")
             os.makedirs(out_dir, exist_ok=True)
-            names = []
             base = Image.open(image_path).convert("RGB")
-            for box in boxes:
+            for name, box, color in zip(names, boxes, self._BOX_COLORS):
                 img = base.copy()
                 ImageDraw.Draw(img).rectangle(
-                    [box[0], box[1], box[2], box[3]], outline=(255, 0, 0), width=3
+                    [box[0], box[1], box[2], box[3]], outline=color, width=4
                 )
-                name = f"{uuid.uuid4().hex}.jpg"
-                img.save(os.path.join(out_dir, name), "JPEG")
-                names.append(os.path.join(out_dir, name))
-            return names
+                img.save(os.path.join(out_dir, f"{name}.jpg"), "JPEG")
         except Exception:
             traceback.print_exc()
-            return []
+        return names
 
     def run_forever(self, poll_s: float = 0.02) -> None:
         """Blocking consume loop; SIGTERM/SIGINT finish the in-flight batch
-        then exit (unacked messages redeliver — at-least-once)."""
+        then exit (unacked messages redeliver — at-least-once). Transient
+        errors (e.g. sqlite contention under competing consumers) are logged
+        and retried instead of killing the worker process — leased messages
+        redeliver after the lease timeout either way."""
         import signal
 
         stop = {"flag": False}
@@ -342,5 +360,11 @@ class ServingWorker:
         except ValueError:
             pass  # not the main thread
         while not stop["flag"]:
-            if self.process_once(max_wait_s=poll_s) == 0:
+            try:
+                served = self.process_once(max_wait_s=poll_s)
+            except Exception:
+                traceback.print_exc()
+                served = 0
+                time.sleep(poll_s * 5)
+            if served == 0:
                 time.sleep(poll_s)

@@ -21,6 +21,13 @@ from urllib.parse import unquote, urlsplit
 
 _GUID = b"258EAFA5-E914-47DA-95CA-C5AB0DC85B11"
 
+# DoS bounds: a client frame/assembled message (or the pre-handshake header
+# block) above this is rejected with close 1009 (message too big) instead of
+# growing self.buf/_frag_data without limit. The demo's messages are small
+# JSON; 1 MiB is generous.
+MAX_MESSAGE_BYTES = 1 << 20
+MAX_HANDSHAKE_BYTES = 64 << 10
+
 
 def _accept_key(key: str) -> str:
     return base64.b64encode(hashlib.sha1(key.encode() + _GUID).digest()).decode()
@@ -65,10 +72,27 @@ class MinimalWebSocketProtocol(asyncio.Protocol):
         self.buf += data
         if not self.handshaken:
             if b"\r\n\r\n" not in self.buf:
+                if len(self.buf) > MAX_HANDSHAKE_BYTES:
+                    self.transport.write(
+                        b"HTTP/1.1 431 Request Header Fields Too Large\r\n"
+                        b"content-length: 0\r\n\r\n"
+                    )
+                    self.transport.close()
                 return
             head, self.buf = self.buf.split(b"\r\n\r\n", 1)
             self._handshake(head)
         self._parse_frames()
+
+    def _protocol_error(self, code: int) -> None:
+        """Close the connection with the given code (1009 too-big /
+        1007 invalid-utf8) and drop buffered input."""
+        if not self.closed:
+            self._send_close(code)  # before the closed flag gates _send_frame
+            self.closed = True
+            self.recv_q.put_nowait({"type": "websocket.disconnect", "code": code})
+        self.buf = b""
+        self._frag_op, self._frag_data = 0, b""
+        self.transport.close()
 
     # ---- handshake --------------------------------------------------------
     def _handshake(self, head: bytes) -> None:
@@ -189,6 +213,9 @@ class MinimalWebSocketProtocol(asyncio.Protocol):
                     return
                 ln = int.from_bytes(self.buf[2:10], "big")
                 off = 10
+            if ln > MAX_MESSAGE_BYTES:
+                self._protocol_error(1009)
+                return
             mask = b""
             if masked:
                 if len(self.buf) < off + 4:
@@ -205,6 +232,9 @@ class MinimalWebSocketProtocol(asyncio.Protocol):
 
     def _on_frame(self, fin: int, opcode: int, payload: bytes) -> None:
         if opcode == 0x0:  # continuation
+            if len(self._frag_data) + len(payload) > MAX_MESSAGE_BYTES:
+                self._protocol_error(1009)  # endless-continuation DoS guard
+                return
             self._frag_data += payload
             if fin:
                 opcode, payload = self._frag_op, self._frag_data
@@ -215,14 +245,19 @@ class MinimalWebSocketProtocol(asyncio.Protocol):
             self._frag_op, self._frag_data = opcode, payload
             return
         if opcode == 0x1:
-            self.recv_q.put_nowait({"type": "websocket.receive", "text": payload.decode()})
+            try:
+                text = payload.decode()
+            except UnicodeDecodeError:
+                self._protocol_error(1007)  # RFC6455: invalid UTF-8 in text
+                return
+            self.recv_q.put_nowait({"type": "websocket.receive", "text": text})
         elif opcode == 0x2:
             self.recv_q.put_nowait({"type": "websocket.receive", "bytes": payload})
         elif opcode == 0x8:
             code = int.from_bytes(payload[:2], "big") if len(payload) >= 2 else 1000
             if not self.closed:
+                self._send_close(code)  # echo close before gating _send_frame
                 self.closed = True
-                self._send_close(code)
                 self.recv_q.put_nowait({"type": "websocket.disconnect", "code": code})
             self.transport.close()
         elif opcode == 0x9:
