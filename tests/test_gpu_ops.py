@@ -343,3 +343,107 @@ def test_ln_train_bwd_deterministic():
         grads.append((wg.grad.clone(), bg.grad.clone()))
     assert torch.equal(grads[0][0], grads[1][0])
     assert torch.equal(grads[0][1], grads[1][1])
+
+
+# ---------------------------------------------------------------------------
+# v3 bh-loop prefetch kernel (round-2): parity at every serving shape family
+# ---------------------------------------------------------------------------
+
+@pytest.mark.parametrize(
+    "B,H,Lq,Lk,D,mask_mode",
+    [
+        (3, 12, 38, 38, 64, 1),     # text self
+        (2, 8, 101, 101, 128, 1),   # vision self
+        (2, 8, 38, 101, 128, 1),    # co-attn text->vision
+        (96, 12, 38, 38, 64, 1),    # BH=1152: auto path, >=2 grid-stride iters
+        (24, 8, 101, 101, 128, 1),  # BH=192 forced: ragged iters w/ prefetch
+        (5, 8, 20, 33, 128, 0),     # no mask, ragged lengths
+    ],
+)
+def test_attention_bhloop_vs_reference(ext, monkeypatch, B, H, Lq, Lk, D, mask_mode):
+    monkeypatch.setenv("VILBERT_ATTN_BHLOOP", "1")
+    q = _rand_bf16(B, Lq, H * D, seed=B * 101 + Lq)
+    k = _rand_bf16(B, Lk, H * D, seed=B * 101 + Lk + 1)
+    v = _rand_bf16(B, Lk, H * D, seed=B * 101 + Lk + 2)
+    if mask_mode == 1:
+        keep = torch.ones(B, Lk)
+        keep[:, Lk - 5:] = 0
+        mask = ((1 - keep) * -1e9).to(torch.bfloat16).cuda().view(B, 1, 1, Lk)
+        ref_mask = mask.float()
+    else:
+        mask = ref_mask = None
+    out = torch.ops.vilbert_amd.attention(q, k, v, H, mask)
+    ref = _attn_ref(q, k, v, H, ref_mask)
+    err = (out.float() - ref).abs().max().item()
+    assert err < 4e-2, f"max err {err}"
+
+
+def test_attention_bhloop_matches_base_kernel(ext, monkeypatch):
+    """The bh-loop path must agree with the per-bh kernel to rounding (same
+    MFMA order and swizzles; only the mask-add association differs)."""
+    B, H, Lq, Lk, D = 16, 8, 101, 101, 128
+    q = _rand_bf16(B, Lq, H * D, seed=900)
+    k = _rand_bf16(B, Lk, H * D, seed=901)
+    v = _rand_bf16(B, Lk, H * D, seed=902)
+    keep = torch.ones(B, Lk)
+    keep[:, 90:] = 0
+    mask = ((1 - keep) * -1e9).to(torch.bfloat16).cuda().view(B, 1, 1, Lk)
+    monkeypatch.setenv("VILBERT_ATTN_BHLOOP", "0")
+    base = torch.ops.vilbert_amd.attention(q, k, v, H, mask)
+    monkeypatch.setenv("VILBERT_ATTN_BHLOOP", "1")
+    loop = torch.ops.vilbert_amd.attention(q, k, v, H, mask)
+    assert (base.float() - loop.float()).abs().max().item() < 2e-3
+
+
+# ---------------------------------------------------------------------------
+# attention-prob export (output_all_attention_masks=True, worker.py:288)
+# ---------------------------------------------------------------------------
+
+@pytest.mark.parametrize(
+    "B,H,Lq,Lk,D,force_bhloop",
+    [
+        (3, 12, 38, 38, 64, False),
+        (2, 8, 101, 101, 128, False),
+        (24, 8, 101, 101, 128, True),   # prob export through the bh-loop path
+    ],
+)
+def test_attention_probs_vs_reference(ext, monkeypatch, B, H, Lq, Lk, D, force_bhloop):
+    monkeypatch.setenv("VILBERT_ATTN_BHLOOP", "1" if force_bhloop else "0")
+    q = _rand_bf16(B, Lq, H * D, seed=700 + B)
+    k = _rand_bf16(B, Lk, H * D, seed=701 + B)
+    v = _rand_bf16(B, Lk, H * D, seed=702 + B)
+    keep = torch.ones(B, Lk)
+    keep[:, Lk - 5:] = 0
+    mask = ((1 - keep) * -1e9).to(torch.bfloat16).cuda().view(B, 1, 1, Lk)
+    out, probs = torch.ops.vilbert_amd.attention_probs(q, k, v, H, mask)
+    assert probs.shape == (B, H, Lq, Lk)
+    # fp32 reference probs
+    d = (H * D) // H
+    qh = q.float().view(B, Lq, H, D).transpose(1, 2)
+    kh = k.float().view(B, Lk, H, D).transpose(1, 2)
+    vh = v.float().view(B, Lk, H, D).transpose(1, 2)
+    s = qh @ kh.transpose(-1, -2) / math.sqrt(D) + mask.float()
+    p_ref = torch.softmax(s, dim=-1)
+    ref = (p_ref @ vh).transpose(1, 2).reshape(B, Lq, H * D)
+    assert (out.float() - ref).abs().max().item() < 4e-2
+    assert (probs.float() - p_ref).abs().max().item() < 2e-2
+    # masked tail keys carry zero probability
+    assert probs.float()[..., Lk - 5:].abs().max().item() == 0.0
+
+
+def test_functional_need_probs_uses_hip(ext):
+    """functional.attention(need_probs=True) must run the HIP prob-export
+    kernel (round 1 silently fell back to torch math — VERDICT item 7)."""
+    from vilbert_multi_task_amd.ops import functional as F_ops
+
+    B, H, L, D = 2, 8, 101, 128
+    q = _rand_bf16(B, L, H * D, seed=30)
+    k = _rand_bf16(B, L, H * D, seed=31)
+    v = _rand_bf16(B, L, H * D, seed=32)
+    ctx, probs = F_ops.attention(q, k, v, H, None, need_probs=True)
+    assert probs is not None and probs.shape == (B, H, L, L)
+    ctx_ref, probs_ref = F_ops.attention(
+        q.float().cpu(), k.float().cpu(), v.float().cpu(), H, None, need_probs=True
+    )
+    assert (ctx.float().cpu() - ctx_ref).abs().max().item() < 4e-2
+    assert (probs.float().cpu() - probs_ref).abs().max().item() < 2e-2

@@ -74,7 +74,8 @@ __global__ __launch_bounds__(THREADS) void attn_kernel(
     float scale, int qs, int ks, int vs, int nsplit,
     unsigned char* __restrict__ out8 = nullptr,
     const float* __restrict__ fp8_scales = nullptr,
-    float* __restrict__ fp8_amaxes = nullptr, int fp8_site = 0) {
+    float* __restrict__ fp8_amaxes = nullptr, int fp8_site = 0,
+    bf16* __restrict__ probs_out = nullptr) {
   const float fp8_inv = FP8OUT ? 1.0f / fp8_scales[fp8_site] : 0.f;
   float fp8_amax = 0.f;
   // qs/ks/vs: row strides (elems) of q/k/v — [B,L,H*D] views into a fused
@@ -251,6 +252,25 @@ __global__ __launch_bounds__(THREADS) void attn_kernel(
     }
     // per-wave P buffer: same wave writes then reads (compiler inserts the
     // lgkm waits through the address dependence); no cross-wave sharing.
+
+    // ---- optional attention-prob export (output_all_attention_masks=True,
+    // worker.py:288): normalized softmax rows to [B,H,Lq,Lk] bf16 ----------
+    if (probs_out != nullptr) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int row = qrow0 + (lane >> 4) * 4 + r;
+        if (row < Lq) {
+          const long pbase = ((long)bh * Lq + row) * Lk;
+#pragma unroll
+          for (int nt = 0; nt < NTMAX; ++nt) {
+            if (nt >= NT) break;
+            const int col = nt * 16 + col0;
+            if (col < Lk)
+              probs_out[pbase + col] = f2bf(acc_s[nt][r] * inv_l[r]);
+          }
+        }
+      }
+    }
 
     // ---- O = P V (V fragments via hardware transpose reads) ---------------
     f32x4 acc_o[D / 16];
@@ -532,6 +552,263 @@ __global__ __launch_bounds__(256) void attn_kernel_pipe(
 }
 
 // ---------------------------------------------------------------------------
+// v3: bh-loop with cross-iteration K/V prefetch (round-2 kernel).
+//
+// Why: PMC (profiles/r06) shows the per-(b,h) kernel at 13% MFMA-busy —
+// staging-latency bound, because every workgroup cold-starts: ALL K+V HBM
+// loads must land and be stored to LDS before its first MFMA. At B=1024 the
+// op moves ~850 MB (HBM roofline ~135 us at 6.3 TB/s) but measured 315 us
+// (43% of roofline): the fix is hiding the staging latency, not more MFMA.
+//
+// Structure (guide T14 async-STAGE split, applied ACROSS bh iterations):
+// each workgroup walks bh pairs grid-stride with DOUBLE-BUFFERED K/V LDS.
+// Per iteration:
+//   1. preload this bh's Q rows + mask values into registers (oldest loads,
+//      so the compute waits on vmcnt leave the prefetch in flight)
+//   2. issue the NEXT bh's K/V global loads into registers (in flight
+//      through the whole compute phase — ~10k cycles of MFMA/softmax cover
+//      the ~1k-cycle HBM latency and the staging bandwidth)
+//   3. compute all stripes from LDS[cur]
+//   4. barrier; write the prefetched K (swizzled) / V (transposed) tiles to
+//      LDS[cur^1]; barrier.
+// All staging loads are ordinary register loads (NOT global_load_lds): the
+// guide's mixed-kind trap ("beside glds hipcc waits vmcnt(0) for any
+// ordinary load's use") would drain the prefetch at the first Q use.
+// ---------------------------------------------------------------------------
+template <int D, int NTMAX, int THREADS = 256>
+__global__ __launch_bounds__(THREADS) void attn_bhloop_kernel(
+    const bf16* __restrict__ q, const bf16* __restrict__ kg,
+    const bf16* __restrict__ vg, const bf16* __restrict__ mask,
+    bf16* __restrict__ out, int BH, int H, int Lq, int Lk, int mask_mode,
+    float scale, int qs, int ks, int vs,
+    bf16* __restrict__ probs_out = nullptr) {
+  constexpr int KCH = D / 8;      // 16B chunks per row
+  constexpr int NW = THREADS / WAVE;
+  constexpr int SMAX = 2;         // stripes per wave (Lq <= 16*NW*SMAX = 128)
+  const int HD = H * D;
+  const int tid = threadIdx.x;
+  const int lane = lane_id();
+  const int wid = wave_id();
+
+  const int LK_PAD = (Lk + 31) & ~31;
+  const int NT = LK_PAD / 16;
+  const int kvbytes = LK_PAD * D * 2;
+
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  // [K0 | V0 | K1 | V1 | P(per wave)]
+  char* P_base = smem + 4 * kvbytes;
+
+  // staging thread roles (same decomposition as v2)
+  const int rows_per_pass = THREADS / KCH;
+  const int r0 = tid / KCH;
+  const int c = tid % KCH;
+  const int npass = (LK_PAD + rows_per_pass - 1) / rows_per_pass;  // <= 8
+
+  uint4 kr[8], vr[8];
+  auto issue_loads = [&](int bh) {
+    const int b = bh / H, h = bh % H;
+    const long kbase0 = (long)b * Lk * ks + (long)h * D;
+    const long vbase0 = (long)b * Lk * vs + (long)h * D;
+#pragma unroll
+    for (int pi = 0; pi < 8; ++pi) {
+      if (pi >= npass) break;
+      const int r = r0 + pi * rows_per_pass;
+      kr[pi] = make_uint4(0, 0, 0, 0);
+      vr[pi] = make_uint4(0, 0, 0, 0);
+      if (r < Lk) {
+        kr[pi] = *reinterpret_cast<const uint4*>(kg + kbase0 + (long)r * ks + c * 8);
+        vr[pi] = *reinterpret_cast<const uint4*>(vg + vbase0 + (long)r * vs + c * 8);
+      }
+    }
+  };
+  auto write_tiles = [&](int buf) {
+    char* K_lds = smem + 2 * buf * kvbytes;
+    char* V_lds = K_lds + kvbytes;
+#pragma unroll
+    for (int pi = 0; pi < 8; ++pi) {
+      if (pi >= npass) break;
+      const int r = r0 + pi * rows_per_pass;
+      if (r >= LK_PAD) break;
+      lds_store_b128(K_lds + r * (D * 2) + ((c * 16) ^ SWZ(r)), kr[pi]);
+      union { uint4 u; short s[8]; } vv;
+      vv.u = vr[pi];
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const int d = c * 8 + j;
+        *reinterpret_cast<short*>(
+            V_lds + d * (LK_PAD * 2) + ((r * 2) ^ SWZ(d))) = vv.s[j];
+      }
+    }
+  };
+
+  // prologue: stage the first bh synchronously
+  if (blockIdx.x >= (unsigned)BH) return;  // grid never exceeds BH (launcher)
+  issue_loads(blockIdx.x);
+  write_tiles(0);
+  __syncthreads();
+
+  const int nstripes = (Lq + 15) / 16;
+  const int col0 = lane & 15;
+
+  for (int bh = blockIdx.x, it = 0; bh < BH; bh += gridDim.x, ++it) {
+    const int buf = it & 1;
+    const char* K_lds = smem + 2 * buf * kvbytes;
+    const char* V_lds = K_lds + kvbytes;
+    char* P_lds = P_base + wid * 16 * LK_PAD * 2;
+    const int b = bh / H;
+
+    // ---- 1. preload Q + mask for this wave's stripes (oldest loads) ------
+    bf16x8 aq[SMAX][D / 32];
+#pragma unroll
+    for (int si = 0; si < SMAX; ++si) {
+      const int s = wid + si * NW;
+      if (s >= nstripes) break;
+      const int row = min(s * 16 + (lane & 15), Lq - 1);
+      const long qoff = ((long)b * Lq + row) * qs + (long)(bh % H) * D + (lane >> 4) * 8;
+#pragma unroll
+      for (int kk = 0; kk < D / 32; ++kk) aq[si][kk] = load_bf16x8(q + qoff + kk * 32);
+    }
+    // mode-1 mask values depend only on (b, col): one register set serves
+    // every stripe and row (v2 reloaded them 4x per stripe inside softmax)
+    float mv[NTMAX];
+#pragma unroll
+    for (int nt = 0; nt < NTMAX; ++nt) {
+      mv[nt] = 0.f;
+      if (mask_mode == 1 && nt < NT) {
+        const int col = nt * 16 + col0;
+        if (col < Lk) mv[nt] = bf2f(mask[(long)b * Lk + col]);
+      }
+    }
+
+    // ---- 2. prefetch the NEXT bh's K/V (stays in flight through compute) -
+    const int nbh = bh + gridDim.x;
+    if (nbh < BH) issue_loads(nbh);
+
+    // ---- 3. stripes from LDS[buf] ----------------------------------------
+#pragma unroll
+    for (int si = 0; si < SMAX; ++si) {
+      const int s = wid + si * NW;
+      if (s >= nstripes) break;
+      const int qrow0 = s * 16;
+
+      f32x4 acc_s[NTMAX];
+#pragma unroll
+      for (int nt = 0; nt < NTMAX; ++nt) acc_s[nt] = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+      for (int nt = 0; nt < NTMAX; ++nt) {
+        if (nt >= NT) break;
+        const int key = nt * 16 + (lane & 15);
+        const char* kbase = K_lds + key * (D * 2);
+        const int ksw = SWZ(key);
+#pragma unroll
+        for (int kk = 0; kk < D / 32; ++kk) {
+          const bf16x8 bk = lds_b128(kbase + (((kk * 64) + ((lane >> 4) * 16)) ^ ksw));
+          acc_s[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              (bf16x8)aq[si][kk], bk, acc_s[nt], 0, 0, 0);
+        }
+      }
+
+      float inv_l[4];
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int row = qrow0 + (lane >> 4) * 4 + r;
+        float mx = -3.0e38f;
+#pragma unroll
+        for (int nt = 0; nt < NTMAX; ++nt) {
+          if (nt >= NT) break;
+          const int col = nt * 16 + col0;
+          float sv = acc_s[nt][r] * scale + mv[nt];
+          if (col < Lk) {
+            acc_s[nt][r] = sv;
+            mx = fmaxf(mx, sv);
+          } else {
+            acc_s[nt][r] = -3.0e38f;
+          }
+        }
+        const float mrow = group16_max(mx);
+        float sum = 0.f;
+#pragma unroll
+        for (int nt = 0; nt < NTMAX; ++nt) {
+          if (nt >= NT) break;
+          const int col = nt * 16 + col0;
+          const float p = (col < Lk) ? __expf(acc_s[nt][r] - mrow) : 0.f;
+          acc_s[nt][r] = p;
+          sum += p;
+        }
+        inv_l[r] = 1.0f / group16_sum(sum);
+      }
+
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int prow = (lane >> 4) * 4 + r;
+        char* prow_base = P_lds + prow * (LK_PAD * 2);
+        const int psw = SWZ(prow);
+#pragma unroll
+        for (int nt = 0; nt < NTMAX; ++nt) {
+          if (nt >= NT) break;
+          const int col = nt * 16 + col0;
+          *reinterpret_cast<short*>(prow_base + ((col * 2) ^ psw)) =
+              (short)f2us(acc_s[nt][r]);
+        }
+      }
+      if (probs_out != nullptr) {
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const int row = qrow0 + (lane >> 4) * 4 + r;
+          if (row < Lq) {
+            const long pbase = ((long)bh * Lq + row) * Lk;
+#pragma unroll
+            for (int nt = 0; nt < NTMAX; ++nt) {
+              if (nt >= NT) break;
+              const int col = nt * 16 + col0;
+              if (col < Lk)
+                probs_out[pbase + col] = f2bf(acc_s[nt][r] * inv_l[r]);
+            }
+          }
+        }
+      }
+
+      f32x4 acc_o[D / 16];
+#pragma unroll
+      for (int nt = 0; nt < D / 16; ++nt) acc_o[nt] = {0.f, 0.f, 0.f, 0.f};
+      const int parow = lane & 15;
+      const char* pa_base = P_lds + parow * (LK_PAD * 2);
+      const int pasw = SWZ(parow);
+#pragma unroll
+      for (int kk = 0; kk < NTMAX / 2; ++kk) {
+        if (kk * 32 >= LK_PAD) break;
+        const bf16x8 ap =
+            lds_b128(pa_base + (((kk * 64) + ((lane >> 4) * 16)) ^ pasw));
+        const int keyoff = (kk * 64) + ((lane >> 4) * 16);
+#pragma unroll
+        for (int nt = 0; nt < D / 16; ++nt) {
+          const int d = nt * 16 + (lane & 15);
+          const bf16x8 bv =
+              lds_b128(V_lds + d * (LK_PAD * 2) + (keyoff ^ SWZ(d)));
+          acc_o[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ap, bv, acc_o[nt], 0, 0, 0);
+        }
+      }
+
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int row = qrow0 + (lane >> 4) * 4 + r;
+        if (row < Lq) {
+          const long obase = ((long)b * Lq + row) * HD + (long)(bh % H) * D;
+#pragma unroll
+          for (int nt = 0; nt < D / 16; ++nt)
+            out[obase + nt * 16 + col0] = f2bf(acc_o[nt][r] * inv_l[r]);
+        }
+      }
+    }
+
+    // ---- 4. publish the prefetched tiles into the other buffer -----------
+    __syncthreads();  // all waves done reading LDS[buf] and P
+    if (nbh < BH) write_tiles(buf ^ 1);
+    __syncthreads();
+  }
+}
+
+// ---------------------------------------------------------------------------
 // mfma layout probe: one-wave 16x16x32 product for on-device layout checks
 // ---------------------------------------------------------------------------
 __global__ void mfma_probe_kernel(const bf16* __restrict__ a,
@@ -577,7 +854,8 @@ void launch_attention_impl(const bf16* q, const bf16* k, const bf16* v,
                            int Lk, int D, int mask_mode, int qs, int ks,
                            int vs, unsigned char* out8,
                            const float* fp8_scales, float* fp8_amaxes,
-                           int fp8_site, hipStream_t stream) {
+                           int fp8_site, hipStream_t stream,
+                           bf16* probs_out = nullptr) {
   const float scale = 1.0f / sqrtf((float)D);
   const int LK_PAD = (Lk + 31) & ~31;
   static const int kglobal_env = [] {
@@ -598,7 +876,36 @@ void launch_attention_impl(const bf16* q, const bf16* k, const bf16* v,
     const char* e = getenv("VILBERT_ATTN_PIPE");
     return e ? atoi(e) : 0;  // 1 = V-staging pipelined behind stripe-0 S
   }();
+  // read dynamically (not static) so tests/A-B harnesses can flip it per
+  // call; getenv cost is noise next to a kernel launch
+  const char* bhloop_e = getenv("VILBERT_ATTN_BHLOOP");
+  const int bhloop_env = bhloop_e ? atoi(bhloop_e) : -1;  // -1 auto, 0 off, 1 force
   const int nwaves = (waves_env == 8 && D == 128 && !kglobal_env) ? 8 : 4;
+
+  // ---- v3 bh-loop prefetch dispatch (the round-2 default for large BH) ----
+  const int BH = B * H;
+  const bool bhloop_ok = mask_mode <= 1 && !kglobal_env && nsplit == 1 &&
+                         !pipe_env && nwaves == 4 && Lq <= 128 &&
+                         out8 == nullptr;
+  const bool use_bhloop =
+      bhloop_ok && (bhloop_env == 1 || (bhloop_env != 0 && BH >= 1024));
+  if (use_bhloop) {
+    // [K0|V0|K1|V1|P x 4 waves]
+    const size_t lds2 =
+        (size_t)4 * LK_PAD * D * 2 + (size_t)4 * 16 * LK_PAD * 2;
+    int nres = (int)(163840 / lds2);
+    nres = nres < 1 ? 1 : (nres > 8 ? 8 : nres);
+    const dim3 grid2(BH < nres * 256 ? BH : nres * 256);
+#define LAUNCH_BHLOOP(DD, NTM)                                               \
+    hipLaunchKernelGGL((attn_bhloop_kernel<DD, NTM>), grid2, dim3(256),      \
+                       lds2, stream, q, k, v, mask, out, BH, H, Lq, Lk,      \
+                       mask_mode, scale, qs, ks, vs, probs_out)
+    const bool small2 = LK_PAD <= 64;
+    if (D == 64) { if (small2) LAUNCH_BHLOOP(64, 4); else LAUNCH_BHLOOP(64, 8); }
+    else         { if (small2) LAUNCH_BHLOOP(128, 4); else LAUNCH_BHLOOP(128, 8); }
+#undef LAUNCH_BHLOOP
+    return;
+  }
   // (K if staged) + V + per-wave P (all bf16)
   const size_t lds = sizeof(bf16) *
       (size_t)((kglobal_env ? 1 : 2) * LK_PAD * D + nwaves * 16 * LK_PAD);
@@ -633,7 +940,8 @@ void launch_attention_impl(const bf16* q, const bf16* k, const bf16* v,
     else                                                                      \
       hipLaunchKernelGGL((attn_kernel<DD, KG, NTM, false>), grid, dim3(256),  \
                          lds, stream, q, k, v, mask, out, B, H, Lq, Lk,       \
-                         mask_mode, scale, qs, ks, vs, nsplit);               \
+                         mask_mode, scale, qs, ks, vs, nsplit, nullptr,       \
+                         nullptr, nullptr, 0, probs_out);                     \
   } while (0)
   const bool small = LK_PAD <= 64;  // NTMAX=4 halves the accumulator VGPRs
   if (D == 64) {
@@ -652,6 +960,14 @@ void launch_attention(const bf16* q, const bf16* k, const bf16* v,
                       hipStream_t stream) {
   launch_attention_impl(q, k, v, mask, out, B, H, Lq, Lk, D, mask_mode, qs, ks,
                         vs, nullptr, nullptr, nullptr, 0, stream);
+}
+
+void launch_attention_probs(const bf16* q, const bf16* k, const bf16* v,
+                            const bf16* mask, bf16* out, bf16* probs, int B,
+                            int H, int Lq, int Lk, int D, int mask_mode,
+                            int qs, int ks, int vs, hipStream_t stream) {
+  launch_attention_impl(q, k, v, mask, out, B, H, Lq, Lk, D, mask_mode, qs, ks,
+                        vs, nullptr, nullptr, nullptr, 0, stream, probs);
 }
 
 void launch_attention_fp8out(const bf16* q, const bf16* k, const bf16* v,

@@ -26,6 +26,9 @@ void launch_embedding_ln(const long*, const long*, const long*, const T*,
                          float, hipStream_t);
 void launch_attention(const bf16*, const bf16*, const bf16*, const bf16*, bf16*,
                       int, int, int, int, int, int, int, int, int, hipStream_t);
+void launch_attention_probs(const bf16*, const bf16*, const bf16*, const bf16*,
+                            bf16*, bf16*, int, int, int, int, int, int, int,
+                            int, int, hipStream_t);
 void launch_attention_fp8out(const bf16*, const bf16*, const bf16*, const bf16*,
                              bf16*, unsigned char*, const float*, float*, int,
                              int, int, int, int, int, int, int, int, int,
@@ -162,6 +165,41 @@ at::Tensor attention(const at::Tensor& q, const at::Tensor& k,
                    (bf16*)out.data_ptr(), B, H, Lq, Lk, D, mask_mode,
                    qs, kss, vss, cur_stream());
   return out;
+}
+
+std::tuple<at::Tensor, at::Tensor> attention_probs(
+    const at::Tensor& q, const at::Tensor& k, const at::Tensor& v,
+    int64_t heads, const c10::optional<at::Tensor>& mask) {
+  // attention-map export path (worker.py:288 output_all_attention_masks=True):
+  // same kernel, plus normalized softmax rows written to [B,H,Lq,Lk] bf16
+  TORCH_CHECK(q.is_cuda() && q.scalar_type() == at::kBFloat16);
+  TORCH_CHECK(q.dim() == 3 && k.dim() == 3 && v.dim() == 3);
+  const int B = (int)q.size(0), Lq = (int)q.size(1), HD = (int)q.size(2);
+  const int Lk = (int)k.size(1);
+  const int H = (int)heads;
+  const int D = HD / H;
+  TORCH_CHECK(HD % H == 0 && (D == 64 || D == 128));
+  TORCH_CHECK(Lq <= 128 && Lk <= 128);
+  const int qs = row_stride_of(q, "q"), kss = row_stride_of(k, "k"),
+            vss = row_stride_of(v, "v");
+  int mask_mode = 0;
+  const bf16* mptr = nullptr;
+  if (mask.has_value() && mask->defined()) {
+    TORCH_CHECK(mask->scalar_type() == at::kBFloat16 && mask->is_contiguous());
+    const long mn = mask->numel();
+    if (mn == (long)B * Lk) mask_mode = 1;
+    else if (mn == (long)B * Lq * Lk) mask_mode = 2;
+    else TORCH_CHECK(false, "attention_probs: bad mask shape");
+    mptr = (const bf16*)mask->data_ptr();
+  }
+  auto out = at::empty({q.size(0), q.size(1), q.size(2)}, q.options());
+  auto probs = at::empty({(long)B, (long)H, (long)Lq, (long)Lk}, q.options());
+  launch_attention_probs(
+      (const bf16*)q.data_ptr(), (const bf16*)k.data_ptr(),
+      (const bf16*)v.data_ptr(), mptr, (bf16*)out.data_ptr(),
+      (bf16*)probs.data_ptr(), B, H, Lq, Lk, D, mask_mode, qs, kss, vss,
+      cur_stream());
+  return {out, probs};
 }
 
 std::tuple<at::Tensor, at::Tensor> attention_fp8out(
@@ -547,6 +585,7 @@ TORCH_LIBRARY(vilbert_amd, m) {
   m.def("residual_layer_norm(Tensor x, Tensor? res, Tensor w, Tensor b, float eps) -> Tensor");
   m.def("bias_gelu(Tensor x, Tensor? bias) -> Tensor");
   m.def("attention(Tensor q, Tensor k, Tensor v, int heads, Tensor? mask) -> Tensor");
+  m.def("attention_probs(Tensor q, Tensor k, Tensor v, int heads, Tensor? mask) -> (Tensor, Tensor)");
   m.def("attention_fp8out(Tensor q, Tensor k, Tensor v, int heads, Tensor? mask, Tensor scales, Tensor(a!) amaxes, int site) -> (Tensor, Tensor)");
   m.def("embedding_ln(Tensor ids, Tensor pos, Tensor type, Tensor word_w, Tensor pos_w, Tensor type_w, Tensor ln_w, Tensor ln_b, float eps) -> Tensor");
   m.def("mfma_probe(Tensor a, Tensor b) -> Tensor");
@@ -571,6 +610,7 @@ TORCH_LIBRARY_IMPL(vilbert_amd, CUDA, m) {
   m.impl("residual_layer_norm", residual_layer_norm);
   m.impl("bias_gelu", bias_gelu);
   m.impl("attention", attention);
+  m.impl("attention_probs", attention_probs);
   m.impl("attention_fp8out", attention_fp8out);
   m.impl("embedding_ln", embedding_ln);
   m.impl("mfma_probe", mfma_probe);

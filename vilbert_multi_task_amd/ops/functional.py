@@ -246,7 +246,6 @@ def attention(
     (ctx [B,Lq,H*D], probs [B,H,Lq,Lk] or None)."""
     use_hip = (
         _want_hip(q, k, v)
-        and not need_probs
         and not (training and dropout_p > 0.0)
     )
     if use_hip:
@@ -256,6 +255,16 @@ def attention(
             mb = mb.contiguous()
         # q/k/v may be strided views into a fused QKV projection (dim-1
         # stride 3*HD); the kernel reads strides directly — no copies.
+        if need_probs:
+            # prob-emitting kernel variant (worker.py:288 passes
+            # output_all_attention_masks=True): normalized softmax rows
+            # written [B,H,Lq,Lk] during the P-store phase
+            out, probs = ext.attention_probs(q, k, v, num_heads, mb)
+            if fp8_out is not None:
+                from ..models.fp8 import attach_quant_pack
+
+                attach_quant_pack(out, fp8_out[0], fp8_out[1])
+            return out, probs
         if fp8_out is not None:
             ctx8, site = fp8_out
             out, out8 = ext.attention_fp8out(
