@@ -17,7 +17,9 @@ sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 import torch
 
-from vilbert_multi_task_amd.ops import functional as F_ops  # loads ext
+from vilbert_multi_task_amd.ops import functional as F_ops
+
+F_ops._load_extension()  # registers torch.ops.vilbert_amd
 
 SHAPES = [
     # (name, H, Lq, Lk, D)

@@ -72,10 +72,10 @@ def test_serving_worker_gpu_end_to_end(tmp_path):
     vilbert_task(broker, ["/a.jpg", "/b.jpg"], "both show dogs", 12, "g2")
     assert worker.process_once() == 2
     results = [json.loads(p["result"]) for s, p in push.messages if "result" in p]
-    assert {r["task_id"] for r in results} == {1, 12}
-    vqa = next(r for r in results if r["task_id"] == 1)
+    assert {r["task_id"] for r in results} == {"1", "12"}
+    vqa = next(r for r in results if r["task_id"] == "1")
     assert len(vqa["result"]) == 3
-    assert all(0 <= e["confidence"] <= 1 for e in vqa["result"])
+    assert all(0 <= e["confidence"] <= 100 for e in vqa["result"])
 
 
 def test_x152_detector_bf16_matches_fp32():
