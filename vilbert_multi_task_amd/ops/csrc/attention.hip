@@ -115,17 +115,16 @@ __global__ __launch_bounds__(THREADS) void attn_kernel(
     // store pass: one batch of HBM round trips instead of one per pass
     const int npass = (LK_PAD + rows_per_pass - 1) / rows_per_pass;  // <= 8
     uint4 kr[8], vr[8];
+    // unconditional clamped loads (no per-element r<Lk branch — §5 trap
+    // (c)); pad rows re-read row Lk-1: K pads are masked to -3e38 in the
+    // softmax and V pads multiply by P == 0
 #pragma unroll
     for (int pi = 0; pi < 8; ++pi) {
       if (pi >= npass) break;
-      const int r = r0 + pi * rows_per_pass;
-      kr[pi] = make_uint4(0, 0, 0, 0);
-      vr[pi] = make_uint4(0, 0, 0, 0);
-      if (r < Lk) {
-        if (!KGLOBAL)
-          kr[pi] = *reinterpret_cast<const uint4*>(kg + kbase0 + (long)r * ks + c * 8);
-        vr[pi] = *reinterpret_cast<const uint4*>(vg + vbase0 + (long)r * vs + c * 8);
-      }
+      const long r = min(r0 + pi * rows_per_pass, Lk - 1);
+      if (!KGLOBAL)
+        kr[pi] = *reinterpret_cast<const uint4*>(kg + kbase0 + r * ks + c * 8);
+      vr[pi] = *reinterpret_cast<const uint4*>(vg + vbase0 + r * vs + c * 8);
     }
 #pragma unroll
     for (int pi = 0; pi < 8; ++pi) {
@@ -201,35 +200,65 @@ __global__ __launch_bounds__(THREADS) void attn_kernel(
     }
 
     // ---- mask + softmax (rows live across the 16-lane group) -------------
+    // Mask values are preloaded UNCONDITIONALLY at clamped addresses, then
+    // folded into acc_s in one register pass. The previous per-element
+    // guarded loads (`if (col < Lk) sv += bf2f(mask[...])`) compiled to 32
+    // dependent load->vmcnt(0) round trips per stripe per wave (guide §5
+    // trap (c)) — the dominant stall of the whole kernel (r2 .s audit).
     const int col0 = lane & 15;
+    if (mask_mode == 1) {
+      const long mb = (long)b * Lk;
+      float mv1[NTMAX];
+#pragma unroll
+      for (int nt = 0; nt < NTMAX; ++nt)
+        mv1[nt] = bf2f(mask[mb + min(nt * 16 + col0, Lk - 1)]);
+#pragma unroll
+      for (int nt = 0; nt < NTMAX; ++nt) {
+        const int col = nt * 16 + col0;
+#pragma unroll
+        for (int r = 0; r < 4; ++r)
+          acc_s[nt][r] = (col < Lk) ? acc_s[nt][r] * scale + mv1[nt] : -3.0e38f;
+      }
+    } else if (mask_mode == 2) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int row = min(qrow0 + (lane >> 4) * 4 + r, Lq - 1);
+        const long mb = ((long)b * Lq + row) * Lk;
+        float mv2[NTMAX];
+#pragma unroll
+        for (int nt = 0; nt < NTMAX; ++nt)
+          mv2[nt] = bf2f(mask[mb + min(nt * 16 + col0, Lk - 1)]);
+#pragma unroll
+        for (int nt = 0; nt < NTMAX; ++nt) {
+          const int col = nt * 16 + col0;
+          acc_s[nt][r] = (col < Lk) ? acc_s[nt][r] * scale + mv2[nt] : -3.0e38f;
+        }
+      }
+    } else {
+#pragma unroll
+      for (int nt = 0; nt < NTMAX; ++nt) {
+        const int col = nt * 16 + col0;
+#pragma unroll
+        for (int r = 0; r < 4; ++r)
+          acc_s[nt][r] = (col < Lk) ? acc_s[nt][r] * scale : -3.0e38f;
+      }
+    }
     float inv_l[4];
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
-      const int row = qrow0 + (lane >> 4) * 4 + r;
       float mx = -3.0e38f;
 #pragma unroll
       for (int nt = 0; nt < NTMAX; ++nt) {
         if (nt >= NT) break;
-        const int col = nt * 16 + col0;
-        float sv = acc_s[nt][r] * scale;
-        if (col < Lk) {
-          if (mask_mode == 1)
-            sv += bf2f(mask[(long)b * Lk + col]);
-          else if (mask_mode == 2)
-            sv += bf2f(mask[((long)b * Lq + min(row, Lq - 1)) * Lk + col]);
-          acc_s[nt][r] = sv;
-          mx = fmaxf(mx, sv);
-        } else {
-          acc_s[nt][r] = -3.0e38f;
-        }
+        mx = fmaxf(mx, acc_s[nt][r]);
       }
       const float mrow = group16_max(mx);
       float sum = 0.f;
 #pragma unroll
       for (int nt = 0; nt < NTMAX; ++nt) {
         if (nt >= NT) break;
-        const int col = nt * 16 + col0;
-        const float p = (col < Lk) ? __expf(acc_s[nt][r] - mrow) : 0.f;
+        // pad cols hold -3e38: exp underflows to exactly 0, no guard needed
+        const float p = __expf(acc_s[nt][r] - mrow);
         acc_s[nt][r] = p;
         sum += p;
       }
@@ -428,33 +457,59 @@ __global__ __launch_bounds__(256) void attn_kernel_pipe(
             (bf16x8)aq[kk], bk, acc_s[nt], 0, 0, 0);
       }
     }
+    // unconditional clamped mask preload (see base kernel note: the guarded
+    // in-loop loads serialized into per-element vmcnt(0) round trips)
+    if (mask_mode == 1) {
+      const long mb = (long)b * Lk;
+      float mv1[NTMAX];
+#pragma unroll
+      for (int nt = 0; nt < NTMAX; ++nt)
+        mv1[nt] = bf2f(mask[mb + min(nt * 16 + col0, Lk - 1)]);
+#pragma unroll
+      for (int nt = 0; nt < NTMAX; ++nt) {
+        const int col = nt * 16 + col0;
+#pragma unroll
+        for (int r = 0; r < 4; ++r)
+          acc_s[nt][r] = (col < Lk) ? acc_s[nt][r] * scale + mv1[nt] : -3.0e38f;
+      }
+    } else if (mask_mode == 2) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int row = min(qrow0 + (lane >> 4) * 4 + r, Lq - 1);
+        const long mb = ((long)b * Lq + row) * Lk;
+        float mv2[NTMAX];
+#pragma unroll
+        for (int nt = 0; nt < NTMAX; ++nt)
+          mv2[nt] = bf2f(mask[mb + min(nt * 16 + col0, Lk - 1)]);
+#pragma unroll
+        for (int nt = 0; nt < NTMAX; ++nt) {
+          const int col = nt * 16 + col0;
+          acc_s[nt][r] = (col < Lk) ? acc_s[nt][r] * scale + mv2[nt] : -3.0e38f;
+        }
+      }
+    } else {
+#pragma unroll
+      for (int nt = 0; nt < NTMAX; ++nt) {
+        const int col = nt * 16 + col0;
+#pragma unroll
+        for (int r = 0; r < 4; ++r)
+          acc_s[nt][r] = (col < Lk) ? acc_s[nt][r] * scale : -3.0e38f;
+      }
+    }
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
-      const int row = qrow0 + (lane >> 4) * 4 + r;
       float mx = -3.0e38f;
 #pragma unroll
       for (int nt = 0; nt < NTMAX; ++nt) {
         if (nt >= NT) break;
-        const int col = nt * 16 + col0;
-        float sv = acc_s[nt][r] * scale;
-        if (col < Lk) {
-          if (mask_mode == 1)
-            sv += bf2f(mask[(long)b * Lk + col]);
-          else if (mask_mode == 2)
-            sv += bf2f(mask[((long)b * Lq + min(row, Lq - 1)) * Lk + col]);
-          acc_s[nt][r] = sv;
-          mx = fmaxf(mx, sv);
-        } else {
-          acc_s[nt][r] = -3.0e38f;
-        }
+        mx = fmaxf(mx, acc_s[nt][r]);
       }
       const float mrow = group16_max(mx);
       float sum = 0.f;
 #pragma unroll
       for (int nt = 0; nt < NTMAX; ++nt) {
         if (nt >= NT) break;
-        const int col = nt * 16 + col0;
-        const float p = (col < Lk) ? __expf(acc_s[nt][r] - mrow) : 0.f;
+        const float p = __expf(acc_s[nt][r] - mrow);
         acc_s[nt][r] = p;
         sum += p;
       }
@@ -584,7 +639,7 @@ __global__ __launch_bounds__(THREADS) void attn_bhloop_kernel(
     bf16* __restrict__ probs_out = nullptr) {
   constexpr int KCH = D / 8;      // 16B chunks per row
   constexpr int NW = THREADS / WAVE;
-  constexpr int SMAX = 2;         // stripes per wave (Lq <= 16*NW*SMAX = 128)
+  constexpr int SMAX = 512 / THREADS;  // stripes per wave (Lq <= 16*NW*SMAX = 128)
   const int HD = H * D;
   const int tid = threadIdx.x;
   const int lane = lane_id();
@@ -605,6 +660,11 @@ __global__ __launch_bounds__(THREADS) void attn_bhloop_kernel(
   const int npass = (LK_PAD + rows_per_pass - 1) / rows_per_pass;  // <= 8
 
   uint4 kr[8], vr[8];
+  // UNCONDITIONAL loads at clamped rows: a per-element `if (r < Lk)` guard
+  // makes hipcc branch around each load and wait per element (guide §5 trap
+  // (c)). Pad rows re-read row Lk-1: K pad rows never matter (their score
+  // cols are forced to -3e38) and V pad rows multiply by P == 0 — real data
+  // instead of zeros avoids 0*NaN from garbage reads without any branch.
   auto issue_loads = [&](int bh) {
     const int b = bh / H, h = bh % H;
     const long kbase0 = (long)b * Lk * ks + (long)h * D;
@@ -612,13 +672,9 @@ __global__ __launch_bounds__(THREADS) void attn_bhloop_kernel(
 #pragma unroll
     for (int pi = 0; pi < 8; ++pi) {
       if (pi >= npass) break;
-      const int r = r0 + pi * rows_per_pass;
-      kr[pi] = make_uint4(0, 0, 0, 0);
-      vr[pi] = make_uint4(0, 0, 0, 0);
-      if (r < Lk) {
-        kr[pi] = *reinterpret_cast<const uint4*>(kg + kbase0 + (long)r * ks + c * 8);
-        vr[pi] = *reinterpret_cast<const uint4*>(vg + vbase0 + (long)r * vs + c * 8);
-      }
+      const long r = min(r0 + pi * rows_per_pass, Lk - 1);
+      kr[pi] = *reinterpret_cast<const uint4*>(kg + kbase0 + r * ks + c * 8);
+      vr[pi] = *reinterpret_cast<const uint4*>(vg + vbase0 + r * vs + c * 8);
     }
   };
   auto write_tiles = [&](int buf) {
@@ -657,27 +713,30 @@ __global__ __launch_bounds__(THREADS) void attn_bhloop_kernel(
     char* P_lds = P_base + wid * 16 * LK_PAD * 2;
     const int b = bh / H;
 
-    // ---- 1. preload Q + mask for this wave's stripes (oldest loads) ------
+    // ---- 1. preload mask for this bh (oldest loads; one register set
+    // serves every stripe and row in mode 1 — v2 reloaded them 4x per
+    // stripe inside the softmax). Unconditional clamped loads (trap (c)).
+    float mv[NTMAX];
+    if (mask_mode == 1) {
+      const long mb = (long)b * Lk;
+#pragma unroll
+      for (int nt = 0; nt < NTMAX; ++nt)
+        mv[nt] = bf2f(mask[mb + min(nt * 16 + col0, Lk - 1)]);
+    } else {
+#pragma unroll
+      for (int nt = 0; nt < NTMAX; ++nt) mv[nt] = 0.f;
+    }
+    // Q fragments for this wave's stripes, loaded BEFORE the prefetch so
+    // the compute phase contains no global loads at all — a compute-phase
+    // load's wait would drain the (older, in-order-retired) prefetch queue
     bf16x8 aq[SMAX][D / 32];
 #pragma unroll
     for (int si = 0; si < SMAX; ++si) {
       const int s = wid + si * NW;
-      if (s >= nstripes) break;
-      const int row = min(s * 16 + (lane & 15), Lq - 1);
+      const int row = min(min(s, nstripes - 1) * 16 + (lane & 15), Lq - 1);
       const long qoff = ((long)b * Lq + row) * qs + (long)(bh % H) * D + (lane >> 4) * 8;
 #pragma unroll
       for (int kk = 0; kk < D / 32; ++kk) aq[si][kk] = load_bf16x8(q + qoff + kk * 32);
-    }
-    // mode-1 mask values depend only on (b, col): one register set serves
-    // every stripe and row (v2 reloaded them 4x per stripe inside softmax)
-    float mv[NTMAX];
-#pragma unroll
-    for (int nt = 0; nt < NTMAX; ++nt) {
-      mv[nt] = 0.f;
-      if (mask_mode == 1 && nt < NT) {
-        const int col = nt * 16 + col0;
-        if (col < Lk) mv[nt] = bf2f(mask[(long)b * Lk + col]);
-      }
     }
 
     // ---- 2. prefetch the NEXT bh's K/V (stays in flight through compute) -
@@ -708,30 +767,28 @@ __global__ __launch_bounds__(THREADS) void attn_bhloop_kernel(
         }
       }
 
+#pragma unroll
+      for (int nt = 0; nt < NTMAX; ++nt) {
+        const int col = nt * 16 + col0;
+#pragma unroll
+        for (int r = 0; r < 4; ++r)
+          acc_s[nt][r] = (col < Lk) ? acc_s[nt][r] * scale + mv[nt] : -3.0e38f;
+      }
       float inv_l[4];
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
-        const int row = qrow0 + (lane >> 4) * 4 + r;
         float mx = -3.0e38f;
 #pragma unroll
         for (int nt = 0; nt < NTMAX; ++nt) {
           if (nt >= NT) break;
-          const int col = nt * 16 + col0;
-          float sv = acc_s[nt][r] * scale + mv[nt];
-          if (col < Lk) {
-            acc_s[nt][r] = sv;
-            mx = fmaxf(mx, sv);
-          } else {
-            acc_s[nt][r] = -3.0e38f;
-          }
+          mx = fmaxf(mx, acc_s[nt][r]);
         }
         const float mrow = group16_max(mx);
         float sum = 0.f;
 #pragma unroll
         for (int nt = 0; nt < NTMAX; ++nt) {
           if (nt >= NT) break;
-          const int col = nt * 16 + col0;
-          const float p = (col < Lk) ? __expf(acc_s[nt][r] - mrow) : 0.f;
+          const float p = __expf(acc_s[nt][r] - mrow);
           acc_s[nt][r] = p;
           sum += p;
         }
@@ -887,17 +944,24 @@ void launch_attention_impl(const bf16* q, const bf16* k, const bf16* v,
   const bool bhloop_ok = mask_mode <= 1 && !kglobal_env && nsplit == 1 &&
                          !pipe_env && nwaves == 4 && Lq <= 128 &&
                          out8 == nullptr;
-  const bool use_bhloop =
-      bhloop_ok && (bhloop_env == 1 || (bhloop_env != 0 && BH >= 1024));
+  // measured (r2 A/B @B1024): v3 first cut 0.42-0.70x of v2 — default OFF
+  // until the prefetch actually pays; VILBERT_ATTN_BHLOOP=1 forces it for
+  // iteration. (v2's 2-6 co-resident WGs/CU already overlap staging with
+  // compute; v3's single-WG double-buffer must beat that to ship.)
+  const bool use_bhloop = bhloop_ok && bhloop_env == 1;
   if (use_bhloop) {
-    // [K0|V0|K1|V1|P x 4 waves]
+    // [K0|V0|K1|V1|P x 8 waves]
     const size_t lds2 =
-        (size_t)4 * LK_PAD * D * 2 + (size_t)4 * 16 * LK_PAD * 2;
+        (size_t)4 * LK_PAD * D * 2 + (size_t)8 * 16 * LK_PAD * 2;
+    // 512 threads (8 waves): halves the per-thread prefetch register
+    // footprint (the 256-thread variant held 64 staging VGPRs across the
+    // whole compute — the allocator pushed them to AGPRs, and the
+    // accvgpr copies forced per-load waits that killed the pipeline)
     int nres = (int)(163840 / lds2);
-    nres = nres < 1 ? 1 : (nres > 8 ? 8 : nres);
+    nres = nres < 1 ? 1 : (nres > 4 ? 4 : nres);  // 512-thread WGs: <=4/CU
     const dim3 grid2(BH < nres * 256 ? BH : nres * 256);
 #define LAUNCH_BHLOOP(DD, NTM)                                               \
-    hipLaunchKernelGGL((attn_bhloop_kernel<DD, NTM>), grid2, dim3(256),      \
+    hipLaunchKernelGGL((attn_bhloop_kernel<DD, NTM, 512>), grid2, dim3(512), \
                        lds2, stream, q, k, v, mask, out, BH, H, Lq, Lk,      \
                        mask_mode, scale, qs, ks, vs, probs_out)
     const bool small2 = LK_PAD <= 64;
