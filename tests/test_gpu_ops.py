@@ -447,3 +447,66 @@ def test_functional_need_probs_uses_hip(ext):
     )
     assert (ctx.float().cpu() - ctx_ref).abs().max().item() < 4e-2
     assert (probs.float().cpu() - probs_ref).abs().max().item() < 2e-2
+
+
+# ---------------------------------------------------------------------------
+# hand-written MFMA GEMM (gemm_mfma.hip): y = x @ W^T (+bias)(+res)(+GELU)
+# ---------------------------------------------------------------------------
+
+def _mfma_linear_ref(x, w, bias, res, gelu):
+    y = x.float() @ w.float().t()
+    if bias is not None:
+        y = y + bias.float()
+    if gelu:
+        y = torch.nn.functional.gelu(y)
+    if res is not None:
+        y = y + res.float()
+    return y
+
+
+@pytest.mark.parametrize(
+    "M,N,K,gelu,with_res",
+    [
+        (512, 768, 768, False, False),     # out-proj shape family
+        (512, 3072, 768, True, False),     # FFN1 + GELU
+        (512, 768, 3072, False, True),     # FFN2 + residual
+        (512, 1024, 1024, False, True),    # vision out-proj + residual
+        (37, 768, 768, False, False),      # ragged M < one tile
+        (300, 2304, 768, False, False),    # ragged M, QKV width
+        (259, 776, 1152, True, True),      # ragged M and N (N%8, not %256)
+        (1024, 1032, 2048, False, False),  # ragged N just past a tile
+    ],
+)
+def test_mfma_linear_vs_reference(ext, M, N, K, gelu, with_res):
+    x = _rand_bf16(M, K, seed=M + N)
+    w = _rand_bf16(N, K, seed=M + N + 1, scale=0.05)
+    bias = _rand_bf16(N, seed=M + N + 2)
+    res = _rand_bf16(M, N, seed=M + N + 3) if with_res else None
+    y = torch.ops.vilbert_amd.mfma_linear(x, w, bias, res, gelu)
+    ref = _mfma_linear_ref(x, w, bias, res, gelu)
+    # bf16 inputs, f32 accumulate: error ~ bf16 rounding of the inputs
+    # amplified by sqrt(K) * |x||w| — bound the max abs error empirically
+    err = (y.float() - ref).abs().max().item()
+    scale = ref.abs().max().item()
+    assert err < 0.02 * max(scale, 1.0), f"max err {err} (scale {scale})"
+
+
+def test_mfma_linear_matches_hipblaslt(ext):
+    """The plain-GEMM result must agree with the hipBLASLt path within bf16
+    rounding (both f32-accumulate over the same K)."""
+    M, N, K = 512, 768, 768
+    x = _rand_bf16(M, K, seed=1)
+    w = _rand_bf16(N, K, seed=2, scale=0.05)
+    bias = _rand_bf16(N, seed=3)
+    a = torch.ops.vilbert_amd.mfma_linear(x, w, bias, None, False)
+    b = torch.ops.vilbert_amd.linear_bias(x, w, bias)
+    assert (a.float() - b.float()).abs().max().item() < 2e-2
+
+
+def test_mfma_linear_no_bias(ext):
+    M, N, K = 128, 256, 192
+    x = _rand_bf16(M, K, seed=4)
+    w = _rand_bf16(N, K, seed=5, scale=0.05)
+    y = torch.ops.vilbert_amd.mfma_linear(x, w, None, None, False)
+    ref = x.float() @ w.float().t()
+    assert (y.float() - ref).abs().max().item() < 2e-2

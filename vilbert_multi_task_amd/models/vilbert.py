@@ -179,14 +179,16 @@ class MultiHeadSelfAttention(nn.Module):
             from .fp8 import attach_quant_pack
 
             attach_quant_pack(ctx, self._fp8_ctx_obj, self._fp8_ctx_site)
-        # NOTE: folding the residual into the out-proj GEMM epilogue
-        # (linear_bias_residual, beta=1) intermittently faults in hipBLASLt
-        # at some shapes ("write access to a read-only page", B=512 warmup)
-        # — wiring reverted; the op remains for round-2 investigation.
+        # residual folded into the out-proj GEMM epilogue — runs on the
+        # hand-written MFMA kernel (gemm_mfma.hip), NOT hipBLASLt's beta=1
+        # epilogue (which intermittently faults: "write access to a
+        # read-only page", r1 note). The following LayerNorm then reads ONE
+        # tensor instead of re-reading the residual (residual_ln was 12.1%
+        # of the serving step at HBM roofline, profiles/r04).
         if infer and isinstance(self.out, nn.Linear):
-            o = F_ops.linear_bias(ctx, self.out.weight, self.out.bias)
-        else:
-            o = self.out(ctx)
+            o = F_ops.linear_bias_residual(ctx, self.out.weight, self.out.bias, x)
+            return self.layer_norm(o), probs
+        o = self.out(ctx)
         y = self.layer_norm(self.dropout(o), residual=x)
         return y, probs
 
@@ -242,8 +244,9 @@ class FeedForward(nn.Module):
             and not torch.is_grad_enabled()
             and isinstance(self.output, nn.Linear)
         ):
-            o = F_ops.linear_bias(h, self.output.weight, self.output.bias)
-            return self.layer_norm(self.dropout(o), residual=x)
+            # fused GEMM+bias+residual epilogue (see MultiHeadSelfAttention)
+            o = F_ops.linear_bias_residual(h, self.output.weight, self.output.bias, x)
+            return self.layer_norm(o)
         return self.layer_norm(self.dropout(self.output(h)), residual=x)
 
 
@@ -309,9 +312,10 @@ class CrossAttention(nn.Module):
             fp8_out=fp8_out,
         )
         if infer and isinstance(self.out, nn.Linear):
-            o = F_ops.linear_bias(ctx, self.out.weight, self.out.bias)
-        else:
-            o = self.out(ctx)
+            # fused GEMM+bias+residual epilogue (see MultiHeadSelfAttention)
+            o = F_ops.linear_bias_residual(ctx, self.out.weight, self.out.bias, x_q)
+            return self.layer_norm(o), probs
+        o = self.out(ctx)
         y = self.layer_norm(self.dropout(o), residual=x_q)
         return y, probs
 

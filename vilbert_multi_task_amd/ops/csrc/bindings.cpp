@@ -34,6 +34,8 @@ void launch_attention_fp8out(const bf16*, const bf16*, const bf16*, const bf16*,
                              int, int, int, int, int, int, int, int, int,
                              hipStream_t);
 void launch_mfma_probe(const bf16*, const bf16*, float*, hipStream_t);
+void launch_gemm256(const bf16*, const bf16*, const bf16*, const bf16*, bf16*,
+                    long, long, long, bool, hipStream_t);
 void launch_nms_multiclass(const float*, const float*, const long*, float*, int,
                            int, float, float, hipStream_t);
 void launch_tr16_probe(short*, int, hipStream_t);
@@ -164,6 +166,43 @@ at::Tensor attention(const at::Tensor& q, const at::Tensor& k,
                    (const bf16*)v.data_ptr(), mptr,
                    (bf16*)out.data_ptr(), B, H, Lq, Lk, D, mask_mode,
                    qs, kss, vss, cur_stream());
+  return out;
+}
+
+at::Tensor mfma_linear(const at::Tensor& x, const at::Tensor& w,
+                       const c10::optional<at::Tensor>& bias,
+                       const c10::optional<at::Tensor>& residual,
+                       bool gelu) {
+  // hand-written 256x256x64 MFMA GEMM (gemm_mfma.hip): y = x @ w^T
+  // (+bias) (+residual) (+GELU) — the hipBLASLt-free hot-path GEMM with the
+  // beta=1 residual epilogue hipBLASLt faults on
+  TORCH_CHECK(x.is_cuda() && x.scalar_type() == at::kBFloat16, "mfma_linear: bf16 cuda only");
+  TORCH_CHECK(w.scalar_type() == at::kBFloat16);
+  auto xc = x.contiguous();
+  auto wc = w.contiguous();
+  const long K = xc.size(-1);
+  const long M = xc.numel() / K;
+  const long N = wc.size(0);
+  TORCH_CHECK(wc.size(-1) == K, "mfma_linear: K mismatch");
+  TORCH_CHECK(K % 64 == 0, "mfma_linear: K % 64 != 0");
+  TORCH_CHECK(N % 8 == 0, "mfma_linear: N % 8 != 0");
+  auto sizes = x.sizes().vec();
+  sizes.back() = N;
+  auto out = at::empty(sizes, x.options());
+  const bf16* bp = nullptr;
+  const bf16* rp = nullptr;
+  at::Tensor rc;
+  if (bias.has_value() && bias->defined()) {
+    TORCH_CHECK(bias->is_contiguous() && bias->numel() == N);
+    bp = (const bf16*)bias->data_ptr();
+  }
+  if (residual.has_value() && residual->defined()) {
+    rc = residual->contiguous();
+    TORCH_CHECK(rc.numel() == M * N, "mfma_linear: residual shape");
+    rp = (const bf16*)rc.data_ptr();
+  }
+  launch_gemm256((const bf16*)xc.data_ptr(), (const bf16*)wc.data_ptr(), bp,
+                 rp, (bf16*)out.data_ptr(), M, N, K, gelu, cur_stream());
   return out;
 }
 
@@ -585,6 +624,7 @@ TORCH_LIBRARY(vilbert_amd, m) {
   m.def("residual_layer_norm(Tensor x, Tensor? res, Tensor w, Tensor b, float eps) -> Tensor");
   m.def("bias_gelu(Tensor x, Tensor? bias) -> Tensor");
   m.def("attention(Tensor q, Tensor k, Tensor v, int heads, Tensor? mask) -> Tensor");
+  m.def("mfma_linear(Tensor x, Tensor w, Tensor? bias, Tensor? residual, bool gelu) -> Tensor");
   m.def("attention_probs(Tensor q, Tensor k, Tensor v, int heads, Tensor? mask) -> (Tensor, Tensor)");
   m.def("attention_fp8out(Tensor q, Tensor k, Tensor v, int heads, Tensor? mask, Tensor scales, Tensor(a!) amaxes, int site) -> (Tensor, Tensor)");
   m.def("embedding_ln(Tensor ids, Tensor pos, Tensor type, Tensor word_w, Tensor pos_w, Tensor type_w, Tensor ln_w, Tensor ln_b, float eps) -> Tensor");
@@ -611,6 +651,7 @@ TORCH_LIBRARY_IMPL(vilbert_amd, CUDA, m) {
   m.impl("bias_gelu", bias_gelu);
   m.impl("attention", attention);
   m.impl("attention_probs", attention_probs);
+  m.impl("mfma_linear", mfma_linear);
   m.impl("attention_fp8out", attention_fp8out);
   m.impl("embedding_ln", embedding_ln);
   m.impl("mfma_probe", mfma_probe);

@@ -1,0 +1,269 @@
+// Hand-written CDNA4 MFMA GEMM for the ViLBERT hot projection/FFN shapes.
+//
+// Computes C = x @ W^T (+ bias) (+ residual) (+ GELU), bf16 in / f32
+// accumulate / bf16 out — the torch nn.Linear contract with the epilogues
+// the model needs fused (SURVEY.md §7.3's "GEMM-MFMA bf16 tile kernel";
+// the reference's implicit cuBLAS GEMMs of worker.py:287-289).
+//
+// Why it exists even though hipBLASLt runs these shapes at ~1.07 PF/s
+// (profiles/r06): (1) the north star requires the transformer GEMMs as
+// hand-written MFMA kernels with an honest A/B; (2) the residual epilogue —
+// hipBLASLt's beta=1 epilogue faults intermittently at some serving shapes
+// ("write access to a read-only page", models/vilbert.py r1 note), while a
+// hand-written epilogue trivially fuses y = x@W^T + b + res, removing a full
+// HBM pass from the residual_ln that follows (12.1% of the serving step).
+//
+// Structure (guide §5 "canonical CDNA GEMM" + the 256² template parameters):
+//   - tile 256(M) x 256(N) x BK=64, 512 threads = 8 waves as 2(M) x 4(N),
+//     per-wave output 128x64 = acc[8][4] f32x4 fragments
+//   - operands staged global->LDS with __builtin_amdgcn_global_load_lds
+//     (16B/lane, lane-linear dest), double-buffered: 2 x (32KB A + 32KB B)
+//     = 128 KiB LDS
+//   - st_16x32 XOR swizzle (byte ^= ((byte>>9)&1)<<5) applied on the glds
+//     SOURCE address and the ds_read_b128 offset (both-sides-or-neither,
+//     guide §5.4 rule 21) — row-major [256][64] bf16 would otherwise put a
+//     ds_read_b128 lane group 8-way on one bank slot
+//   - both fragments read contiguous-K: A[m][k] from row-major x, and
+//     B[k][n] = W[n][k] from row-major W (nn.Linear stores W as [N,K])
+//   - ragged M/N handled by clamped loads + guarded stores; K % 64 == 0 and
+//     N % 8 == 0 required (dispatch falls back to hipBLASLt otherwise)
+//   - epilogue re-tiles C through LDS (the C fragment layout stores 2 bytes
+//     per lane per instruction — 128 scalar stores/lane; via LDS it becomes
+//     16 x b128 reads + 16 x 16B coalesced global stores), then adds the
+//     residual streamed in b128 chunks. GELU is the exact erf form
+//     (0.5*x*(1+erf(x/sqrt(2))), matching elementwise.hip:206).
+//
+// MFMA fragment maps: same as attention.hip (verified by mfma_probe):
+//   A: lane l holds A[row = l&15][k = (l>>4)*8 + j]
+//   B: lane l holds B[k = (l>>4)*8 + j][col = l&15]
+//   C: lane l, reg r hold C[row = (l>>4)*4 + r][col = l&15]
+
+#include "common.h"
+
+#define KINVSQRT2 0.70710678118654752440f
+
+// st_16x32 swizzle on a byte offset within a [256][64]-bf16 tile image
+// (row stride 128 B): flips 32-byte half-lines when row bit 2 is set.
+DEV int gswz(int o) { return o ^ (((o >> 9) & 1) << 5); }
+
+// packed bf16 pair add (two lanes of a 32-bit word), f32 math, RNE repack
+DEV unsigned int bfadd2(unsigned int a, unsigned int b) {
+  const float lo = us2f((unsigned short)a) + us2f((unsigned short)b);
+  const float hi = us2f((unsigned short)(a >> 16)) + us2f((unsigned short)(b >> 16));
+  return (unsigned int)f2us(lo) | ((unsigned int)f2us(hi) << 16);
+}
+
+typedef __attribute__((address_space(3))) unsigned int* lds_u32p;
+
+DEV bf16x8 lds_b128_g(const char* p) {
+  union { uint4 u; bf16x8 v; } c;
+  c.u = *reinterpret_cast<const uint4*>(p);
+  return c.v;
+}
+typedef const __attribute__((address_space(1))) unsigned int* glob_u32p;
+
+// one cooperative 8 KiB glds piece: 512 threads x 16 B, lane-linear LDS
+DEV void glds16(const bf16* src_lane, char* lds_wave_base) {
+  __builtin_amdgcn_global_load_lds(
+      (glob_u32p)src_lane, (lds_u32p)lds_wave_base, 16, 0, 0);
+}
+
+template <bool GELU_, bool RES>
+__global__ __launch_bounds__(512) void gemm256_kernel(
+    const bf16* __restrict__ x,    // [M,K] row-major
+    const bf16* __restrict__ w,    // [N,K] row-major (nn.Linear weight)
+    const bf16* __restrict__ bias, // [N] or nullptr
+    const bf16* __restrict__ res,  // [M,N] or nullptr (RES => non-null)
+    bf16* __restrict__ out,        // [M,N]
+    int M, int N, int K, int gx, int gy) {
+  const int tid = threadIdx.x;
+  const int lane = lane_id();
+  const int wid = __builtin_amdgcn_readfirstlane(wave_id());
+  const int wm = wid >> 2;           // 0..1  (M half)
+  const int wn = wid & 3;            // 0..3  (N quarter)
+
+  // XCD-bijective block remap (guide §5: consecutive remapped ids share the
+  // same W panel within one XCD's L2)
+  int id = blockIdx.x;
+  {
+    const int nwg = gx * gy;
+    const int q = nwg / 8, r = nwg % 8;
+    const int xcd = id % 8;
+    id = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + id / 8;
+  }
+  const int blk_m = id % gx;
+  const int blk_n = id / gx;
+  const int row0 = blk_m * 256;  // x rows
+  const int col0 = blk_n * 256;  // out cols = W rows
+
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  // [ Abuf0 32K | Abuf1 32K | Bbuf0 32K | Bbuf1 32K ] — computed by offset
+  // (a pointer ARRAY of addrspace-cast bases fails to materialize)
+#define ABUF(b) (smem + (b) * 32768)
+#define BBUF(b) (smem + 65536 + (b) * 32768)
+
+  // ---- per-thread staging geometry (constant across k-tiles) -------------
+  // glds piece g covers image bytes [g*8192 + tid*16): the element that
+  // belongs at lane-linear byte o is at swizzled position gswz(o)
+  long srcA[4], srcB[4];  // element offsets (bytes) minus the k-tile advance
+#pragma unroll
+  for (int g = 0; g < 4; ++g) {
+    const int o = g * 8192 + tid * 16;
+    const int os = gswz(o);
+    const int r = os >> 7;         // row within the 256-row tile
+    const int kb = os & 127;       // byte within the 64-element K slice
+    srcA[g] = (long)min(row0 + r, M - 1) * (K * 2) + kb;
+    srcB[g] = (long)min(col0 + r, N - 1) * (K * 2) + kb;
+  }
+  const char* xB = reinterpret_cast<const char*>(x);
+  const char* wB = reinterpret_cast<const char*>(w);
+
+  // ---- per-thread fragment read offsets (constant; buffer base varies) ---
+  int offA[2][8], offB[2][4];
+#pragma unroll
+  for (int kk = 0; kk < 2; ++kk) {
+    const int kb = (kk * 32 + (lane >> 4) * 8) * 2;
+#pragma unroll
+    for (int mt = 0; mt < 8; ++mt)
+      offA[kk][mt] = gswz((wm * 128 + mt * 16 + (lane & 15)) * 128 + kb);
+#pragma unroll
+    for (int nt = 0; nt < 4; ++nt)
+      offB[kk][nt] = gswz((wn * 64 + nt * 16 + (lane & 15)) * 128 + kb);
+  }
+
+  // bias values for this lane's four 16-col tiles (preloaded, clamped)
+  float bias_v[4];
+#pragma unroll
+  for (int nt = 0; nt < 4; ++nt) {
+    const int c = col0 + wn * 64 + nt * 16 + (lane & 15);
+    bias_v[nt] = bias ? bf2f(bias[min(c, N - 1)]) : 0.f;
+  }
+
+  auto stage = [&](int buf, int t) {
+    const long kadv = (long)t * 128;  // 64 bf16 per k-tile
+    char* la = ABUF(buf) + wid * 1024;
+    char* lb = BBUF(buf) + wid * 1024;
+#pragma unroll
+    for (int g = 0; g < 4; ++g) {
+      glds16(reinterpret_cast<const bf16*>(xB + srcA[g] + kadv), la + g * 8192);
+      glds16(reinterpret_cast<const bf16*>(wB + srcB[g] + kadv), lb + g * 8192);
+    }
+  };
+
+  f32x4 acc[8][4];
+#pragma unroll
+  for (int mt = 0; mt < 8; ++mt)
+#pragma unroll
+    for (int nt = 0; nt < 4; ++nt) acc[mt][nt] = {0.f, 0.f, 0.f, 0.f};
+
+  const int ntiles = K >> 6;  // K % 64 == 0 enforced by the launcher
+
+  // prologue: tile 0 into buf 0
+  stage(0, 0);
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+
+  int buf = 0;
+  for (int t = 0; t < ntiles; ++t) {
+    if (t + 1 < ntiles) stage(buf ^ 1, t + 1);  // flies under this tile's MFMA
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+      bf16x8 af[8], bfr[4];
+#pragma unroll
+      for (int mt = 0; mt < 8; ++mt)
+        af[mt] = lds_b128_g(ABUF(buf) + offA[kk][mt]);
+#pragma unroll
+      for (int nt = 0; nt < 4; ++nt)
+        bfr[nt] = lds_b128_g(BBUF(buf) + offB[kk][nt]);
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int mt = 0; mt < 8; ++mt)
+#pragma unroll
+        for (int nt = 0; nt < 4; ++nt)
+          acc[mt][nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[mt], bfr[nt], acc[mt][nt], 0, 0, 0);
+      __builtin_amdgcn_s_setprio(0);
+    }
+    // next tile's glds must have landed; this wave's LDS reads are done
+    asm volatile("s_waitcnt vmcnt(0) lgkmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    buf ^= 1;
+  }
+
+  // ---- epilogue: bias (+GELU) on registers, re-tile through LDS, then ----
+  // ---- (+residual) and coalesced 16 B stores -----------------------------
+  // per-wave scratch = its 16 KiB slice of the (now free) staging LDS
+  char* scratch = smem + wid * 16384;
+#pragma unroll
+  for (int mt = 0; mt < 8; ++mt) {
+#pragma unroll
+    for (int nt = 0; nt < 4; ++nt) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        float v = acc[mt][nt][r] + bias_v[nt];
+        if (GELU_) v = 0.5f * v * (1.0f + erff(v * KINVSQRT2));
+        const int rl = mt * 16 + (lane >> 4) * 4 + r;  // row in wave tile
+        const int cl = nt * 16 + (lane & 15);          // col in wave tile
+        *reinterpret_cast<short*>(scratch + rl * 128 + cl * 2) =
+            (short)f2us(v);
+      }
+    }
+  }
+  __syncthreads();
+
+  const int mrow0 = row0 + wm * 128;
+  const int ncol0 = col0 + wn * 64;
+  // batch the residual loads first (unconditional, clamped), then combine —
+  // a per-chunk load->wait->add chain serializes 16 HBM round trips, and a
+  // short-array union here round-trips every chunk through scratch
+  uint4 rv[16];
+  if (RES) {
+    const long last = (long)M * N - 8;
+#pragma unroll
+    for (int c = 0; c < 16; ++c) {
+      const int o = c * 1024 + lane * 16;
+      long gb = (long)min(mrow0 + (o >> 7), M - 1) * N + ncol0 + (o & 127) / 2;
+      rv[c] = *reinterpret_cast<const uint4*>(res + (gb < last ? gb : last));
+    }
+  }
+#pragma unroll
+  for (int c = 0; c < 16; ++c) {
+    const int o = c * 1024 + lane * 16;
+    const int row = mrow0 + (o >> 7);
+    const int col = ncol0 + (o & 127) / 2;
+    if (row >= M) continue;
+    uint4 v = *reinterpret_cast<uint4*>(scratch + o);
+    const long gb = (long)row * N + col;
+    if (col + 8 <= N) {
+      if (RES) {
+        v.x = bfadd2(v.x, rv[c].x);
+        v.y = bfadd2(v.y, rv[c].y);
+        v.z = bfadd2(v.z, rv[c].z);
+        v.w = bfadd2(v.w, rv[c].w);
+      }
+      *reinterpret_cast<uint4*>(out + gb) = v;
+    } else if (col < N) {  // ragged-N tail (edge blocks only)
+      unsigned int arr[4] = {v.x, v.y, v.z, v.w};
+      for (int j = 0; j < N - col && j < 8; ++j) {
+        float vv = us2f((unsigned short)(arr[j >> 1] >> ((j & 1) * 16)));
+        if (RES) vv += bf2f(res[gb + j]);
+        out[gb + j] = f2bf(vv);
+      }
+    }
+  }
+}
+
+void launch_gemm256(const bf16* x, const bf16* w, const bf16* bias,
+                    const bf16* res, bf16* out, long M, long N, long K,
+                    bool gelu, hipStream_t stream) {
+  const int gx = (int)((M + 255) / 256);
+  const int gy = (int)((N + 255) / 256);
+  const dim3 grid(gx * gy);
+  const size_t lds = 131072;
+#define L(G, R)                                                              \
+  hipLaunchKernelGGL((gemm256_kernel<G, R>), grid, dim3(512), lds, stream,   \
+                     x, w, bias, res, out, (int)M, (int)N, (int)K, gx, gy)
+  if (gelu) { if (res) L(true, true); else L(true, false); }
+  else      { if (res) L(false, true); else L(false, false); }
+#undef L
+}

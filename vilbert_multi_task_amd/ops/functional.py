@@ -182,6 +182,8 @@ def linear_bias_gelu(x: torch.Tensor, w: torch.Tensor, bias: torch.Tensor) -> to
     if _LINEAR_GELU_OK and _want_hip(x, w, bias):
         ext = _load_extension()
         try:
+            if _mfma_gemm_mode() == "mfma" and _mfma_linear_eligible(x, w):
+                return torch.ops.vilbert_amd.mfma_linear(x, w, bias, None, True)
             return ext.linear_bias_gelu(x, w, bias)
         except RuntimeError:
             _LINEAR_GELU_OK = False  # no algo for this arch/shape: fall back
@@ -201,6 +203,8 @@ def linear_bias(x: torch.Tensor, w: torch.Tensor, bias: torch.Tensor) -> torch.T
     if _LINEAR_OK and _want_hip(x, w, bias):
         ext = _load_extension()
         try:
+            if _mfma_gemm_mode() == "mfma" and _mfma_linear_eligible(x, w):
+                return torch.ops.vilbert_amd.mfma_linear(x, w, bias, None, False)
             return ext.linear_bias(x, w, bias)
         except RuntimeError:
             _LINEAR_OK = False
@@ -210,15 +214,34 @@ def linear_bias(x: torch.Tensor, w: torch.Tensor, bias: torch.Tensor) -> torch.T
 _LINEAR_RES_OK = True
 
 
+def _mfma_gemm_mode() -> str:
+    """VILBERT_GEMM selects the plain-GEMM backend: 'hipblaslt' (default) or
+    'mfma' (the hand-written 256x256x64 kernel, gemm_mfma.hip) — A/B lever.
+    The residual-fused path always prefers the MFMA kernel (hipBLASLt's
+    beta=1 epilogue faults intermittently — VERDICT r1 item 3)."""
+    import os
+
+    return os.environ.get("VILBERT_GEMM", "hipblaslt")
+
+
+def _mfma_linear_eligible(x: torch.Tensor, w: torch.Tensor) -> bool:
+    return x.shape[-1] % 64 == 0 and w.shape[0] % 8 == 0
+
+
 def linear_bias_residual(
     x: torch.Tensor, w: torch.Tensor, bias: torch.Tensor, residual: torch.Tensor
 ) -> torch.Tensor:
     """y = x @ w.T + bias + residual — the residual add fused into the GEMM
-    epilogue (beta=1) so the following LayerNorm reads ONE tensor."""
+    epilogue so the following LayerNorm reads ONE tensor. Runs on the
+    hand-written MFMA kernel: hipBLASLt's beta=1 epilogue faults at some
+    serving shapes ("write access to a read-only page"), the in-house
+    epilogue does not."""
     global _LINEAR_RES_OK
     if _LINEAR_RES_OK and _want_hip(x, w, bias, residual):
         ext = _load_extension()
         try:
+            if _mfma_linear_eligible(x, w):
+                return torch.ops.vilbert_amd.mfma_linear(x, w, bias, residual, False)
             return ext.linear_bias_residual(x, w, bias, residual)
         except RuntimeError:
             _LINEAR_RES_OK = False
