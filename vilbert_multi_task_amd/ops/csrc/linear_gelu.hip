@@ -212,7 +212,11 @@ static int run_epilogue_gemm(int kind, const void* x, const void* w,
     if (!p.has_algo) return 1;
     HIPBLASLT_CHECK(hipblasLtMatmulDescSetAttribute(
         p.op, HIPBLASLT_MATMUL_DESC_BIAS_POINTER, &bias, sizeof(bias)));
-    const void* cmat = kind == 0 ? y : residual;
+    // beta=0 kinds still need a VALID C pointer (hipBLASLt rejects NULL):
+    // kind 2 used to pass residual=nullptr here, which made every direct
+    // ext.linear_bias call fail ("no algo") and silently fall back to
+    // torch F.linear in the wrapper — r2 .s/A-B audit finding
+    const void* cmat = (kind == 0 || kind == 2) ? y : residual;
     int capturing = 0;
     hipStreamCaptureStatus st = hipStreamCaptureStatusNone;
     if (hipStreamIsCapturing(stream, &st) == hipSuccess &&
