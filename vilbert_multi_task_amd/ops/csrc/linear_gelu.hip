@@ -116,14 +116,19 @@ Plan& get_plan(long M, long N, long K, int kind, void* workspace, size_t ws_byte
     p.has_algo = true;
     p.beta = kind == 1 ? 1.0f : 0.0f;
     p.candidates.assign(results, results + found);
-    // Full-catalogue autotune (default ON; VILBERT_GEMM_TUNE_FULL=0 keeps
-    // just the 32-candidate heuristic shortlist): sweep every library algo
-    // supported for this problem. Warmup-only cost — autotune() times each
-    // candidate pre-capture — measured +1.1% steady-state @B1024 over the
-    // heuristic shortlist.
+    // Full-catalogue autotune (default OFF; VILBERT_GEMM_TUNE_FULL=1 opts
+    // in): sweep every library algo supported for this problem. Measured
+    // +1.1% steady-state @B1024 over the 32-candidate heuristic shortlist,
+    // BUT the catalogue contains algos that pass matmulIsAlgoSupported and
+    // then fault at run time ("Memory access fault" / "write access to a
+    // read-only page") at several serving shapes — the round-1 beta=1
+    // faults and the round-2 kind-2 faults are this one class. The
+    // heuristic shortlist has never faulted; +1.1% is not worth a
+    // process-killing fault lottery. scripts/debug_hipblaslt_algos.py pins
+    // individual catalogue algos for isolation.
     static const bool tune_full = [] {
       const char* e = std::getenv("VILBERT_GEMM_TUNE_FULL");
-      return !(e && e[0] == '0');
+      return e && e[0] == '1';
     }();
     if (tune_full) {
       std::vector<hipblasLtMatmulHeuristicResult_t> all;
