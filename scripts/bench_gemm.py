@@ -47,6 +47,8 @@ def main():
         w = torch.randn(N, K, device="cuda", dtype=torch.bfloat16) * 0.05
         bias = torch.randn(N, device="cuda", dtype=torch.bfloat16)
         res = torch.randn(M, N, device="cuda", dtype=torch.bfloat16) if args.res else None
+        lnw = torch.ones(N, device="cuda", dtype=torch.bfloat16)
+        lnb = torch.zeros(N, device="cuda", dtype=torch.bfloat16)
         flops = 2.0 * M * N * K
         res_t = {"mfma": [], "blaslt": []}
         for rnd in range(args.iters):
@@ -54,17 +56,21 @@ def main():
                 torch.cuda.synchronize()
                 t0 = time.perf_counter()
                 if mode == "mfma":
-                    torch.ops.vilbert_amd.mfma_linear(x, w, bias, res, args.gelu)
+                    y = torch.ops.vilbert_amd.mfma_linear(x, w, bias, res, args.gelu)
+                    if args.res:
+                        torch.ops.vilbert_amd.residual_layer_norm(
+                            y, None, lnw, lnb, 1e-12)
                 else:
+                    # baseline = the shipped default: torch F.linear
+                    # (hipBLASLt's own BIAS epilogue faults at large M)
                     if args.gelu:
                         torch.ops.vilbert_amd.linear_bias_gelu(x, w, bias)
                     elif args.res:
-                        # hipBLASLt residual path: plain GEMM + separate add
-                        # (the beta=1 epilogue faults — r1 finding)
-                        y = torch.ops.vilbert_amd.linear_bias(x, w, bias)
-                        y += res
+                        y = torch.nn.functional.linear(x, w, bias)
+                        torch.ops.vilbert_amd.residual_layer_norm(
+                            y, res, lnw, lnb, 1e-12)
                     else:
-                        torch.ops.vilbert_amd.linear_bias(x, w, bias)
+                        torch.nn.functional.linear(x, w, bias)
                 torch.cuda.synchronize()
                 dt = (time.perf_counter() - t0) * 1e6
                 if rnd >= 5:

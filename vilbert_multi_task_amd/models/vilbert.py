@@ -186,8 +186,11 @@ class MultiHeadSelfAttention(nn.Module):
         # tensor instead of re-reading the residual (residual_ln was 12.1%
         # of the serving step at HBM roofline, profiles/r04).
         if infer and isinstance(self.out, nn.Linear):
-            o = F_ops.linear_bias_residual(ctx, self.out.weight, self.out.bias, x)
-            return self.layer_norm(o), probs
+            if F_ops.res_fusion_active(ctx, self.out.weight):
+                o = F_ops.linear_bias_residual(ctx, self.out.weight, self.out.bias, x)
+                return self.layer_norm(o), probs
+            o = F_ops.linear_bias(ctx, self.out.weight, self.out.bias)
+            return self.layer_norm(o, residual=x), probs
         o = self.out(ctx)
         y = self.layer_norm(self.dropout(o), residual=x)
         return y, probs
@@ -244,9 +247,11 @@ class FeedForward(nn.Module):
             and not torch.is_grad_enabled()
             and isinstance(self.output, nn.Linear)
         ):
-            # fused GEMM+bias+residual epilogue (see MultiHeadSelfAttention)
-            o = F_ops.linear_bias_residual(h, self.output.weight, self.output.bias, x)
-            return self.layer_norm(o)
+            if F_ops.res_fusion_active(h, self.output.weight):
+                o = F_ops.linear_bias_residual(h, self.output.weight, self.output.bias, x)
+                return self.layer_norm(o)
+            o = F_ops.linear_bias(h, self.output.weight, self.output.bias)
+            return self.layer_norm(o, residual=x)
         return self.layer_norm(self.dropout(self.output(h)), residual=x)
 
 
@@ -312,9 +317,11 @@ class CrossAttention(nn.Module):
             fp8_out=fp8_out,
         )
         if infer and isinstance(self.out, nn.Linear):
-            # fused GEMM+bias+residual epilogue (see MultiHeadSelfAttention)
-            o = F_ops.linear_bias_residual(ctx, self.out.weight, self.out.bias, x_q)
-            return self.layer_norm(o), probs
+            if F_ops.res_fusion_active(ctx, self.out.weight):
+                o = F_ops.linear_bias_residual(ctx, self.out.weight, self.out.bias, x_q)
+                return self.layer_norm(o), probs
+            o = F_ops.linear_bias(ctx, self.out.weight, self.out.bias)
+            return self.layer_norm(o, residual=x_q), probs
         o = self.out(ctx)
         y = self.layer_norm(self.dropout(o), residual=x_q)
         return y, probs

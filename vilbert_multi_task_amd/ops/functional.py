@@ -197,15 +197,24 @@ _LINEAR_OK = True
 
 
 def linear_bias(x: torch.Tensor, w: torch.Tensor, bias: torch.Tensor) -> torch.Tensor:
-    """Autotuned hipBLASLt linear (bias epilogue) for the serving path;
-    plain F.linear otherwise."""
+    """Serving linear. Backend by VILBERT_GEMM:
+      - 'torch' (default): torch F.linear (Tensile/hipBLASLt inside torch —
+        the fastest measured stable path at the serving shapes, r06 1.07 PF)
+      - 'mfma': the hand-written 256x256x64 kernel (gemm_mfma.hip) — the
+        A/B lever the north star requires
+      - 'hipblaslt': the in-house autotuned BIAS-epilogue path. ISOLATED r2:
+        hipBLASLt's BIAS epilogue (beta 0 AND 1, heuristic algos included)
+        memory-faults at large M (e.g. 38912x2304x768) — kept opt-in for
+        the fault-isolation harness only."""
     global _LINEAR_OK
     if _LINEAR_OK and _want_hip(x, w, bias):
         ext = _load_extension()
+        mode = _mfma_gemm_mode()
         try:
-            if _mfma_gemm_mode() == "mfma" and _mfma_linear_eligible(x, w):
+            if mode == "mfma" and _mfma_linear_eligible(x, w):
                 return torch.ops.vilbert_amd.mfma_linear(x, w, bias, None, False)
-            return ext.linear_bias(x, w, bias)
+            if mode == "hipblaslt":
+                return ext.linear_bias(x, w, bias)
         except RuntimeError:
             _LINEAR_OK = False
     return torch.nn.functional.linear(x, w, bias)
@@ -215,13 +224,12 @@ _LINEAR_RES_OK = True
 
 
 def _mfma_gemm_mode() -> str:
-    """VILBERT_GEMM selects the plain-GEMM backend: 'hipblaslt' (default) or
-    'mfma' (the hand-written 256x256x64 kernel, gemm_mfma.hip) — A/B lever.
-    The residual-fused path always prefers the MFMA kernel (hipBLASLt's
-    beta=1 epilogue faults intermittently — VERDICT r1 item 3)."""
+    """VILBERT_GEMM selects the plain-GEMM backend: 'torch' (default — see
+    linear_bias docstring), 'mfma' (hand-written kernel), or 'hipblaslt'
+    (in-house BIAS-epilogue path; faults at large M — isolation only)."""
     import os
 
-    return os.environ.get("VILBERT_GEMM", "hipblaslt")
+    return os.environ.get("VILBERT_GEMM", "torch")
 
 
 def _mfma_linear_eligible(x: torch.Tensor, w: torch.Tensor) -> bool:
@@ -238,14 +246,29 @@ def linear_bias_residual(
     epilogue does not."""
     global _LINEAR_RES_OK
     if _LINEAR_RES_OK and _want_hip(x, w, bias, residual):
-        ext = _load_extension()
         try:
             if _mfma_linear_eligible(x, w):
                 return torch.ops.vilbert_amd.mfma_linear(x, w, bias, residual, False)
-            return ext.linear_bias_residual(x, w, bias, residual)
         except RuntimeError:
             _LINEAR_RES_OK = False
+    # NOT ext.linear_bias_residual: hipBLASLt's beta=1 BIAS epilogue
+    # memory-faults at large M (r2 isolation; see linear_bias docstring)
     return torch.nn.functional.linear(x, w, bias) + residual
+
+
+def res_fusion_active(x: torch.Tensor, w: torch.Tensor) -> bool:
+    """Whether the model should take the fused GEMM+bias+residual epilogue
+    (the MFMA kernel) instead of linear + LN-fused-residual. A/B lever:
+    VILBERT_RES_FUSION=0 keeps the unfused structure."""
+    import os
+
+    return (
+        os.environ.get("VILBERT_RES_FUSION", "1") == "1"
+        and x.is_cuda
+        and x.dtype == torch.bfloat16
+        and _mfma_linear_eligible(x, w)
+        and extension_available()
+    )
 
 
 # --------------------------------------------------------------------------
