@@ -39,6 +39,7 @@
 //   C: lane l, reg r hold C[row = (l>>4)*4 + r][col = l&15]
 
 #include "common.h"
+#include <cstdlib>
 
 #define KINVSQRT2 0.70710678118654752440f
 
@@ -66,6 +67,29 @@ typedef const __attribute__((address_space(1))) unsigned int* glob_u32p;
 DEV void glds16(const bf16* src_lane, char* lds_wave_base) {
   __builtin_amdgcn_global_load_lds(
       (glob_u32p)src_lane, (lds_u32p)lds_wave_base, 16, 0, 0);
+}
+
+DEV unsigned lds_byte_addr(const char* p) {
+  return (unsigned)(size_t)(__attribute__((address_space(3))) const char*)p;
+}
+
+// inline-asm LDS-DMA (guide recipe): INVISIBLE to hipcc's waitcnt
+// bookkeeping -- the compiler otherwise inserts s_waitcnt vmcnt(0) before
+// every ds_read that might alias an outstanding glds, draining the 8-phase
+// pipeline each phase. Completion is counted BY HAND with the schedule's
+// vmcnt(4)/vmcnt(2) waits. M0 (LDS base) is saved/restored in-statement;
+// per-lane dest = M0 + lane*16.
+DEV void glds16_asm(const bf16* src_lane, unsigned lds_base) {
+  unsigned keep;
+  asm volatile(
+      "s_mov_b32 %0, m0\n\t"
+      "s_mov_b32 m0, %2\n\t"
+      "s_nop 0\n\t"
+      "global_load_lds_dwordx4 %1, off\n\t"
+      "s_mov_b32 m0, %0"
+      : "=&s"(keep)
+      : "v"(src_lane), "s"(lds_base)
+      : "memory");
 }
 
 template <bool GELU_, bool RES>
@@ -253,6 +277,233 @@ __global__ __launch_bounds__(512) void gemm256_kernel(
   }
 }
 
+// ---------------------------------------------------------------------------
+// 8-phase pipelined variant (guide §5 "256² 8-phase template" structure):
+// each k-tile runs as 4 phases of {ds_read new fragments | stage 2 pieces of
+// the NEXT k-tile | barrier | 16 MFMA | counted vmcnt | barrier}, so staging
+// glds stay in flight ACROSS barriers (never drained to 0 in the loop — the
+// counted wait is the whole gain per the guide: 8-phase-with-drain0 ≈
+// 1-phase). Quadrant order (0,0),(0,1),(1,0),(1,1) over (mh,nh) halves of
+// the wave's 128x64 output keeps one A-half + both B-halves live (64 frag
+// VGPRs + 128 acc).
+//
+// Staging schedule (piece = 8 KiB = one cooperative glds = 64 rows of an
+// operand image; ktile t stages ktile t+1 into buf^1):
+//   ph0: B pieces 0,1   ph1: B pieces 2,3   ph2: A pieces 0,2   ph3: A 1,3
+// Need times at ktile t+1: ph0 reads A piece 2*wm + B piece wn (covered by
+// t.ph2/ph0-1); ph2 reads A pieces 1,3 (staged t.ph3).
+// Waits (own-wave counters; the following barrier certifies block-wide):
+//   end ph1: vmcnt(4)  -> A pieces 1,3 of THIS ktile landed (read at ph2)
+//   end ph3: vmcnt(2)  -> B0-3 + A0,2 of the NEXT ktile landed (read ph0)
+// ---------------------------------------------------------------------------
+template <bool GELU_, bool RES>
+__global__ __launch_bounds__(512) void gemm256p_kernel(
+    const bf16* __restrict__ x, const bf16* __restrict__ w,
+    const bf16* __restrict__ bias, const bf16* __restrict__ res,
+    bf16* __restrict__ out, int M, int N, int K, int gx, int gy) {
+  const int tid = threadIdx.x;
+  const int lane = lane_id();
+  const int wid = __builtin_amdgcn_readfirstlane(wave_id());
+  const int wm = wid >> 2;
+  const int wn = wid & 3;
+
+  int id = blockIdx.x;
+  {
+    const int nwg = gx * gy;
+    const int q = nwg / 8, r = nwg % 8;
+    const int xcd = id % 8;
+    id = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + id / 8;
+  }
+  const int blk_m = id % gx;
+  const int blk_n = id / gx;
+  const int row0 = blk_m * 256;
+  const int col0 = blk_n * 256;
+
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+#define ABUF(b) (smem + (b) * 32768)
+#define BBUF(b) (smem + 65536 + (b) * 32768)
+
+  long srcA[4], srcB[4];
+#pragma unroll
+  for (int g = 0; g < 4; ++g) {
+    const int o = g * 8192 + tid * 16;
+    const int os = gswz(o);
+    const int r = os >> 7;
+    const int kb = os & 127;
+    srcA[g] = (long)min(row0 + r, M - 1) * (K * 2) + kb;
+    srcB[g] = (long)min(col0 + r, N - 1) * (K * 2) + kb;
+  }
+  const char* xB = reinterpret_cast<const char*>(x);
+  const char* wB = reinterpret_cast<const char*>(w);
+
+  int offA[2][8], offB[2][4];
+#pragma unroll
+  for (int kk = 0; kk < 2; ++kk) {
+    const int kb = (kk * 32 + (lane >> 4) * 8) * 2;
+#pragma unroll
+    for (int mt = 0; mt < 8; ++mt)
+      offA[kk][mt] = gswz((wm * 128 + mt * 16 + (lane & 15)) * 128 + kb);
+#pragma unroll
+    for (int nt = 0; nt < 4; ++nt)
+      offB[kk][nt] = gswz((wn * 64 + nt * 16 + (lane & 15)) * 128 + kb);
+  }
+
+  float bias_v[4];
+#pragma unroll
+  for (int nt = 0; nt < 4; ++nt) {
+    const int c = col0 + wn * 64 + nt * 16 + (lane & 15);
+    bias_v[nt] = bias ? bf2f(bias[min(c, N - 1)]) : 0.f;
+  }
+
+  auto stageA = [&](int buf, int t, int p) {
+    glds16_asm(reinterpret_cast<const bf16*>(xB + srcA[p] + (long)t * 128),
+               lds_byte_addr(ABUF(buf) + p * 8192) + wid * 1024);
+  };
+  auto stageB = [&](int buf, int t, int p) {
+    glds16_asm(reinterpret_cast<const bf16*>(wB + srcB[p] + (long)t * 128),
+               lds_byte_addr(BBUF(buf) + p * 8192) + wid * 1024);
+  };
+
+  f32x4 acc[8][4];
+#pragma unroll
+  for (int mt = 0; mt < 8; ++mt)
+#pragma unroll
+    for (int nt = 0; nt < 4; ++nt) acc[mt][nt] = {0.f, 0.f, 0.f, 0.f};
+
+  const int ntiles = K >> 6;
+
+  // prologue: whole ktile 0 into buf 0
+#pragma unroll
+  for (int p = 0; p < 4; ++p) { stageA(0, 0, p); stageB(0, 0, p); }
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+
+  bf16x8 a_[8], b0_[4], b1_[4];
+  auto rdA = [&](int buf, int mh) {
+#pragma unroll
+    for (int mtl = 0; mtl < 4; ++mtl)
+#pragma unroll
+      for (int kk = 0; kk < 2; ++kk)
+        a_[mtl * 2 + kk] = lds_b128_g(ABUF(buf) + offA[kk][mh * 4 + mtl]);
+  };
+  auto rdB = [&](int buf, int nh, bf16x8 (&dst)[4]) {
+#pragma unroll
+    for (int ntl = 0; ntl < 2; ++ntl)
+#pragma unroll
+      for (int kk = 0; kk < 2; ++kk)
+        dst[ntl * 2 + kk] = lds_b128_g(BBUF(buf) + offB[kk][nh * 2 + ntl]);
+  };
+  auto mm = [&](int mh, int nh, bf16x8 (&bf_)[4]) {
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int mtl = 0; mtl < 4; ++mtl)
+#pragma unroll
+      for (int ntl = 0; ntl < 2; ++ntl)
+#pragma unroll
+        for (int kk = 0; kk < 2; ++kk)
+          acc[mh * 4 + mtl][nh * 2 + ntl] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a_[mtl * 2 + kk], bf_[ntl * 2 + kk], acc[mh * 4 + mtl][nh * 2 + ntl], 0, 0, 0);
+    __builtin_amdgcn_s_setprio(0);
+  };
+#define PH_BAR() __builtin_amdgcn_s_barrier()
+#define LGKM0() asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory")
+
+  int buf = 0;
+  for (int t = 0; t < ntiles; ++t) {
+    const bool st = t + 1 < ntiles;
+    const int nb = buf ^ 1;
+    // ph0: (0,0)
+    rdA(buf, 0); rdB(buf, 0, b0_);
+    if (st) { stageB(nb, t + 1, 0); stageB(nb, t + 1, 1); }
+    LGKM0(); PH_BAR();
+    mm(0, 0, b0_);
+    PH_BAR();
+    // ph1: (0,1)
+    rdB(buf, 1, b1_);
+    if (st) { stageB(nb, t + 1, 2); stageB(nb, t + 1, 3); }
+    LGKM0(); PH_BAR();
+    mm(0, 1, b1_);
+    asm volatile("s_waitcnt vmcnt(4)" ::: "memory");  // own A1,A3 landed
+    PH_BAR();
+    // ph2: (1,0)
+    rdA(buf, 1);
+    if (st) { stageA(nb, t + 1, 0); stageA(nb, t + 1, 2); }
+    LGKM0(); PH_BAR();
+    mm(1, 0, b0_);
+    PH_BAR();
+    // ph3: (1,1)
+    if (st) { stageA(nb, t + 1, 1); stageA(nb, t + 1, 3); }
+    LGKM0(); PH_BAR();
+    mm(1, 1, b1_);
+    asm volatile("s_waitcnt vmcnt(2)" ::: "memory");  // next ktile's B0-3,A0,A2
+    PH_BAR();
+    buf = nb;
+  }
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");  // drain tail stages
+#undef PH_BAR
+#undef LGKM0
+
+  // ---- epilogue: identical to gemm256_kernel ------------------------------
+  char* scratch = smem + wid * 16384;
+  __builtin_amdgcn_s_barrier();
+#pragma unroll
+  for (int mt = 0; mt < 8; ++mt) {
+#pragma unroll
+    for (int nt = 0; nt < 4; ++nt) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        float v = acc[mt][nt][r] + bias_v[nt];
+        if (GELU_) v = 0.5f * v * (1.0f + erff(v * KINVSQRT2));
+        const int rl = mt * 16 + (lane >> 4) * 4 + r;
+        const int cl = nt * 16 + (lane & 15);
+        *reinterpret_cast<short*>(scratch + rl * 128 + cl * 2) =
+            (short)f2us(v);
+      }
+    }
+  }
+  __syncthreads();
+
+  const int mrow0 = row0 + wm * 128;
+  const int ncol0 = col0 + wn * 64;
+  uint4 rv[16];
+  if (RES) {
+    const long last = (long)M * N - 8;
+#pragma unroll
+    for (int c = 0; c < 16; ++c) {
+      const int o = c * 1024 + lane * 16;
+      long gb = (long)min(mrow0 + (o >> 7), M - 1) * N + ncol0 + (o & 127) / 2;
+      rv[c] = *reinterpret_cast<const uint4*>(res + (gb < last ? gb : last));
+    }
+  }
+#pragma unroll
+  for (int c = 0; c < 16; ++c) {
+    const int o = c * 1024 + lane * 16;
+    const int row = mrow0 + (o >> 7);
+    const int col = ncol0 + (o & 127) / 2;
+    if (row >= M) continue;
+    uint4 v = *reinterpret_cast<uint4*>(scratch + o);
+    const long gb = (long)row * N + col;
+    if (col + 8 <= N) {
+      if (RES) {
+        v.x = bfadd2(v.x, rv[c].x);
+        v.y = bfadd2(v.y, rv[c].y);
+        v.z = bfadd2(v.z, rv[c].z);
+        v.w = bfadd2(v.w, rv[c].w);
+      }
+      *reinterpret_cast<uint4*>(out + gb) = v;
+    } else if (col < N) {
+      unsigned int arr[4] = {v.x, v.y, v.z, v.w};
+      for (int j = 0; j < N - col && j < 8; ++j) {
+        float vv = us2f((unsigned short)(arr[j >> 1] >> ((j & 1) * 16)));
+        if (RES) vv += bf2f(res[gb + j]);
+        out[gb + j] = f2bf(vv);
+      }
+    }
+  }
+#undef ABUF
+#undef BBUF
+}
+
 void launch_gemm256(const bf16* x, const bf16* w, const bf16* bias,
                     const bf16* res, bf16* out, long M, long N, long K,
                     bool gelu, hipStream_t stream) {
@@ -260,10 +511,21 @@ void launch_gemm256(const bf16* x, const bf16* w, const bf16* bias,
   const int gy = (int)((N + 255) / 256);
   const dim3 grid(gx * gy);
   const size_t lds = 131072;
-#define L(G, R)                                                              \
-  hipLaunchKernelGGL((gemm256_kernel<G, R>), grid, dim3(512), lds, stream,   \
+  // 8-phase pipelined schedule is the default; VILBERT_GEMM_PIPE=0 falls
+  // back to the simple 2-phase loop (A/B + bisection lever)
+  static const int pipe = [] {
+    const char* e = getenv("VILBERT_GEMM_PIPE");
+    return e ? atoi(e) : 1;
+  }();
+#define L(KRN, G, R)                                                         \
+  hipLaunchKernelGGL((KRN<G, R>), grid, dim3(512), lds, stream,              \
                      x, w, bias, res, out, (int)M, (int)N, (int)K, gx, gy)
-  if (gelu) { if (res) L(true, true); else L(true, false); }
-  else      { if (res) L(false, true); else L(false, false); }
+  if (pipe) {
+    if (gelu) { if (res) L(gemm256p_kernel, true, true); else L(gemm256p_kernel, true, false); }
+    else      { if (res) L(gemm256p_kernel, false, true); else L(gemm256p_kernel, false, false); }
+  } else {
+    if (gelu) { if (res) L(gemm256_kernel, true, true); else L(gemm256_kernel, true, false); }
+    else      { if (res) L(gemm256_kernel, false, true); else L(gemm256_kernel, false, false); }
+  }
 #undef L
 }
