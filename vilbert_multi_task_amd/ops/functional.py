@@ -258,12 +258,14 @@ def linear_bias_residual(
 
 def res_fusion_active(x: torch.Tensor, w: torch.Tensor) -> bool:
     """Whether the model should take the fused GEMM+bias+residual epilogue
-    (the MFMA kernel) instead of linear + LN-fused-residual. A/B lever:
-    VILBERT_RES_FUSION=0 keeps the unfused structure."""
+    (the MFMA kernel) instead of linear + LN-fused-residual. Measured
+    @B1024: fused 16.9k q/s vs unfused 18.8k (torch's GEMM is enough
+    faster that the saved LN re-read doesn't pay) — default OFF;
+    VILBERT_RES_FUSION=1 is the A/B arm."""
     import os
 
     return (
-        os.environ.get("VILBERT_RES_FUSION", "1") == "1"
+        os.environ.get("VILBERT_RES_FUSION", "0") == "1"
         and x.is_cuda
         and x.dtype == torch.bfloat16
         and _mfma_linear_eligible(x, w)
