@@ -510,3 +510,123 @@ def test_mfma_linear_no_bias(ext):
     y = torch.ops.vilbert_amd.mfma_linear(x, w, None, None, False)
     ref = x.float() @ w.float().t()
     assert (y.float() - ref).abs().max().item() < 2e-2
+
+
+# ---------------------------------------------------------------------------
+# attention training: fused fwd (in-kernel dropout) + hand-written backward
+# ---------------------------------------------------------------------------
+
+@pytest.mark.parametrize(
+    "B,H,Lq,Lk,D",
+    [
+        (2, 12, 38, 38, 64),     # text self
+        (2, 8, 101, 101, 128),   # vision self
+        (2, 8, 38, 101, 128),    # co-attn t->v (rectangular)
+        (2, 8, 101, 38, 128),    # co-attn v->t
+    ],
+)
+def test_attention_train_bwd_matches_fp32_autograd(ext, B, H, Lq, Lk, D):
+    from vilbert_multi_task_amd.ops.functional import _AttentionTrainFn
+
+    torch.manual_seed(B * 10 + Lq)
+    qf = torch.randn(B, Lq, H * D) * 0.5
+    kf = torch.randn(B, Lk, H * D) * 0.5
+    vf = torch.randn(B, Lk, H * D) * 0.5
+    keep = torch.ones(B, Lk)
+    keep[:, Lk - 4:] = 0
+    mask = ((1 - keep) * -1e9).view(B, 1, 1, Lk)
+    go = torch.randn(B, Lq, H * D) * 0.1
+
+    # fp32 autograd reference
+    q32 = qf.clone().requires_grad_()
+    k32 = kf.clone().requires_grad_()
+    v32 = vf.clone().requires_grad_()
+    qh = q32.view(B, Lq, H, D).transpose(1, 2)
+    kh = k32.view(B, Lk, H, D).transpose(1, 2)
+    vh = v32.view(B, Lk, H, D).transpose(1, 2)
+    s = qh @ kh.transpose(-1, -2) / math.sqrt(D) + mask
+    p = torch.softmax(s, dim=-1)
+    ref = (p @ vh).transpose(1, 2).reshape(B, Lq, H * D)
+    ref.backward(go)
+
+    # HIP path
+    q16 = qf.to(torch.bfloat16).cuda().requires_grad_()
+    k16 = kf.to(torch.bfloat16).cuda().requires_grad_()
+    v16 = vf.to(torch.bfloat16).cuda().requires_grad_()
+    m16 = mask.to(torch.bfloat16).cuda().view(B, 1, 1, Lk)
+    out = _AttentionTrainFn.apply(q16, k16, v16, H, m16.contiguous(), 0.0)
+    out.backward(go.to(torch.bfloat16).cuda())
+
+    assert (out.float().cpu() - ref.detach()).abs().max().item() < 4e-2
+    for hip, refg in ((q16.grad, q32.grad), (k16.grad, k32.grad), (v16.grad, v32.grad)):
+        err = (hip.float().cpu() - refg).abs().max().item()
+        scale = refg.abs().max().item()
+        assert err < 0.04 * max(scale, 1.0), f"grad err {err} (scale {scale})"
+
+
+def test_attention_train_dropout_consistency(ext):
+    """With an EXPLICIT keep-scale mask, fwd ctx and all three grads must
+    match the fp32 autograd of the same dropped-probs expression."""
+    B, H, Lq, Lk, D = 2, 8, 38, 101, 128
+    torch.manual_seed(77)
+    qf = torch.randn(B, Lq, H * D) * 0.5
+    kf = torch.randn(B, Lk, H * D) * 0.5
+    vf = torch.randn(B, Lk, H * D) * 0.5
+    dm_f = ((torch.rand(B, H, Lq, Lk) >= 0.1).float() / 0.9)
+    dm = dm_f.to(torch.bfloat16)
+    go = torch.randn(B, Lq, H * D) * 0.1
+
+    q32 = qf.clone().requires_grad_()
+    k32 = kf.clone().requires_grad_()
+    v32 = vf.clone().requires_grad_()
+    qh = q32.view(B, Lq, H, D).transpose(1, 2)
+    kh = k32.view(B, Lk, H, D).transpose(1, 2)
+    vh = v32.view(B, Lk, H, D).transpose(1, 2)
+    p = torch.softmax(qh @ kh.transpose(-1, -2) / math.sqrt(D), dim=-1)
+    ref = ((dm.float() * p) @ vh).transpose(1, 2).reshape(B, Lq, H * D)
+    ref.backward(go)
+
+    q16 = qf.to(torch.bfloat16).cuda()
+    k16 = kf.to(torch.bfloat16).cuda()
+    v16 = vf.to(torch.bfloat16).cuda()
+    out, probs = torch.ops.vilbert_amd.attention_train_fwd(
+        q16, k16, v16, H, None, dm.cuda().contiguous()
+    )
+    assert (out.float().cpu() - ref.detach()).abs().max().item() < 4e-2
+    assert (probs.float().cpu() - p.detach()).abs().max().item() < 2e-2
+    dq, dk, dv = torch.ops.vilbert_amd.attention_bwd(
+        q16, k16, v16, probs, dm.cuda().contiguous(),
+        go.to(torch.bfloat16).cuda(), H
+    )
+    for hip, refg in ((dq, q32.grad), (dk, k32.grad), (dv, v32.grad)):
+        err = (hip.float().cpu() - refg).abs().max().item()
+        assert err < 0.04 * max(refg.abs().max().item(), 1.0), err
+
+
+# ---------------------------------------------------------------------------
+# fused AdamW kernel vs the (CPU-tested) fallback math
+# ---------------------------------------------------------------------------
+
+def test_adamw_kernel_matches_reference(ext):
+    n = 12345
+    torch.manual_seed(5)
+    base = torch.randn(n)
+    p = base.to(torch.bfloat16).cuda()
+    master = p.float()
+    m = torch.zeros(n, device="cuda")
+    v = torch.zeros(n, device="cuda")
+    # fp32 reference of the same recurrence
+    rm, rv2, rp = torch.zeros(n), torch.zeros(n), master.cpu().clone()
+    lr, b1, b2, eps, wd = 1e-3, 0.9, 0.999, 1e-8, 0.01
+    for t in range(1, 6):
+        torch.manual_seed(50 + t)
+        g = (torch.randn(n) * 1e-2)
+        torch.ops.vilbert_amd.adamw_step(
+            p, g.to(torch.bfloat16).cuda(), master, m, v, lr, b1, b2, eps, wd, t
+        )
+        gf = g.to(torch.bfloat16).float()  # kernel sees the bf16 grad
+        rm = b1 * rm + (1 - b1) * gf
+        rv2 = b2 * rv2 + (1 - b2) * gf * gf
+        rp -= lr * ((rm / (1 - b1 ** t)) / ((rv2 / (1 - b2 ** t)).sqrt() + eps) + wd * rp)
+    assert (master.cpu() - rp).abs().max().item() < 1e-5
+    assert (p.float().cpu() - rp.to(torch.bfloat16).float()).abs().max().item() == 0.0

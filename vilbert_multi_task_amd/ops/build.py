@@ -22,7 +22,7 @@ CSRC = os.path.join(HERE, "csrc")
 OUT_DIR = os.path.join(HERE, "_C")
 OUT_SO = os.path.join(OUT_DIR, "vilbert_hip.so")
 
-SOURCES = ["elementwise.hip", "attention.hip", "gemm_mfma.hip", "nms.hip", "roialign.hip", "linear_gelu.hip", "fp8_quant.hip",
+SOURCES = ["elementwise.hip", "attention.hip", "gemm_mfma.hip", "optim.hip", "attn_bwd.hip", "nms.hip", "roialign.hip", "linear_gelu.hip", "fp8_quant.hip",
     "train_bwd.hip", "bindings.cpp"]
 ARCH = os.environ.get("PYTORCH_ROCM_ARCH", "gfx950")
 

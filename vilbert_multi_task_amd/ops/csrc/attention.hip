@@ -75,7 +75,8 @@ __global__ __launch_bounds__(THREADS) void attn_kernel(
     unsigned char* __restrict__ out8 = nullptr,
     const float* __restrict__ fp8_scales = nullptr,
     float* __restrict__ fp8_amaxes = nullptr, int fp8_site = 0,
-    bf16* __restrict__ probs_out = nullptr) {
+    bf16* __restrict__ probs_out = nullptr,
+    const bf16* __restrict__ dropm = nullptr) {
   const float fp8_inv = FP8OUT ? 1.0f / fp8_scales[fp8_site] : 0.f;
   float fp8_amax = 0.f;
   // qs/ks/vs: row strides (elems) of q/k/v — [B,L,H*D] views into a fused
@@ -266,17 +267,31 @@ __global__ __launch_bounds__(THREADS) void attn_kernel(
     }
 
     // ---- P -> LDS (bf16, swizzled row-major; A-frag readable) -------------
+    // training (dropm != nullptr): PV consumes the DROPPED P~ = dm .* P
+    // while the probs export below stays pre-dropout (what the backward
+    // kernels need). dm values preload batched at clamped addresses.
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
       const int prow = (lane >> 4) * 4 + r;
       char* prow_base = P_lds + prow * (LK_PAD * 2);
       const int psw = SWZ(prow);
+      float dmv[NTMAX];
+      if (dropm != nullptr) {
+        const int rowc = min(qrow0 + (lane >> 4) * 4 + r, Lq - 1);
+        const long db = ((long)bh * Lq + rowc) * Lk;
+#pragma unroll
+        for (int nt = 0; nt < NTMAX; ++nt)
+          dmv[nt] = bf2f(dropm[db + min(nt * 16 + col0, Lk - 1)]);
+      } else {
+#pragma unroll
+        for (int nt = 0; nt < NTMAX; ++nt) dmv[nt] = 1.f;
+      }
 #pragma unroll
       for (int nt = 0; nt < NTMAX; ++nt) {
         if (nt >= NT) break;
         const int col = nt * 16 + col0;
         *reinterpret_cast<short*>(prow_base + ((col * 2) ^ psw)) =
-            (short)f2us(acc_s[nt][r]);
+            (short)f2us(acc_s[nt][r] * dmv[nt]);
       }
     }
     // per-wave P buffer: same wave writes then reads (compiler inserts the
@@ -912,7 +927,8 @@ void launch_attention_impl(const bf16* q, const bf16* k, const bf16* v,
                            int vs, unsigned char* out8,
                            const float* fp8_scales, float* fp8_amaxes,
                            int fp8_site, hipStream_t stream,
-                           bf16* probs_out = nullptr) {
+                           bf16* probs_out = nullptr,
+                           const bf16* dropm = nullptr) {
   const float scale = 1.0f / sqrtf((float)D);
   const int LK_PAD = (Lk + 31) & ~31;
   static const int kglobal_env = [] {
@@ -943,7 +959,7 @@ void launch_attention_impl(const bf16* q, const bf16* k, const bf16* v,
   const int BH = B * H;
   const bool bhloop_ok = mask_mode <= 1 && !kglobal_env && nsplit == 1 &&
                          !pipe_env && nwaves == 4 && Lq <= 128 &&
-                         out8 == nullptr;
+                         out8 == nullptr && dropm == nullptr;
   // measured (r2 A/B @B1024): v3 first cut 0.42-0.70x of v2 — default OFF
   // until the prefetch actually pays; VILBERT_ATTN_BHLOOP=1 forces it for
   // iteration. (v2's 2-6 co-resident WGs/CU already overlap staging with
@@ -1005,7 +1021,7 @@ void launch_attention_impl(const bf16* q, const bf16* k, const bf16* v,
       hipLaunchKernelGGL((attn_kernel<DD, KG, NTM, false>), grid, dim3(256),  \
                          lds, stream, q, k, v, mask, out, B, H, Lq, Lk,       \
                          mask_mode, scale, qs, ks, vs, nsplit, nullptr,       \
-                         nullptr, nullptr, 0, probs_out);                     \
+                         nullptr, nullptr, 0, probs_out, dropm);              \
   } while (0)
   const bool small = LK_PAD <= 64;  // NTMAX=4 halves the accumulator VGPRs
   if (D == 64) {
@@ -1032,6 +1048,15 @@ void launch_attention_probs(const bf16* q, const bf16* k, const bf16* v,
                             int qs, int ks, int vs, hipStream_t stream) {
   launch_attention_impl(q, k, v, mask, out, B, H, Lq, Lk, D, mask_mode, qs, ks,
                         vs, nullptr, nullptr, nullptr, 0, stream, probs);
+}
+
+void launch_attention_train_fwd(const bf16* q, const bf16* k, const bf16* v,
+                                const bf16* mask, const bf16* dropm,
+                                bf16* out, bf16* probs, int B, int H, int Lq,
+                                int Lk, int D, int mask_mode, int qs, int ks,
+                                int vs, hipStream_t stream) {
+  launch_attention_impl(q, k, v, mask, out, B, H, Lq, Lk, D, mask_mode, qs, ks,
+                        vs, nullptr, nullptr, nullptr, 0, stream, probs, dropm);
 }
 
 void launch_attention_fp8out(const bf16* q, const bf16* k, const bf16* v,

@@ -29,6 +29,13 @@ void launch_attention(const bf16*, const bf16*, const bf16*, const bf16*, bf16*,
 void launch_attention_probs(const bf16*, const bf16*, const bf16*, const bf16*,
                             bf16*, bf16*, int, int, int, int, int, int, int,
                             int, int, hipStream_t);
+void launch_attention_train_fwd(const bf16*, const bf16*, const bf16*,
+                                const bf16*, const bf16*, bf16*, bf16*, int,
+                                int, int, int, int, int, int, int, int,
+                                hipStream_t);
+void launch_attention_bwd(const bf16*, const bf16*, const bf16*, const bf16*,
+                          const bf16*, const bf16*, bf16*, bf16*, bf16*,
+                          bf16*, int, int, int, int, int, hipStream_t);
 void launch_attention_fp8out(const bf16*, const bf16*, const bf16*, const bf16*,
                              bf16*, unsigned char*, const float*, float*, int,
                              int, int, int, int, int, int, int, int, int,
@@ -36,6 +43,8 @@ void launch_attention_fp8out(const bf16*, const bf16*, const bf16*, const bf16*,
 void launch_mfma_probe(const bf16*, const bf16*, float*, hipStream_t);
 void launch_gemm256(const bf16*, const bf16*, const bf16*, const bf16*, bf16*,
                     long, long, long, bool, hipStream_t);
+void launch_adamw(bf16*, const void*, bool, float*, float*, float*, long,
+                  float, float, float, float, float, long, hipStream_t);
 void launch_nms_multiclass(const float*, const float*, const long*, float*, int,
                            int, float, float, hipStream_t);
 void launch_tr16_probe(short*, int, hipStream_t);
@@ -206,6 +215,25 @@ at::Tensor mfma_linear(const at::Tensor& x, const at::Tensor& w,
   return out;
 }
 
+void adamw_step(at::Tensor param, at::Tensor grad, at::Tensor master,
+                at::Tensor m, at::Tensor v, double lr, double b1, double b2,
+                double eps, double wd, int64_t step) {
+  TORCH_CHECK(param.is_cuda() && param.scalar_type() == at::kBFloat16);
+  TORCH_CHECK(master.scalar_type() == at::kFloat &&
+              m.scalar_type() == at::kFloat && v.scalar_type() == at::kFloat);
+  const long n = param.numel();
+  TORCH_CHECK(grad.numel() == n && master.numel() == n && m.numel() == n &&
+              v.numel() == n);
+  TORCH_CHECK(param.is_contiguous() && grad.is_contiguous() &&
+              master.is_contiguous() && m.is_contiguous() && v.is_contiguous());
+  const bool gb = grad.scalar_type() == at::kBFloat16;
+  TORCH_CHECK(gb || grad.scalar_type() == at::kFloat);
+  launch_adamw((bf16*)param.data_ptr(), grad.data_ptr(), gb,
+               master.data_ptr<float>(), m.data_ptr<float>(),
+               v.data_ptr<float>(), n, (float)lr, (float)b1, (float)b2,
+               (float)eps, (float)wd, step, cur_stream());
+}
+
 std::tuple<at::Tensor, at::Tensor> attention_probs(
     const at::Tensor& q, const at::Tensor& k, const at::Tensor& v,
     int64_t heads, const c10::optional<at::Tensor>& mask) {
@@ -239,6 +267,78 @@ std::tuple<at::Tensor, at::Tensor> attention_probs(
       (bf16*)probs.data_ptr(), B, H, Lq, Lk, D, mask_mode, qs, kss, vss,
       cur_stream());
   return {out, probs};
+}
+
+std::tuple<at::Tensor, at::Tensor> attention_train_fwd(
+    const at::Tensor& q, const at::Tensor& k, const at::Tensor& v,
+    int64_t heads, const c10::optional<at::Tensor>& mask,
+    const c10::optional<at::Tensor>& dropm) {
+  // training forward: context uses the DROPPED probs, the returned probs
+  // tensor is pre-dropout (what attention_bwd consumes)
+  TORCH_CHECK(q.is_cuda() && q.scalar_type() == at::kBFloat16);
+  const int B = (int)q.size(0), Lq = (int)q.size(1), HD = (int)q.size(2);
+  const int Lk = (int)k.size(1);
+  const int H = (int)heads;
+  const int D = HD / H;
+  TORCH_CHECK(HD % H == 0 && (D == 64 || D == 128));
+  TORCH_CHECK(Lq <= 128 && Lk <= 128);
+  const int qs = row_stride_of(q, "q"), kss = row_stride_of(k, "k"),
+            vss = row_stride_of(v, "v");
+  int mask_mode = 0;
+  const bf16* mptr = nullptr;
+  if (mask.has_value() && mask->defined()) {
+    TORCH_CHECK(mask->scalar_type() == at::kBFloat16 && mask->is_contiguous());
+    const long mn = mask->numel();
+    if (mn == (long)B * Lk) mask_mode = 1;
+    else if (mn == (long)B * Lq * Lk) mask_mode = 2;
+    else TORCH_CHECK(false, "attention_train_fwd: bad mask shape");
+    mptr = (const bf16*)mask->data_ptr();
+  }
+  const bf16* dmp = nullptr;
+  if (dropm.has_value() && dropm->defined()) {
+    TORCH_CHECK(dropm->is_contiguous() &&
+                dropm->numel() == (long)B * H * Lq * Lk);
+    dmp = (const bf16*)dropm->data_ptr();
+  }
+  auto out = at::empty({q.size(0), q.size(1), q.size(2)}, q.options());
+  auto probs = at::empty({(long)B, (long)H, (long)Lq, (long)Lk}, q.options());
+  launch_attention_train_fwd(
+      (const bf16*)q.data_ptr(), (const bf16*)k.data_ptr(),
+      (const bf16*)v.data_ptr(), mptr, dmp, (bf16*)out.data_ptr(),
+      (bf16*)probs.data_ptr(), B, H, Lq, Lk, D, mask_mode, qs, kss, vss,
+      cur_stream());
+  return {out, probs};
+}
+
+std::tuple<at::Tensor, at::Tensor, at::Tensor> attention_bwd(
+    const at::Tensor& q, const at::Tensor& k, const at::Tensor& v,
+    const at::Tensor& probs, const c10::optional<at::Tensor>& dropm,
+    const at::Tensor& dout, int64_t heads) {
+  TORCH_CHECK(q.is_cuda() && q.scalar_type() == at::kBFloat16);
+  TORCH_CHECK(q.is_contiguous() && k.is_contiguous() && v.is_contiguous());
+  TORCH_CHECK(probs.is_contiguous());
+  const int B = (int)q.size(0), Lq = (int)q.size(1), HD = (int)q.size(2);
+  const int Lk = (int)k.size(1);
+  const int H = (int)heads;
+  const int D = HD / H;
+  TORCH_CHECK(HD % H == 0 && (D == 64 || D == 128));
+  auto dc = dout.contiguous();
+  const bf16* dmp = nullptr;
+  if (dropm.has_value() && dropm->defined()) {
+    TORCH_CHECK(dropm->is_contiguous());
+    dmp = (const bf16*)dropm->data_ptr();
+  }
+  auto dq = at::empty_like(q);
+  auto dk = at::empty_like(k);
+  auto dv = at::empty_like(v);
+  auto ds = at::empty({(long)B, (long)H, (long)Lq, (long)Lk}, q.options());
+  launch_attention_bwd(
+      (const bf16*)q.data_ptr(), (const bf16*)k.data_ptr(),
+      (const bf16*)v.data_ptr(), (const bf16*)probs.data_ptr(), dmp,
+      (const bf16*)dc.data_ptr(), (bf16*)ds.data_ptr(),
+      (bf16*)dq.data_ptr(), (bf16*)dk.data_ptr(), (bf16*)dv.data_ptr(), B, H,
+      Lq, Lk, D, cur_stream());
+  return {dq, dk, dv};
 }
 
 std::tuple<at::Tensor, at::Tensor> attention_fp8out(
@@ -624,7 +724,10 @@ TORCH_LIBRARY(vilbert_amd, m) {
   m.def("residual_layer_norm(Tensor x, Tensor? res, Tensor w, Tensor b, float eps) -> Tensor");
   m.def("bias_gelu(Tensor x, Tensor? bias) -> Tensor");
   m.def("attention(Tensor q, Tensor k, Tensor v, int heads, Tensor? mask) -> Tensor");
+  m.def("adamw_step(Tensor(a!) param, Tensor grad, Tensor(b!) master, Tensor(c!) m, Tensor(d!) v, float lr, float b1, float b2, float eps, float wd, int step) -> ()");
   m.def("mfma_linear(Tensor x, Tensor w, Tensor? bias, Tensor? residual, bool gelu) -> Tensor");
+  m.def("attention_train_fwd(Tensor q, Tensor k, Tensor v, int heads, Tensor? mask, Tensor? dropm) -> (Tensor, Tensor)");
+  m.def("attention_bwd(Tensor q, Tensor k, Tensor v, Tensor probs, Tensor? dropm, Tensor dout, int heads) -> (Tensor, Tensor, Tensor)");
   m.def("attention_probs(Tensor q, Tensor k, Tensor v, int heads, Tensor? mask) -> (Tensor, Tensor)");
   m.def("attention_fp8out(Tensor q, Tensor k, Tensor v, int heads, Tensor? mask, Tensor scales, Tensor(a!) amaxes, int site) -> (Tensor, Tensor)");
   m.def("embedding_ln(Tensor ids, Tensor pos, Tensor type, Tensor word_w, Tensor pos_w, Tensor type_w, Tensor ln_w, Tensor ln_b, float eps) -> Tensor");
@@ -651,7 +754,10 @@ TORCH_LIBRARY_IMPL(vilbert_amd, CUDA, m) {
   m.impl("bias_gelu", bias_gelu);
   m.impl("attention", attention);
   m.impl("attention_probs", attention_probs);
+  m.impl("attention_train_fwd", attention_train_fwd);
+  m.impl("attention_bwd", attention_bwd);
   m.impl("mfma_linear", mfma_linear);
+  m.impl("adamw_step", adamw_step);
   m.impl("attention_fp8out", attention_fp8out);
   m.impl("embedding_ln", embedding_ln);
   m.impl("mfma_probe", mfma_probe);

@@ -273,6 +273,43 @@ def res_fusion_active(x: torch.Tensor, w: torch.Tensor) -> bool:
     )
 
 
+class _AttentionTrainFn(torch.autograd.Function):
+    """Fused attention fwd (attention.hip, probs export + in-kernel dropout)
+    with the hand-written backward (attn_bwd.hip): dS from the saved
+    pre-dropout probs, then dQ = dS@K, dK = dS^T@Q, dV = P~^T@dO."""
+
+    @staticmethod
+    def forward(ctx, q, k, v, heads, mask, drop_p):
+        b, lq, hd = q.shape
+        lk = k.shape[1]
+        dm = None
+        if drop_p > 0.0:
+            keep = torch.rand(b, heads, lq, lk, device=q.device) >= drop_p
+            dm = (keep.to(torch.bfloat16) / (1.0 - drop_p)).contiguous()
+        out, probs = torch.ops.vilbert_amd.attention_train_fwd(
+            q, k, v, heads, mask, dm
+        )
+        if dm is None:
+            ctx.save_for_backward(q, k, v, probs)
+        else:
+            ctx.save_for_backward(q, k, v, probs, dm)
+        ctx.heads = heads
+        ctx.has_dm = dm is not None
+        return out
+
+    @staticmethod
+    def backward(ctx, gout):
+        if ctx.has_dm:
+            q, k, v, probs, dm = ctx.saved_tensors
+        else:
+            q, k, v, probs = ctx.saved_tensors
+            dm = None
+        dq, dk, dv = torch.ops.vilbert_amd.attention_bwd(
+            q, k, v, probs, dm, gout, ctx.heads
+        )
+        return dq, dk, dv, None, None, None
+
+
 # --------------------------------------------------------------------------
 # Multi-head scaled-dot-product attention with additive mask.
 # Flattened [B, L, H*D] layout so the GPU path needs no transpose copies.
@@ -292,6 +329,30 @@ def attention(
     """q: [B,Lq,H*D], k/v: [B,Lk,H*D]; mask_bias additive [B,1,1,Lk] or
     [B,1,Lq,Lk] (0 keep, large-negative masked). Returns
     (ctx [B,Lq,H*D], probs [B,H,Lq,Lk] or None)."""
+    # ---- training path: fused HIP forward + hand-written backward --------
+    # (attn_bwd.hip; r1 fell back to torch autograd math here). Dropout is
+    # applied IN-KERNEL via a keep-scale mask so PV consumes the dropped
+    # probs while the saved probs stay pre-dropout for the backward.
+    if (
+        torch.is_grad_enabled()
+        and q.is_cuda
+        and q.dtype == torch.bfloat16
+        and not need_probs
+        and (q.requires_grad or k.requires_grad or v.requires_grad)
+        and os.environ.get("VILBERT_AMD_EAGER_BWD") != "1"
+        and os.environ.get("VILBERT_AMD_FORCE_EAGER") != "1"
+        and q.shape[1] <= 128
+        and k.shape[1] <= 128
+        and (q.shape[2] // num_heads) in (64, 128)
+        and _load_extension() is not None
+    ):
+        mb = mask_bias.contiguous() if mask_bias is not None else None
+        out = _AttentionTrainFn.apply(
+            q.contiguous(), k.contiguous(), v.contiguous(), num_heads, mb,
+            dropout_p if training else 0.0,
+        )
+        return out, None
+
     use_hip = (
         _want_hip(q, k, v)
         and not (training and dropout_p > 0.0)

@@ -3,7 +3,8 @@
 The engine the reference links to but does not vendor (SURVEY.md §0
 obligation 2: the 12-in-1 round-robin training engine), built MI355X-first:
 one process per GPU over RCCL (`nccl` backend on ROCm), bucketed all-reduce
-overlapped with backward (parallel/ddp.py), bf16 params with fp32 AdamW
+overlapped with backward (parallel/ddp.py), bf16 params with fp32-master
+fused AdamW (parallel/optim.py)
 master state, checkpoints that round-trip to the upstream .bin layout
 (models/checkpoint.py).
 
@@ -132,7 +133,11 @@ class MultiTaskTrainer:
         self.grad_clip = grad_clip
         self.model = model.to(device)
         self.ddp = BucketedDataParallel(self.model, bucket_bytes=bucket_bytes)
-        self.opt = torch.optim.AdamW(self.model.parameters(), lr=lr, weight_decay=0.01)
+        # fused AdamW with fp32 master weights (parallel/optim.py): one HIP
+        # kernel per tensor on GPU; identical-math torch fallback elsewhere
+        from .optim import FusedAdamW
+
+        self.opt = FusedAdamW(self.model.parameters(), lr=lr, weight_decay=0.01)
         self.sampler = RoundRobinTaskSampler(rank=rank, world_size=world_size)
         self.base_lr = lr
         self.warmup_steps = warmup_steps
