@@ -106,3 +106,30 @@ def test_media_path_traversal_blocked(client):
     assert r.status_code == 404
     r = c.get("/media/%2e%2e/%2e%2e/etc/passwd")
     assert r.status_code == 404
+
+
+def test_index_renders_frontend(client):
+    """GET / serves the full demo page with a fresh socketid and the demo
+    image list substituted (views.py:39-42 template-context contract)."""
+    client, _app = client
+    r = client.get("/")
+    assert r.status_code == 200
+    body = r.text
+    assert "__SOCKET_ID__" not in body and "__DEMO_IMAGES__" not in body
+    assert 'id="selected-task"' in body       # the 8-task dropdown
+    assert 'option value="16"' in body        # GuessWhat entry
+    assert 'id="Console"' in body             # terminal panel
+    assert 'id="ReferExpressionsTaskResultImage_1"' in body
+    assert "/static/app.js" in body
+
+
+def test_static_assets_served(client):
+    client, _app = client
+    js = client.get("/static/app.js")
+    assert js.status_code == 200
+    assert "addImagesToSampleList" in js.text
+    assert "Only a maximum of 4 files are allowed!" in js.text  # upload cap
+    css = client.get("/static/style.css")
+    assert css.status_code == 200
+    r = client.get("/static/../app.py")
+    assert r.status_code == 404  # traversal guarded

@@ -118,15 +118,27 @@ def test_http_worker_ws_across_processes(tmp_path):
         time.sleep(0.05)
 
     try:
+        # full demo loop: take the socketid the SERVED PAGE injects
+        # (views.py:39-42 context contract; app.js sends it on ws open)
+        import re
+
+        page = urllib.request.urlopen(
+            f"http://127.0.0.1:{http_port}/", timeout=10
+        ).read().decode()
+        m = re.search(r'data-socketid="([0-9a-f-]{36})"', page)
+        assert m, "served page must carry a fresh socketid"
+        sock_id = m.group(1)
+        assert 'id="selected-task"' in page  # frontend really served
+
         ws = _ws_connect(http_port)
-        _ws_send(ws, "xp1")
+        _ws_send(ws, sock_id)
         time.sleep(0.3)
 
         ctx = mp.get_context("spawn")
         proc = ctx.Process(target=_worker_proc, args=(queue_path, db_path, hub_port))
         proc.start()
 
-        body = ("socket_id=xp1&task_id=1&question=what+is+this"
+        body = (f"socket_id={sock_id}&task_id=1&question=what+is+this"
                 "&image_list%5B%5D=demo/z.jpg").encode()
         urllib.request.urlopen(urllib.request.Request(
             f"http://127.0.0.1:{http_port}/", data=body,

@@ -97,29 +97,54 @@ def create_app(
         async def _stop_hub():
             await hub.stop()
 
-    static_index = os.path.join(os.path.dirname(__file__), "static", "index.html")
+    static_dir = os.path.join(os.path.dirname(__file__), "static")
+    static_index = os.path.join(static_dir, "index.html")
 
-    @app.get("/", response_class=HTMLResponse)
-    async def index() -> str:
-        if os.path.exists(static_index):
-            with open(static_index, encoding="utf-8") as f:
-                return f.read()
-        return _INDEX_HTML
-
-    @app.get("/demo_images/")
-    async def demo_images():
-        """Random demo gallery (views.py:64-81: 6 random COCO test2014 images,
-        falling back to static samples when the directory is absent)."""
+    def _demo_image_urls():
+        """views.py:64-81: 6 random COCO test2014 images, falling back to the
+        uploaded-demo directory, then to an empty gallery."""
         import random
 
         coco_dir = os.path.join(media_root, "test2014")
         if os.path.isdir(coco_dir):
             pool = [f for f in os.listdir(coco_dir) if f.lower().endswith((".jpg", ".png"))]
             picks = random.sample(pool, min(6, len(pool)))
-            return JSONResponse({"images": [f"test2014/{p}" for p in picks]})
+            return [f"/media/test2014/{p}" for p in picks]
         demo_dir = os.path.join(media_root, "demo")
         pool = sorted(os.listdir(demo_dir)) if os.path.isdir(demo_dir) else []
-        return JSONResponse({"images": [f"demo/{p}" for p in pool[:6]]})
+        return [f"/media/demo/{p}" for p in pool[:6]]
+
+    def _render_index() -> str:
+        """Server-side substitution standing in for the Django template
+        context {demo_images, socketid} (views.py:39-42)."""
+        if not os.path.exists(static_index):
+            return _INDEX_HTML
+        with open(static_index, encoding="utf-8") as f:
+            page = f.read()
+        sock = str(uuid.uuid4())
+        return page.replace("__SOCKET_ID__", sock).replace(
+            "__DEMO_IMAGES__", json.dumps(_demo_image_urls())
+        )
+
+    @app.get("/", response_class=HTMLResponse)
+    async def index() -> str:
+        return _render_index()
+
+    @app.get("/static/{name}")
+    async def static_file(name: str):
+        from fastapi.responses import FileResponse
+
+        full = os.path.normpath(os.path.join(static_dir, name))
+        if os.path.commonpath([full, static_dir]) != static_dir or not os.path.isfile(full):
+            return JSONResponse({"error": "not found"}, status_code=404)
+        media_types = {".js": "text/javascript", ".css": "text/css", ".html": "text/html"}
+        return FileResponse(full, media_type=media_types.get(os.path.splitext(name)[1]))
+
+    @app.get("/demo_images/")
+    async def demo_images():
+        """Random demo gallery as JSON (views.py:64-81 equivalent for
+        API consumers; the index page gets the list inlined)."""
+        return JSONResponse({"images": [p[len("/media/"):] for p in _demo_image_urls()]})
 
     @app.get("/admin/")
     async def admin_index():
@@ -154,11 +179,15 @@ def create_app(
         task_id = fields.get("task_id", [""])[0]
         question = fields.get("question", [""])[0].lower()  # views.py:28
         image_list: List[str] = fields.get("image_list[]") or fields.get("image_list") or []
-        # absolute-path resolution analogue of views.py:30-32
-        paths = [
-            p if os.path.isabs(p) else os.path.join(media_root, p.lstrip("/"))
-            for p in image_list
-        ]
+        # absolute-path resolution analogue of views.py:30-32: the frontend
+        # POSTs /media/... URL paths (upload response + <img> pathname) which
+        # resolve under media_root; anything else absolute passes through
+        def _resolve(p: str) -> str:
+            if p.startswith("/media/"):
+                return os.path.join(media_root, p[len("/media/"):])
+            return p if os.path.isabs(p) else os.path.join(media_root, p.lstrip("/"))
+
+        paths = [_resolve(p) for p in image_list]
         from ..utils.trace import log_json, new_trace_id
 
         trace_id = new_trace_id()
@@ -193,7 +222,10 @@ def create_app(
             with open(dst, "wb") as out:
                 out.write(payload)
             app.state.db.add_attachment(dst)  # Attachment row (models.py:45-46)
-            file_paths.append(dst)
+            # reference returns URL paths the frontend uses as <img> src and
+            # later POSTs back (views.py:106 + demo_images.html done handler);
+            # the submit handler resolves them back under media_root
+            file_paths.append(f"/media/demo/{name}")
         return JSONResponse({"file_paths": file_paths})
 
     @app.websocket("/chat/")

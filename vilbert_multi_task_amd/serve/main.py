@@ -66,7 +66,10 @@ def build_worker(args):
         device = f"cuda:{args.device}"
     if device.startswith("cuda"):
         model = model.to(device=device, dtype=torch.bfloat16)
-    runner = GraphRunner(model, device=device, use_graphs=device.startswith("cuda"), serving_fast=True, fp8=args.fp8)
+    runner = GraphRunner(
+        model, device=device, use_graphs=device.startswith("cuda"),
+        serving_fast=True, fp8=args.fp8, feat_dim=cfg.v_feature_size,
+    )
 
     provider = None
     if args.detector:
@@ -74,7 +77,10 @@ def build_worker(args):
 
         provider = DetectorFeatureProvider(device=device)
     else:
-        provider = SyntheticFeatureProvider(device=device)
+        # feature dim must match the CONFIG (the tiny smoke config is 128-d)
+        provider = SyntheticFeatureProvider(device=device, feat_dim=cfg.v_feature_size)
+
+    from ..data.tokenizer import BertWordPieceTokenizer
 
     return ServingWorker(
         runner,
@@ -82,6 +88,9 @@ def build_worker(args):
         Database(args.db),
         PushClient(port=args.hub_port),
         provider=provider,
+        # tokenizer ids must stay inside the CONFIG's vocab (a tiny config
+        # with the default 30522-sized hash fallback crashed the embedding)
+        tokenizer=BertWordPieceTokenizer(vocab_size=cfg.vocab_size),
         vqa_vocab=AnswerVocab(cfg.num_labels_vqa, args.vqa_answers),
         gqa_vocab=AnswerVocab(cfg.num_labels_gqa, args.gqa_answers),
         max_batch_rows=args.max_batch,
