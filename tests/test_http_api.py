@@ -133,3 +133,18 @@ def test_static_assets_served(client):
     assert css.status_code == 200
     r = client.get("/static/../app.py")
     assert r.status_code == 404  # traversal guarded
+
+
+def test_admin_crud(client):
+    """Admin CRUD surface (demo/admin.py equivalent): edit a Tasks row,
+    delete a QuestionAnswer row."""
+    client, app = client
+    r = client.post("/admin/tasks/1/", json={"placeholder": "edited!", "bogus": 1})
+    assert r.status_code == 200 and r.json()["placeholder"] == "edited!"
+    assert client.get("/get_task_details/1/").json()["placeholder"] == "edited!"
+    # no editable fields -> 400
+    assert client.post("/admin/tasks/1/", json={"bogus": 1}).status_code == 400
+    # question delete
+    qa = app.state.db.create_question(1, "q", "[]", "s")
+    assert client.delete(f"/admin/questions/{qa}/").json() == {"deleted": qa}
+    assert client.delete(f"/admin/questions/{qa}/").status_code == 404

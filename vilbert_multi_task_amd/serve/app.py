@@ -148,8 +148,9 @@ def create_app(
 
     @app.get("/admin/")
     async def admin_index():
-        """Read-only admin surface (the reference exposes Django admin list
-        views for Tasks and QuestionAnswer — demo/admin.py:1-34)."""
+        """Admin surface (the reference exposes Django admin CRUD for Tasks
+        and QuestionAnswer — demo/admin.py:1-34). List view here; edits via
+        POST /admin/tasks/{id}/ and DELETE /admin/questions/{id}/."""
         return JSONResponse(
             {
                 "tasks": app.state.db.list_tasks(),
@@ -160,6 +161,27 @@ def create_app(
                 },
             }
         )
+
+    @app.post("/admin/tasks/{task_id}/")
+    async def admin_update_task(task_id: int, request: Request):
+        """Edit a Tasks row (Django admin change-view equivalent): JSON body
+        with any of name/placeholder/description/num_of_images/example."""
+        try:
+            fields = await request.json()
+        except Exception:
+            return JSONResponse({"error": "json body required"}, status_code=400)
+        if not isinstance(fields, dict):
+            return JSONResponse({"error": "json object required"}, status_code=400)
+        if not app.state.db.update_task(task_id, fields):
+            return JSONResponse({"error": "no editable fields or unknown task"},
+                                status_code=400)
+        return JSONResponse(app.state.db.get_task(task_id))
+
+    @app.delete("/admin/questions/{qa_id}/")
+    async def admin_delete_question(qa_id: int):
+        if not app.state.db.delete_question(qa_id):
+            return JSONResponse({"error": "not found"}, status_code=404)
+        return JSONResponse({"deleted": qa_id})
 
     @app.get("/media/{path:path}")
     async def media(path: str):

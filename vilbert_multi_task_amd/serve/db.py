@@ -155,6 +155,25 @@ class Database:
         rows = self._conn().execute("SELECT * FROM tasks ORDER BY unique_id").fetchall()
         return [dict(r) for r in rows]
 
+    def update_task(self, unique_id: int, fields: Dict[str, Any]) -> bool:
+        """Admin CRUD (demo/admin.py equivalent): edit a Tasks row's
+        editable columns."""
+        allowed = {"name", "placeholder", "description", "num_of_images", "example"}
+        cols = {k: v for k, v in fields.items() if k in allowed}
+        if not cols:
+            return False
+        sets = ", ".join(f"{k}=?" for k in cols) + ", modified=?"
+        with self._conn() as c:
+            cur = c.execute(
+                f"UPDATE tasks SET {sets} WHERE unique_id=?",
+                [*cols.values(), _now(), unique_id],
+            )
+            return cur.rowcount > 0
+
+    def delete_question(self, qa_id: int) -> bool:
+        with self._conn() as c:
+            return c.execute("DELETE FROM questionanswer WHERE id=?", (qa_id,)).rowcount > 0
+
     def get_question(self, qa_id: int) -> Optional[Dict[str, Any]]:
         row = self._conn().execute(
             "SELECT * FROM questionanswer WHERE id=?", (qa_id,)
