@@ -149,3 +149,42 @@ def test_fp8_delayed_scaling_stabilizes(base_model_pair):
     cos = torch.nn.functional.cosine_similarity(a.flatten(), b.flatten(), dim=0)
     assert cos > 0.995, cos.item()
     assert torch.isfinite(b).all()
+
+
+def test_270m_forward_with_attention_maps_on_hip():
+    """The reference serving forward always passes
+    output_all_attention_masks=True (worker.py:288). The full 270M model on
+    GPU must serve that path through the HIP prob-export kernel (r1 fell
+    back to torch math) and return per-layer maps of the right shapes."""
+    import torch
+
+    from vilbert_multi_task_amd.config import ViLBertConfig
+    from vilbert_multi_task_amd.data.synthetic import forward_args, synthetic_batch
+    from vilbert_multi_task_amd.models import VILBertForVLTasks
+
+    torch.manual_seed(0)
+    cfg = ViLBertConfig.base_12in1()
+    model = VILBertForVLTasks(cfg).to("cuda", torch.bfloat16).eval()
+    batch = synthetic_batch(2, seed=5)
+    gbatch = {k: v.cuda() for k, v in batch.items()}
+    gbatch["features"] = gbatch["features"].to(torch.bfloat16)
+    gbatch["spatials"] = gbatch["spatials"].to(torch.bfloat16)
+    with torch.no_grad():
+        out = model(*forward_args(gbatch), output_all_attention_masks=True)
+    attn_data = out[9]
+    kinds = {}
+
+    def check(probs):
+        assert probs is not None and probs.is_cuda
+        s = probs.float().sum(-1)
+        assert (s - 1.0).abs().max().item() < 2e-2  # normalized rows
+
+    for entry in attn_data:
+        kinds[entry["type"]] = kinds.get(entry["type"], 0) + 1
+        if entry["type"] == "co":
+            check(entry["probs_tv"])
+            check(entry["probs_vt"])
+        else:
+            check(entry["probs"])
+    # 12 text self + 6 vision self + 6 co-attention layers
+    assert kinds == {"t_self": 12, "v_self": 6, "co": 6}
