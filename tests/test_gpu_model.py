@@ -170,7 +170,7 @@ def test_270m_forward_with_attention_maps_on_hip():
     gbatch["features"] = gbatch["features"].to(torch.bfloat16)
     gbatch["spatials"] = gbatch["spatials"].to(torch.bfloat16)
     with torch.no_grad():
-        out = model(*forward_args(gbatch), output_all_attention_masks=True)
+        out = model(*forward_args(gbatch, output_all_attention_masks=True))
     attn_data = out[9]
     kinds = {}
 
