@@ -39,7 +39,7 @@ def needs_build() -> bool:
     return os.path.getmtime(OUT_SO) < _newest_src_mtime()
 
 
-def build(force: bool = False, verbose: bool = True) -> str:
+def build(force: bool = False, verbose: bool = True, asan: bool = False) -> str:
     if not force and not needs_build():
         return OUT_SO
     os.makedirs(OUT_DIR, exist_ok=True)
@@ -60,6 +60,10 @@ def build(force: bool = False, verbose: bool = True) -> str:
         "-D__HIP_PLATFORM_AMD__",
         "-DTORCH_EXTENSION_NAME=vilbert_hip",
     ]
+    if asan:
+        # host-side AddressSanitizer (scripts/sanitize.sh); device code is
+        # not instrumented — -fsanitize applies to the x86 host pass only
+        cmd += ["-fsanitize=address", "-shared-libsan", "-g1"]
     for inc in torch_inc:
         cmd.append(f"-I{inc}")
     cmd += [os.path.join(CSRC, s) for s in SOURCES]
@@ -83,5 +87,5 @@ def build(force: bool = False, verbose: bool = True) -> str:
 
 
 if __name__ == "__main__":
-    build(force="--force" in sys.argv)
+    build(force="--force" in sys.argv, asan="--asan" in sys.argv)
     print(f"built {OUT_SO}")
