@@ -296,7 +296,7 @@ __global__ __launch_bounds__(512) void gemm256_kernel(
 //   end ph1: vmcnt(4)  -> A pieces 1,3 of THIS ktile landed (read at ph2)
 //   end ph3: vmcnt(2)  -> B0-3 + A0,2 of the NEXT ktile landed (read ph0)
 // ---------------------------------------------------------------------------
-template <bool GELU_, bool RES>
+template <bool GELU_, bool RES, int SCHED = 0>
 __global__ __launch_bounds__(512) void gemm256p_kernel(
     const bf16* __restrict__ x, const bf16* __restrict__ w,
     const bf16* __restrict__ bias, const bf16* __restrict__ res,
@@ -407,6 +407,16 @@ __global__ __launch_bounds__(512) void gemm256p_kernel(
   };
 #define PH_BAR() __builtin_amdgcn_s_barrier()
 #define LGKM0() asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory")
+  // pre-MFMA sync by schedule form (see launcher): SCHED 1 defers the lgkm
+  // drain past the barrier (read latency spans it); SCHED 2 drops the
+  // post-MFMA barrier (safe only with the drain BEFORE the remaining one:
+  // a wave ahead in the next k-tile stages into the buffer a lagging
+  // wave's un-drained reads still target)
+#define PH_SYNC() do {                                                        \
+    if (SCHED == 1) { PH_BAR(); LGKM0(); }                                    \
+    else { LGKM0(); PH_BAR(); }                                               \
+  } while (0)
+#define PH_END() do { if (SCHED != 2) PH_BAR(); } while (0)
 
   int buf = 0;
   for (int t = 0; t < ntiles; ++t) {
@@ -415,30 +425,32 @@ __global__ __launch_bounds__(512) void gemm256p_kernel(
     // ph0: (0,0)
     rdA(buf, 0); rdB(buf, 0, b0_);
     if (st) { stageB(nb, t + 1, 0); stageB(nb, t + 1, 1); }
-    LGKM0(); PH_BAR();
+    PH_SYNC();
     mm(0, 0, b0_);
-    PH_BAR();
+    PH_END();
     // ph1: (0,1)
     rdB(buf, 1, b1_);
     if (st) { stageB(nb, t + 1, 2); stageB(nb, t + 1, 3); }
-    LGKM0(); PH_BAR();
+    PH_SYNC();
     mm(0, 1, b1_);
     asm volatile("s_waitcnt vmcnt(4)" ::: "memory");  // own A1,A3 landed
-    PH_BAR();
+    PH_END();
     // ph2: (1,0)
     rdA(buf, 1);
     if (st) { stageA(nb, t + 1, 0); stageA(nb, t + 1, 2); }
-    LGKM0(); PH_BAR();
+    PH_SYNC();
     mm(1, 0, b0_);
-    PH_BAR();
+    PH_END();
     // ph3: (1,1)
     if (st) { stageA(nb, t + 1, 1); stageA(nb, t + 1, 3); }
-    LGKM0(); PH_BAR();
+    PH_SYNC();
     mm(1, 1, b1_);
     asm volatile("s_waitcnt vmcnt(2)" ::: "memory");  // next ktile's B0-3,A0,A2
-    PH_BAR();
+    PH_END();
     buf = nb;
   }
+#undef PH_SYNC
+#undef PH_END
   asm volatile("s_waitcnt vmcnt(0)" ::: "memory");  // drain tail stages
 #undef PH_BAR
 #undef LGKM0
@@ -515,17 +527,25 @@ void launch_gemm256(const bf16* x, const bf16* w, const bf16* bias,
   // back to the simple 2-phase loop (A/B + bisection lever)
   static const int pipe = [] {
     const char* e = getenv("VILBERT_GEMM_PIPE");
-    return e ? atoi(e) : 1;
+    return e ? atoi(e) : 1;  // 0 = 2-phase loop; 1/2/3 = 8-phase SCHED 0/1/2
   }();
 #define L(KRN, G, R)                                                         \
   hipLaunchKernelGGL((KRN<G, R>), grid, dim3(512), lds, stream,              \
                      x, w, bias, res, out, (int)M, (int)N, (int)K, gx, gy)
-  if (pipe) {
-    if (gelu) { if (res) L(gemm256p_kernel, true, true); else L(gemm256p_kernel, true, false); }
-    else      { if (res) L(gemm256p_kernel, false, true); else L(gemm256p_kernel, false, false); }
-  } else {
+#define LP(SCH)                                                              \
+  do {                                                                       \
+    if (gelu) { if (res) hipLaunchKernelGGL((gemm256p_kernel<true, true, SCH>), grid, dim3(512), lds, stream, x, w, bias, res, out, (int)M, (int)N, (int)K, gx, gy); \
+                else hipLaunchKernelGGL((gemm256p_kernel<true, false, SCH>), grid, dim3(512), lds, stream, x, w, bias, res, out, (int)M, (int)N, (int)K, gx, gy); }  \
+    else      { if (res) hipLaunchKernelGGL((gemm256p_kernel<false, true, SCH>), grid, dim3(512), lds, stream, x, w, bias, res, out, (int)M, (int)N, (int)K, gx, gy); \
+                else hipLaunchKernelGGL((gemm256p_kernel<false, false, SCH>), grid, dim3(512), lds, stream, x, w, bias, res, out, (int)M, (int)N, (int)K, gx, gy); } \
+  } while (0)
+  if (pipe == 1) LP(0);
+  else if (pipe == 2) LP(1);
+  else if (pipe == 3) LP(2);
+  else {
     if (gelu) { if (res) L(gemm256_kernel, true, true); else L(gemm256_kernel, true, false); }
     else      { if (res) L(gemm256_kernel, false, true); else L(gemm256_kernel, false, false); }
   }
+#undef LP
 #undef L
 }
