@@ -395,14 +395,35 @@ __global__ __launch_bounds__(512) void gemm256p_kernel(
   };
   auto mm = [&](int mh, int nh, bf16x8 (&bf_)[4]) {
     __builtin_amdgcn_s_setprio(1);
+    // kk OUTER: 8 independent accumulators between reuses of one (mtl,ntl)
+    // (kk-inner issued back-to-back dependent MFMAs on the same acc)
 #pragma unroll
-    for (int mtl = 0; mtl < 4; ++mtl)
+    for (int kk = 0; kk < 2; ++kk)
 #pragma unroll
-      for (int ntl = 0; ntl < 2; ++ntl)
+      for (int mtl = 0; mtl < 4; ++mtl)
 #pragma unroll
-        for (int kk = 0; kk < 2; ++kk)
+        for (int ntl = 0; ntl < 2; ++ntl)
           acc[mh * 4 + mtl][nh * 2 + ntl] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               a_[mtl * 2 + kk], bf_[ntl * 2 + kk], acc[mh * 4 + mtl][nh * 2 + ntl], 0, 0, 0);
+    __builtin_amdgcn_s_setprio(0);
+  };
+  // whole-mh variant for the 2-phase-per-ktile schedule: 32 MFMAs per
+  // cluster over both B halves (SCHED 3 uses 2 barriers per ktile, not 8)
+  auto mm2 = [&](int mh) {
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk)
+#pragma unroll
+      for (int mtl = 0; mtl < 4; ++mtl) {
+#pragma unroll
+        for (int ntl = 0; ntl < 2; ++ntl)
+          acc[mh * 4 + mtl][ntl] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a_[mtl * 2 + kk], b0_[ntl * 2 + kk], acc[mh * 4 + mtl][ntl], 0, 0, 0);
+#pragma unroll
+        for (int ntl = 0; ntl < 2; ++ntl)
+          acc[mh * 4 + mtl][2 + ntl] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a_[mtl * 2 + kk], b1_[ntl * 2 + kk], acc[mh * 4 + mtl][2 + ntl], 0, 0, 0);
+      }
     __builtin_amdgcn_s_setprio(0);
   };
 #define PH_BAR() __builtin_amdgcn_s_barrier()
@@ -419,6 +440,29 @@ __global__ __launch_bounds__(512) void gemm256p_kernel(
 #define PH_END() do { if (SCHED != 2) PH_BAR(); } while (0)
 
   int buf = 0;
+  if (SCHED == 3) {
+    for (int t = 0; t < ntiles; ++t) {
+      const bool st = t + 1 < ntiles;
+      const int nb = buf ^ 1;
+      // ph0: mh=0 x both B halves (32 MFMAs)
+      rdA(buf, 0); rdB(buf, 0, b0_); rdB(buf, 1, b1_);
+      if (st) { stageB(nb, t + 1, 0); stageB(nb, t + 1, 1);
+                stageB(nb, t + 1, 2); stageB(nb, t + 1, 3); }
+      LGKM0(); PH_BAR();
+      mm2(0);
+      asm volatile("s_waitcnt vmcnt(4)" ::: "memory");  // own A1,A3 landed
+      PH_BAR();
+      // ph1: mh=1 (A-half swap only)
+      rdA(buf, 1);
+      if (st) { stageA(nb, t + 1, 0); stageA(nb, t + 1, 2);
+                stageA(nb, t + 1, 1); stageA(nb, t + 1, 3); }
+      LGKM0(); PH_BAR();
+      mm2(1);
+      asm volatile("s_waitcnt vmcnt(2)" ::: "memory");  // next B0-3 + A0,A2
+      PH_BAR();
+      buf = nb;
+    }
+  } else
   for (int t = 0; t < ntiles; ++t) {
     const bool st = t + 1 < ntiles;
     const int nb = buf ^ 1;
@@ -542,6 +586,7 @@ void launch_gemm256(const bf16* x, const bf16* w, const bf16* bias,
   if (pipe == 1) LP(0);
   else if (pipe == 2) LP(1);
   else if (pipe == 3) LP(2);
+  else if (pipe == 4) LP(3);
   else {
     if (gelu) { if (res) L(gemm256_kernel, true, true); else L(gemm256_kernel, true, false); }
     else      { if (res) L(gemm256_kernel, false, true); else L(gemm256_kernel, false, false); }
