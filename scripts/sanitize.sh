@@ -19,5 +19,5 @@ python -m vilbert_multi_task_amd.ops.build --force --asan
 ASAN_LIB=$(ls /opt/rocm/lib/llvm/lib/clang/*/lib/linux/libclang_rt.asan-x86_64.so | head -1)
 echo "== running GPU op tests under host ASan ($ASAN_LIB) =="
 LD_PRELOAD="$ASAN_LIB" \
-ASAN_OPTIONS=detect_leaks=0:protect_shadow_gap=0:replace_intrin=0:alloc_dealloc_mismatch=0 \
+ASAN_OPTIONS=detect_leaks=0:protect_shadow_gap=0:replace_intrin=0:alloc_dealloc_mismatch=0:allocator_may_return_null=1 \
 python -m pytest tests/test_gpu_ops.py -q "${@:--x}"
