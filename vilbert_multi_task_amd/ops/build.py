@@ -39,7 +39,8 @@ def needs_build() -> bool:
     return os.path.getmtime(OUT_SO) < _newest_src_mtime()
 
 
-def build(force: bool = False, verbose: bool = True, asan: bool = False) -> str:
+def build(force: bool = False, verbose: bool = True, asan: bool = False,
+          ubsan: bool = False) -> str:
     if not force and not needs_build():
         return OUT_SO
     os.makedirs(OUT_DIR, exist_ok=True)
@@ -64,6 +65,11 @@ def build(force: bool = False, verbose: bool = True, asan: bool = False) -> str:
         # host-side AddressSanitizer (scripts/sanitize.sh); device code is
         # not instrumented — -fsanitize applies to the x86 host pass only
         cmd += ["-fsanitize=address", "-shared-libsan", "-g1"]
+    if ubsan:
+        # UndefinedBehaviorSanitizer, host side: no allocator interception,
+        # coexists with the HIP runtime (ASan's interceptors do not)
+        cmd += ["-fsanitize=undefined", "-fno-sanitize=vptr,function",
+                "-fno-sanitize-recover=undefined", "-shared-libsan", "-g1"]
     for inc in torch_inc:
         cmd.append(f"-I{inc}")
     cmd += [os.path.join(CSRC, s) for s in SOURCES]
@@ -87,5 +93,6 @@ def build(force: bool = False, verbose: bool = True, asan: bool = False) -> str:
 
 
 if __name__ == "__main__":
-    build(force="--force" in sys.argv, asan="--asan" in sys.argv)
+    build(force="--force" in sys.argv, asan="--asan" in sys.argv,
+          ubsan="--ubsan" in sys.argv)
     print(f"built {OUT_SO}")
