@@ -89,3 +89,72 @@ def test_evaluate_script_smoke():
     )
     assert r.returncode == 0, r.stderr[-800:]
     assert '"event": "eval_summary"' in r.stdout
+
+
+def _write_corpus(root, n=6, feat_dim=64, regions_src=8):
+    """Tiny on-disk corpus fixture (data/corpus.py layout)."""
+    import json
+    import os
+
+    import numpy as np
+
+    os.makedirs(os.path.join(root, "features"), exist_ok=True)
+    os.makedirs(os.path.join(root, "boxes"), exist_ok=True)
+    rng = np.random.default_rng(3)
+    with open(os.path.join(root, "captions.jsonl"), "w") as f:
+        for i in range(n):
+            f.write(json.dumps({
+                "image_id": f"img{i}",
+                "caption": f"a dog number {i} on the grass",
+                "label": i % 3,
+            }) + "\n")
+            np.save(os.path.join(root, "features", f"img{i}.npy"),
+                    rng.standard_normal((regions_src, feat_dim)).astype("float32"))
+            np.save(os.path.join(root, "boxes", f"img{i}.npy"),
+                    rng.uniform(0, 1, (regions_src, 4)).astype("float32"))
+    return root
+
+
+def test_conceptcap_loader_file_corpus(tmp_path, tiny_config):
+    """The real-corpus backend feeds the SAME pretraining batch schema:
+    tokens/features come from disk, the masking machinery is unchanged."""
+    import numpy as np
+
+    from vilbert_multi_task_amd.data.loaders import ConceptCapLoaderTrain
+
+    root = _write_corpus(str(tmp_path), feat_dim=tiny_config.v_feature_size)
+    loader = ConceptCapLoaderTrain(
+        tiny_config, batch_size=3, num_batches=2, seq_len=20, regions=12,
+        corpus=root,
+    )
+    batches = list(loader)
+    assert len(batches) == 2
+    b = batches[0]
+    assert b["features"].shape == (3, 12, tiny_config.v_feature_size)
+    assert b["question"].shape == (3, 20)
+    # masking applied on the REAL tokens: [MASK]=103 appears where labels set
+    masked = b["lm_labels"] >= 0
+    assert (b["question"][masked] == 103).all()
+    # global region = mean of the on-disk features for that image
+    feats0 = np.load(str(tmp_path / "features" / "img0.npy"))
+    means = torch.tensor(np.stack([
+        np.load(str(tmp_path / "features" / f"img{i}.npy")).mean(0)
+        for i in range(6)
+    ]))
+    g0 = b["features"][:, 0, :]
+    # every row's global region equals one of the corpus images' mean
+    for r in range(3):
+        d = (means - g0[r]).abs().max(dim=1).values
+        assert d.min() < 1e-5
+
+
+def test_load_dataset_eval_file_corpus(tmp_path, tiny_config):
+    from vilbert_multi_task_amd.data.loaders import LoadDatasetEval
+
+    root = _write_corpus(str(tmp_path), feat_dim=tiny_config.v_feature_size)
+    it = LoadDatasetEval(tiny_config, "snli_ve", batch_size=3, corpus=root)
+    batch, targets = next(it)
+    assert targets.tolist() == [0, 1, 2]          # jsonl labels, in order
+    assert int(batch["task_tokens"][0, 0]) == 13  # snli_ve -> task 13
+    assert batch["question"].shape == (3, 37)
+    assert batch["features"].shape[0] == 3
