@@ -176,15 +176,22 @@ _LINEAR_GELU_OK = True
 
 
 def linear_bias_gelu(x: torch.Tensor, w: torch.Tensor, bias: torch.Tensor) -> torch.Tensor:
-    """y = gelu(x @ w.T + bias) — hipBLASLt fused epilogue on GPU (tanh-form
-    GELU, ~3e-3 from erf, below bf16 resolution); unfused erf path otherwise."""
+    """y = gelu(x @ w.T + bias), fused. Backend by VILBERT_GEMM_GELU:
+      - 'mfma' (default): the hand-written kernel's GELU epilogue — the
+        only path measured fault-free at EVERY serving/retrieval shape
+      - 'hipblaslt': the GELU_BIAS epilogue. r2 isolation: its autotune
+        sweep (and some winning algos) memory-fault at several large-M
+        shapes (5 of 6 probed B>=1024 shapes crash, profiles/r07 §3 class)
+      - 'torch': unfused F.linear + erf-GELU pass"""
     global _LINEAR_GELU_OK
     if _LINEAR_GELU_OK and _want_hip(x, w, bias):
         ext = _load_extension()
+        mode = os.environ.get("VILBERT_GEMM_GELU", "mfma")
         try:
-            if _mfma_gemm_mode() == "mfma" and _mfma_linear_eligible(x, w):
+            if mode == "mfma" and _mfma_linear_eligible(x, w):
                 return torch.ops.vilbert_amd.mfma_linear(x, w, bias, None, True)
-            return ext.linear_bias_gelu(x, w, bias)
+            if mode == "hipblaslt":
+                return ext.linear_bias_gelu(x, w, bias)
         except RuntimeError:
             _LINEAR_GELU_OK = False  # no algo for this arch/shape: fall back
     if _want_hip_train(x, w, bias):
