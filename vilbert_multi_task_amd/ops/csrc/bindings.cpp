@@ -449,7 +449,12 @@ at::Tensor linear_bias_gelu(const at::Tensor& x, const at::Tensor& w,
   auto sizes = x.sizes().vec();
   sizes.back() = N;
   auto y = at::empty(sizes, x.options());
-  constexpr size_t kWs = 32L * 1024 * 1024;
+  // 256 MB (was 32): hipBLASLt's heuristic can select algos whose real
+  // workspace need exceeds the preference cap at large M — a 32 MB
+  // workspace then overruns (standalone: "write access to a read-only
+  // page"; inside the allocator pool: silent scribble). See the r2 fault
+  // isolation in profiles/r07 §3.
+  constexpr size_t kWs = 256L * 1024 * 1024;
   hipStream_t stream = cur_stream();
   void* ws = ws_for_stream(stream, x.options(), kWs);
   int rc = hipblaslt_linear_gelu(
@@ -524,7 +529,12 @@ at::Tensor fp8_linear(const at::Tensor& x8, const at::Tensor& w8,
   auto sizes = x8.sizes().vec();
   sizes.back() = N;
   auto y = at::empty(sizes, x8.options().dtype(at::kBFloat16));
-  constexpr size_t kWs = 32L * 1024 * 1024;
+  // 256 MB (was 32): hipBLASLt's heuristic can select algos whose real
+  // workspace need exceeds the preference cap at large M — a 32 MB
+  // workspace then overruns (standalone: "write access to a read-only
+  // page"; inside the allocator pool: silent scribble). See the r2 fault
+  // isolation in profiles/r07 §3.
+  constexpr size_t kWs = 256L * 1024 * 1024;
   hipStream_t stream = cur_stream();
   void* ws = ws_for_stream(stream, x8.options(), kWs);
   int rc = hipblaslt_fp8_linear(
@@ -546,7 +556,12 @@ at::Tensor fp8_linear_gelu_fp8out(const at::Tensor& x8, const at::Tensor& w8,
   auto sizes = x8.sizes().vec();
   sizes.back() = N;
   auto y8 = at::empty(sizes, x8.options());
-  constexpr size_t kWs = 32L * 1024 * 1024;
+  // 256 MB (was 32): hipBLASLt's heuristic can select algos whose real
+  // workspace need exceeds the preference cap at large M — a 32 MB
+  // workspace then overruns (standalone: "write access to a read-only
+  // page"; inside the allocator pool: silent scribble). See the r2 fault
+  // isolation in profiles/r07 §3.
+  constexpr size_t kWs = 256L * 1024 * 1024;
   hipStream_t stream = cur_stream();
   void* ws = ws_for_stream(stream, x8.options(), kWs);
   int rc = hipblaslt_fp8_linear_gelu_fp8out(
@@ -568,7 +583,12 @@ at::Tensor linear_bias(const at::Tensor& x, const at::Tensor& w,
   auto sizes = x.sizes().vec();
   sizes.back() = N;
   auto y = at::empty(sizes, x.options());
-  constexpr size_t kWs = 32L * 1024 * 1024;
+  // 256 MB (was 32): hipBLASLt's heuristic can select algos whose real
+  // workspace need exceeds the preference cap at large M — a 32 MB
+  // workspace then overruns (standalone: "write access to a read-only
+  // page"; inside the allocator pool: silent scribble). See the r2 fault
+  // isolation in profiles/r07 §3.
+  constexpr size_t kWs = 256L * 1024 * 1024;
   hipStream_t stream = cur_stream();
   void* ws = ws_for_stream(stream, x.options(), kWs);
   int rc = hipblaslt_linear_bias(
@@ -591,7 +611,12 @@ at::Tensor linear_bias_residual(const at::Tensor& x, const at::Tensor& w,
   auto sizes = x.sizes().vec();
   sizes.back() = N;
   auto y = at::empty(sizes, x.options());
-  constexpr size_t kWs = 32L * 1024 * 1024;
+  // 256 MB (was 32): hipBLASLt's heuristic can select algos whose real
+  // workspace need exceeds the preference cap at large M — a 32 MB
+  // workspace then overruns (standalone: "write access to a read-only
+  // page"; inside the allocator pool: silent scribble). See the r2 fault
+  // isolation in profiles/r07 §3.
+  constexpr size_t kWs = 256L * 1024 * 1024;
   hipStream_t stream = cur_stream();
   void* ws = ws_for_stream(stream, x.options(), kWs);
   int rc2 = hipblaslt_linear_bias_add(

@@ -177,23 +177,17 @@ _LINEAR_GELU_OK = True
 
 def linear_bias_gelu(x: torch.Tensor, w: torch.Tensor, bias: torch.Tensor) -> torch.Tensor:
     """y = gelu(x @ w.T + bias), fused. Backend by VILBERT_GEMM_GELU:
-      - 'mfma' (default): the hand-written kernel's GELU epilogue — the
-        only path measured fault-free at EVERY serving/retrieval shape
-      - 'hipblaslt': the GELU_BIAS epilogue. r2 isolation: its autotune
-        sweep (and some winning algos) memory-fault at several large-M
-        shapes (5 of 6 probed B>=1024 shapes crash, profiles/r07 §3 class)
+      - 'hipblaslt' (default): the GELU_BIAS fused epilogue
+      - 'mfma': the hand-written kernel's GELU epilogue
       - 'torch': unfused F.linear + erf-GELU pass"""
     global _LINEAR_GELU_OK
     if _LINEAR_GELU_OK and _want_hip(x, w, bias):
         ext = _load_extension()
-        mode = os.environ.get("VILBERT_GEMM_GELU", "auto")
-        if mode == "auto":
-            # every isolated hipBLASLt GELU_BIAS fault is at M >= 77824
-            # (and 103424 standalone); M = 38912 has been fault-free across
-            # every standalone sweep and bench run. Keep the faster fused
-            # epilogue below the safety line, unfused torch above it.
-            M = x.numel() // x.shape[-1]
-            mode = "hipblaslt" if M < 50000 else "torch"
+        mode = os.environ.get("VILBERT_GEMM_GELU", "hipblaslt")
+        # (r2: the "GELU_BIAS faults at large M" were a 32 MB WORKSPACE
+        # OVERRUN — the heuristic picks algos needing more at large M.
+        # Fixed with a 256 MB workspace in bindings.cpp; every previously
+        # faulting shape passes with correct numerics.)
         try:
             if mode == "mfma" and _mfma_linear_eligible(x, w):
                 return torch.ops.vilbert_amd.mfma_linear(x, w, bias, None, True)
@@ -260,13 +254,17 @@ def linear_bias_residual(
     epilogue does not."""
     global _LINEAR_RES_OK
     if _LINEAR_RES_OK and _want_hip(x, w, bias, residual):
+        ext = _load_extension()
+        backend = os.environ.get("VILBERT_RES_BACKEND", "hipblaslt")
         try:
+            if backend == "hipblaslt":
+                # beta=1 BIAS epilogue — the r1 faults were the 32 MB
+                # workspace overrun, fixed in bindings.cpp (256 MB)
+                return ext.linear_bias_residual(x, w, bias, residual)
             if _mfma_linear_eligible(x, w):
                 return torch.ops.vilbert_amd.mfma_linear(x, w, bias, residual, False)
         except RuntimeError:
             _LINEAR_RES_OK = False
-    # NOT ext.linear_bias_residual: hipBLASLt's beta=1 BIAS epilogue
-    # memory-faults at large M (r2 isolation; see linear_bias docstring)
     return torch.nn.functional.linear(x, w, bias) + residual
 
 
