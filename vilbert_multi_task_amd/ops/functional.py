@@ -206,14 +206,14 @@ _LINEAR_OK = True
 
 def linear_bias(x: torch.Tensor, w: torch.Tensor, bias: torch.Tensor) -> torch.Tensor:
     """Serving linear. Backend by VILBERT_GEMM:
-      - 'torch' (default): torch F.linear (Tensile/hipBLASLt inside torch —
-        the fastest measured stable path at the serving shapes, r06 1.07 PF)
+      - 'torch' (default): torch F.linear — ties the in-house path end to
+        end (19.2k q/s both, r2)
+      - 'hipblaslt': the in-house autotuned BIAS-epilogue path — wins
+        1.00-1.13x per shape standalone after the r2 workspace-overrun fix
+        (the r1/r2 "epilogue faults" were a 32 MB workspace overrun at
+        large M, fixed at 256 MB in bindings.cpp)
       - 'mfma': the hand-written 256x256x64 kernel (gemm_mfma.hip) — the
-        A/B lever the north star requires
-      - 'hipblaslt': the in-house autotuned BIAS-epilogue path. ISOLATED r2:
-        hipBLASLt's BIAS epilogue (beta 0 AND 1, heuristic algos included)
-        memory-faults at large M (e.g. 38912x2304x768) — kept opt-in for
-        the fault-isolation harness only."""
+        honest-A/B lever (profiles/r07)"""
     global _LINEAR_OK
     if _LINEAR_OK and _want_hip(x, w, bias):
         ext = _load_extension()
