@@ -60,10 +60,11 @@ class GraphRunner:
         self.dtype = dtype
         self.use_graphs = use_graphs and device.startswith("cuda")
         self._graphs: Dict[int, Tuple[torch.cuda.CUDAGraph, dict, tuple]] = {}
-        # fp8: specific hipBLASLt fp8-epilogue algos fault ("write access to
-        # a read-only page") at the very large M of bucket 2048 — cap the
+        # (r2: the fp8-epilogue "faults" at bucket-2048 M sizes were the
+        # 32 MB workspace overrun, fixed at 256 MB in bindings.cpp; the cap
+        # is lifted — kept as a comment trail.) Previous note: cap the
         # bucket and serve bigger batches as chunked replays (ROADMAP.md).
-        self.max_bucket = 1024 if fp8 else 4096
+        self.max_bucket = 4096
 
     # -- bucket management -------------------------------------------------
     def bucket_for(self, batch: int) -> int:
