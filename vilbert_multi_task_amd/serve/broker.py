@@ -64,6 +64,16 @@ class Broker:
         self._conn = sqlite3.connect(path, timeout=30.0, check_same_thread=False)
         self._conn.row_factory = sqlite3.Row
         self._conn.execute("PRAGMA journal_mode=WAL")
+        # WAL + NORMAL: fsync on checkpoint, not per-commit. Process-crash
+        # durability is unchanged (the at-least-once contract's failure
+        # model); only whole-machine power loss can drop the tail of the
+        # queue. The mixed-task serving config is otherwise fsync-bound on
+        # slow disks (measured 2.0k-2.9k req/s tracking the box's disk).
+        # VILBERT_SQLITE_FULL_SYNC=1 restores FULL.
+        import os as _os
+
+        if _os.environ.get("VILBERT_SQLITE_FULL_SYNC") != "1":
+            self._conn.execute("PRAGMA synchronous=NORMAL")
         self._conn.executescript(_SCHEMA)
         self._conn.commit()
 

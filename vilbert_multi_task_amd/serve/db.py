@@ -65,6 +65,11 @@ class Database:
             conn = sqlite3.connect(self.path, timeout=30.0)
             conn.row_factory = sqlite3.Row
             conn.execute("PRAGMA journal_mode=WAL")
+            # see broker.py: WAL+NORMAL keeps process-crash durability
+            import os as _os
+
+            if _os.environ.get("VILBERT_SQLITE_FULL_SYNC") != "1":
+                conn.execute("PRAGMA synchronous=NORMAL")
             self._local.conn = conn
         return conn
 
