@@ -339,52 +339,20 @@ __global__ __launch_bounds__(THREADS) void attn_kernel(
     }
 
     // ---- normalize + store O ---------------------------------------------
-    // wide path (bf16-only, P scratch big enough): re-tile the C-layout
-    // through the per-wave P_lds (dead after PV) and emit 16 B coalesced
-    // stores — the scalar path issues 32 narrow stores per stripe per wave
-    // (4 x 32 B segments each), a measurable share of the issue-bound tail
-    if (!FP8OUT && LK_PAD >= D) {
 #pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        const int prow = (lane >> 4) * 4 + r;
-        char* rowb = P_lds + prow * (D * 2);
-        const int psw = SWZ(prow);
+    for (int r = 0; r < 4; ++r) {
+      const int row = qrow0 + (lane >> 4) * 4 + r;
+      if (row < Lq) {
+        const long obase = ((long)b * Lq + row) * HD + (long)h * D;
 #pragma unroll
-        for (int nt = 0; nt < D / 16; ++nt)
-          *reinterpret_cast<short*>(rowb + (((nt * 16 + col0) * 2) ^ psw)) =
-              (short)f2us(acc_o[nt][r] * inv_l[r]);
-      }
-      // lane-linear read-back + guarded coalesced stores (per-wave scratch:
-      // the compiler orders the same-array write->read with lgkm waits)
-#pragma unroll
-      for (int pass = 0; pass < D / 32; ++pass) {
-        const int o = pass * 1024 + lane * 16;
-        const int rl = o / (D * 2);
-        const int cb = o % (D * 2);
-        const int row = qrow0 + rl;
-        const uint4 v = *reinterpret_cast<const uint4*>(
-            P_lds + rl * (D * 2) + (cb ^ SWZ(rl)));
-        if (row < Lq) {
-          *reinterpret_cast<uint4*>(
-              out + ((long)b * Lq + row) * HD + (long)h * D + cb / 2) = v;
-        }
-      }
-    } else {
-#pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        const int row = qrow0 + (lane >> 4) * 4 + r;
-        if (row < Lq) {
-          const long obase = ((long)b * Lq + row) * HD + (long)h * D;
-#pragma unroll
-          for (int nt = 0; nt < D / 16; ++nt) {
-            const float o = acc_o[nt][r] * inv_l[r];
-            out[obase + nt * 16 + col0] = f2bf(o);
-            if (FP8OUT) {
-              fp8_amax = fmaxf(fp8_amax, fabsf(o));
-              const float c = fminf(fmaxf(o * fp8_inv, -448.f), 448.f);
-              int w = __builtin_amdgcn_cvt_pk_fp8_f32(c, c, 0, false);
-              out8[obase + nt * 16 + col0] = (unsigned char)(w & 0xff);
-            }
+        for (int nt = 0; nt < D / 16; ++nt) {
+          const float o = acc_o[nt][r] * inv_l[r];
+          out[obase + nt * 16 + col0] = f2bf(o);
+          if (FP8OUT) {
+            fp8_amax = fmaxf(fp8_amax, fabsf(o));
+            const float c = fminf(fmaxf(o * fp8_inv, -448.f), 448.f);
+            int w = __builtin_amdgcn_cvt_pk_fp8_f32(c, c, 0, false);
+            out8[obase + nt * 16 + col0] = (unsigned char)(w & 0xff);
           }
         }
       }
