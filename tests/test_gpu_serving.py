@@ -104,8 +104,11 @@ def _scaleout_worker_body(queue_path, db_path, n_expect, q):
     torch.manual_seed(0)
     cfg = ViLBertConfig.tiny()
     model = VILBertForVLTasks(cfg).to("cuda", torch.bfloat16).eval()
+    # eager in the children: concurrent per-process hipGraph capture on one
+    # device proved flaky on this stack; the 270M single-process graph path
+    # is covered by test_gpu_serving_end_to_end
     runner = GraphRunner(
-        model, device="cuda", use_graphs=True, feat_dim=cfg.v_feature_size,
+        model, device="cuda", use_graphs=False, feat_dim=cfg.v_feature_size,
         seq_len=20, regions=12,
     )
     worker = ServingWorker(
