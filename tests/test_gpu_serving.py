@@ -102,13 +102,21 @@ def _scaleout_worker_body(queue_path, db_path, n_expect, q):
     from vilbert_multi_task_amd.serve.worker import ServingWorker
 
     torch.manual_seed(0)
+    # GPU-valid small config: the CPU tiny() has head_dim 16/24 which the
+    # attention kernel rejects (D must be 64/128)
     cfg = ViLBertConfig.tiny()
+    cfg.hidden_size = 128
+    cfg.num_attention_heads = 2       # D = 64
+    cfg.intermediate_size = 256
+    cfg.v_hidden_size = 128
+    cfg.v_num_attention_heads = 1     # D = 128
+    cfg.v_intermediate_size = 128
+    cfg.bi_hidden_size = 128
+    cfg.bi_num_attention_heads = 2    # D = 64
+    cfg.bi_intermediate_size = 128
     model = VILBertForVLTasks(cfg).to("cuda", torch.bfloat16).eval()
-    # eager in the children: concurrent per-process hipGraph capture on one
-    # device proved flaky on this stack; the 270M single-process graph path
-    # is covered by test_gpu_serving_end_to_end
     runner = GraphRunner(
-        model, device="cuda", use_graphs=False, feat_dim=cfg.v_feature_size,
+        model, device="cuda", use_graphs=True, feat_dim=cfg.v_feature_size,
         seq_len=20, regions=12,
     )
     worker = ServingWorker(
