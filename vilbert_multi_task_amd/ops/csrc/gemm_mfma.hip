@@ -560,6 +560,197 @@ __global__ __launch_bounds__(512) void gemm256p_kernel(
 #undef BBUF
 }
 
+// ---------------------------------------------------------------------------
+// 4-wave variant: the structure rocBLAS/Tensile's winning MT256x256x64 kernel
+// uses on this chip (256 threads, giant register tile). Wave grid 2(M)x2(N),
+// per-wave output 128x128 = acc[8][8] f32x4 = 256 accumulation registers —
+// at __launch_bounds__(256,1) there is 1 wave/SIMD and the full 512-register
+// unified file, so the accumulators live in AGPRs and the arch VGPRs carry
+// fragments + addressing. vs the 8-wave kernel: 2x the MFMA work per LDS
+// byte read (128 MFMAs vs 64 per wave per ktile against the same 16 KiB of
+// A reads + half the B traffic), and ONE barrier per ktile instead of 8 —
+// the r08 PMC profile showed gemm256p's waves parked 37% of cycles at the
+// per-phase barriers.
+// Schedule per ktile t: {16 glds -> buf^1 (tile t+1, in flight under this
+// tile's 128 MFMAs) | 2x(16 ds_read + 64 MFMA) | vmcnt(0) | barrier}. The
+// counted-wait subtlety of the 8-phase schedule is unnecessary here: at the
+// single wait point the only outstanding vmem IS the prefetch, and it had
+// the whole MFMA cluster (~2k cycles) to land.
+// ---------------------------------------------------------------------------
+template <bool GELU_, bool RES>
+__global__ __launch_bounds__(256, 1) void gemm256w_kernel(
+    const bf16* __restrict__ x, const bf16* __restrict__ w,
+    const bf16* __restrict__ bias, const bf16* __restrict__ res,
+    bf16* __restrict__ out, int M, int N, int K, int gx, int gy) {
+  const int tid = threadIdx.x;
+  const int lane = lane_id();
+  const int wid = __builtin_amdgcn_readfirstlane(wave_id());
+  const int wm = wid >> 1;  // 0..1 (M half)
+  const int wn = wid & 1;   // 0..1 (N half)
+
+  int id = blockIdx.x;
+  {
+    const int nwg = gx * gy;
+    const int q = nwg / 8, r = nwg % 8;
+    const int xcd = id % 8;
+    id = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + id / 8;
+  }
+  const int blk_m = id % gx;
+  const int blk_n = id / gx;
+  const int row0 = blk_m * 256;
+  const int col0 = blk_n * 256;
+
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+#define ABUF(b) (smem + (b) * 32768)
+#define BBUF(b) (smem + 65536 + (b) * 32768)
+
+  // staging geometry: piece g = 4 KiB (256 threads x 16 B), 8 pieces/operand
+  long srcA[8], srcB[8];
+#pragma unroll
+  for (int g = 0; g < 8; ++g) {
+    const int o = g * 4096 + tid * 16;
+    const int os = gswz(o);
+    const int r = os >> 7;
+    const int kb = os & 127;
+    srcA[g] = (long)min(row0 + r, M - 1) * (K * 2) + kb;
+    srcB[g] = (long)min(col0 + r, N - 1) * (K * 2) + kb;
+  }
+  const char* xB = reinterpret_cast<const char*>(x);
+  const char* wB = reinterpret_cast<const char*>(w);
+
+  int offA[2][8], offB[2][8];
+#pragma unroll
+  for (int kk = 0; kk < 2; ++kk) {
+    const int kb = (kk * 32 + (lane >> 4) * 8) * 2;
+#pragma unroll
+    for (int mt = 0; mt < 8; ++mt)
+      offA[kk][mt] = gswz((wm * 128 + mt * 16 + (lane & 15)) * 128 + kb);
+#pragma unroll
+    for (int nt = 0; nt < 8; ++nt)
+      offB[kk][nt] = gswz((wn * 128 + nt * 16 + (lane & 15)) * 128 + kb);
+  }
+
+  float bias_v[8];
+#pragma unroll
+  for (int nt = 0; nt < 8; ++nt) {
+    const int c = col0 + wn * 128 + nt * 16 + (lane & 15);
+    bias_v[nt] = bias ? bf2f(bias[min(c, N - 1)]) : 0.f;
+  }
+
+  auto stage = [&](int buf, int t) {
+    const long kadv = (long)t * 128;
+#pragma unroll
+    for (int g = 0; g < 8; ++g) {
+      glds16_asm(reinterpret_cast<const bf16*>(xB + srcA[g] + kadv),
+                 lds_byte_addr(ABUF(buf) + g * 4096) + wid * 1024);
+      glds16_asm(reinterpret_cast<const bf16*>(wB + srcB[g] + kadv),
+                 lds_byte_addr(BBUF(buf) + g * 4096) + wid * 1024);
+    }
+  };
+
+  f32x4 acc[8][8];
+#pragma unroll
+  for (int mt = 0; mt < 8; ++mt)
+#pragma unroll
+    for (int nt = 0; nt < 8; ++nt) acc[mt][nt] = {0.f, 0.f, 0.f, 0.f};
+
+  const int ntiles = K >> 6;
+
+  stage(0, 0);
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+
+  int buf = 0;
+  for (int t = 0; t < ntiles; ++t) {
+    if (t + 1 < ntiles) stage(buf ^ 1, t + 1);
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+      bf16x8 af[8], bfr[8];
+#pragma unroll
+      for (int mt = 0; mt < 8; ++mt)
+        af[mt] = lds_b128_g(ABUF(buf) + offA[kk][mt]);
+#pragma unroll
+      for (int nt = 0; nt < 8; ++nt)
+        bfr[nt] = lds_b128_g(BBUF(buf) + offB[kk][nt]);
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int mt = 0; mt < 8; ++mt)
+#pragma unroll
+        for (int nt = 0; nt < 8; ++nt)
+          acc[mt][nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[mt], bfr[nt], acc[mt][nt], 0, 0, 0);
+      __builtin_amdgcn_s_setprio(0);
+    }
+    // prefetch landed (only outstanding vmem); all waves' reads of buf done
+    asm volatile("s_waitcnt vmcnt(0) lgkmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    buf ^= 1;
+  }
+
+  // ---- epilogue: per-wave 32 KiB scratch (128x128 bf16), re-tile + store --
+  char* scratch = smem + wid * 32768;
+#pragma unroll
+  for (int mt = 0; mt < 8; ++mt) {
+#pragma unroll
+    for (int nt = 0; nt < 8; ++nt) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        float v = acc[mt][nt][r] + bias_v[nt];
+        if (GELU_) v = 0.5f * v * (1.0f + erff(v * KINVSQRT2));
+        const int rl = mt * 16 + (lane >> 4) * 4 + r;
+        const int cl = nt * 16 + (lane & 15);
+        *reinterpret_cast<short*>(scratch + rl * 256 + cl * 2) =
+            (short)f2us(v);
+      }
+    }
+  }
+  __syncthreads();
+
+  const int mrow0 = row0 + wm * 128;
+  const int ncol0 = col0 + wn * 128;
+  // 32 chunks of 1 KiB; residual batched in halves of 16 to bound VGPRs
+#pragma unroll
+  for (int h = 0; h < 2; ++h) {
+    uint4 rv[16];
+    if (RES) {
+      const long last = (long)M * N - 8;
+#pragma unroll
+      for (int c = 0; c < 16; ++c) {
+        const int o = (h * 16 + c) * 1024 + lane * 16;
+        long gb = (long)min(mrow0 + (o >> 8), M - 1) * N + ncol0 + (o & 255) / 2;
+        rv[c] = *reinterpret_cast<const uint4*>(res + (gb < last ? gb : last));
+      }
+    }
+#pragma unroll
+    for (int c = 0; c < 16; ++c) {
+      const int o = (h * 16 + c) * 1024 + lane * 16;
+      const int row = mrow0 + (o >> 8);
+      const int col = ncol0 + (o & 255) / 2;
+      if (row >= M) continue;
+      uint4 v = *reinterpret_cast<uint4*>(scratch + o);
+      const long gb = (long)row * N + col;
+      if (col + 8 <= N) {
+        if (RES) {
+          v.x = bfadd2(v.x, rv[c].x);
+          v.y = bfadd2(v.y, rv[c].y);
+          v.z = bfadd2(v.z, rv[c].z);
+          v.w = bfadd2(v.w, rv[c].w);
+        }
+        *reinterpret_cast<uint4*>(out + gb) = v;
+      } else if (col < N) {
+        unsigned int arr[4] = {v.x, v.y, v.z, v.w};
+        for (int j = 0; j < N - col && j < 8; ++j) {
+          float vv = us2f((unsigned short)(arr[j >> 1] >> ((j & 1) * 16)));
+          if (RES) vv += bf2f(res[gb + j]);
+          out[gb + j] = f2bf(vv);
+        }
+      }
+    }
+  }
+#undef ABUF
+#undef BBUF
+}
+
 void launch_gemm256(const bf16* x, const bf16* w, const bf16* bias,
                     const bf16* res, bf16* out, long M, long N, long K,
                     bool gelu, hipStream_t stream) {
@@ -587,6 +778,14 @@ void launch_gemm256(const bf16* x, const bf16* w, const bf16* bias,
   else if (pipe == 2) LP(1);
   else if (pipe == 3) LP(2);
   else if (pipe == 4) LP(3);
+  else if (pipe == 5) {  // 4-wave / AGPR-accumulator variant (256 threads)
+#define LW(G, R)                                                             \
+    hipLaunchKernelGGL((gemm256w_kernel<G, R>), grid, dim3(256), lds, stream, \
+                       x, w, bias, res, out, (int)M, (int)N, (int)K, gx, gy)
+    if (gelu) { if (res) LW(true, true); else LW(true, false); }
+    else      { if (res) LW(false, true); else LW(false, false); }
+#undef LW
+  }
   else {
     if (gelu) { if (res) L(gemm256_kernel, true, true); else L(gemm256_kernel, true, false); }
     else      { if (res) L(gemm256_kernel, false, true); else L(gemm256_kernel, false, false); }
