@@ -577,7 +577,7 @@ __global__ __launch_bounds__(512) void gemm256p_kernel(
 // single wait point the only outstanding vmem IS the prefetch, and it had
 // the whole MFMA cluster (~2k cycles) to land.
 // ---------------------------------------------------------------------------
-template <bool GELU_, bool RES>
+template <bool GELU_, bool RES, bool ILV = false>
 __global__ __launch_bounds__(256, 1) void gemm256w_kernel(
     const bf16* __restrict__ x, const bf16* __restrict__ w,
     const bf16* __restrict__ bias, const bf16* __restrict__ res,
@@ -660,32 +660,51 @@ __global__ __launch_bounds__(256, 1) void gemm256w_kernel(
   asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
   __builtin_amdgcn_s_barrier();
 
-  int buf = 0;
-  for (int t = 0; t < ntiles; ++t) {
-    if (t + 1 < ntiles) stage(buf ^ 1, t + 1);
+  // VILBERT_GEMM_W_ILV=1: interleave the 16 staging glds 2-per-8-MFMAs
+  // through the kk=0 cluster (sched_group_barrier pins) instead of issuing
+  // them as one burst — evens out the per-ktile 64 KB HBM demand spike all
+  // 256 CUs otherwise emit in lockstep at cluster start.
+  auto cluster = [&](int cbuf, int t1, bool stg, bool spread) {
 #pragma unroll
     for (int kk = 0; kk < 2; ++kk) {
       bf16x8 af[8], bfr[8];
 #pragma unroll
       for (int mt = 0; mt < 8; ++mt)
-        af[mt] = lds_b128_g(ABUF(buf) + offA[kk][mt]);
+        af[mt] = lds_b128_g(ABUF(cbuf) + offA[kk][mt]);
 #pragma unroll
       for (int nt = 0; nt < 8; ++nt)
-        bfr[nt] = lds_b128_g(BBUF(buf) + offB[kk][nt]);
+        bfr[nt] = lds_b128_g(BBUF(cbuf) + offB[kk][nt]);
+      if (stg && !spread && kk == 0) stage(cbuf ^ 1, t1);
       __builtin_amdgcn_s_setprio(1);
 #pragma unroll
-      for (int mt = 0; mt < 8; ++mt)
+      for (int mt = 0; mt < 8; ++mt) {
 #pragma unroll
         for (int nt = 0; nt < 8; ++nt)
           acc[mt][nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               af[mt], bfr[nt], acc[mt][nt], 0, 0, 0);
+        if (stg && spread && kk == 0) {
+          __builtin_amdgcn_sched_group_barrier(0x8 /*MFMA*/, 8, 0);
+          const long kadv = (long)t1 * 128;
+          glds16_asm(reinterpret_cast<const bf16*>(xB + srcA[mt] + kadv),
+                     lds_byte_addr(ABUF(cbuf ^ 1) + mt * 4096) + wid * 1024);
+          glds16_asm(reinterpret_cast<const bf16*>(wB + srcB[mt] + kadv),
+                     lds_byte_addr(BBUF(cbuf ^ 1) + mt * 4096) + wid * 1024);
+        }
+      }
       __builtin_amdgcn_s_setprio(0);
     }
+  };
+  int buf = 0;
+  for (int t = 0; t < ntiles - 1; ++t) {
+    cluster(buf, t + 1, true, ILV);
     // prefetch landed (only outstanding vmem); all waves' reads of buf done
     asm volatile("s_waitcnt vmcnt(0) lgkmcnt(0)" ::: "memory");
     __builtin_amdgcn_s_barrier();
     buf ^= 1;
   }
+  cluster(buf, 0, false, false);
+  asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
 
   // ---- epilogue: per-wave 32 KiB scratch (128x128 bf16), re-tile + store --
   char* scratch = smem + wid * 32768;
@@ -778,12 +797,18 @@ void launch_gemm256(const bf16* x, const bf16* w, const bf16* bias,
   else if (pipe == 2) LP(1);
   else if (pipe == 3) LP(2);
   else if (pipe == 4) LP(3);
-  else if (pipe == 5) {  // 4-wave / AGPR-accumulator variant (256 threads)
-#define LW(G, R)                                                             \
-    hipLaunchKernelGGL((gemm256w_kernel<G, R>), grid, dim3(256), lds, stream, \
-                       x, w, bias, res, out, (int)M, (int)N, (int)K, gx, gy)
-    if (gelu) { if (res) LW(true, true); else LW(true, false); }
-    else      { if (res) LW(false, true); else LW(false, false); }
+  else if (pipe == 5 || pipe == 6) {  // 4-wave / AGPR variant (256 threads);
+                                       // 6 = glds interleaved 2-per-8-MFMAs
+#define LW(G, R, I)                                                          \
+    hipLaunchKernelGGL((gemm256w_kernel<G, R, I>), grid, dim3(256), lds,     \
+                       stream, x, w, bias, res, out, (int)M, (int)N, (int)K, gx, gy)
+    if (pipe == 6) {
+      if (gelu) { if (res) LW(true, true, true); else LW(true, false, true); }
+      else      { if (res) LW(false, true, true); else LW(false, false, true); }
+    } else {
+      if (gelu) { if (res) LW(true, true, false); else LW(true, false, false); }
+      else      { if (res) LW(false, true, false); else LW(false, false, false); }
+    }
 #undef LW
   }
   else {
