@@ -149,6 +149,9 @@ def _attn_ref(q, k, v, heads, mask_bias):
         (2, 8, 101, 38, 128, 2),   # co-attn: vision queries x text keys, [B,Lq,Lk] bias
         (1, 12, 37, 37, 64, 0),    # no mask, odd length
         (5, 8, 20, 33, 128, 1),    # ragged non-multiple-of-16 lengths
+        (2, 4, 16, 32, 64, 1),     # LK_PAD=32: V/P image swizzle must stay in-row
+        (3, 4, 17, 23, 64, 1),     # LK_PAD=32 ragged (the r2 SWZR regression)
+        (2, 2, 24, 30, 128, 2),    # LK_PAD=32 at D=128, [B,Lq,Lk] bias
     ],
 )
 def test_attention_vs_reference(ext, B, H, Lq, Lk, D, mask_mode):
@@ -161,6 +164,44 @@ def test_attention_vs_reference(ext, B, H, Lq, Lk, D, mask_mode):
     elif mask_mode == 1:
         keep = torch.ones(B, Lk)
         keep[:, Lk - 5 :] = 0  # mask the tail keys
+        mask = ((1 - keep) * -1e9).to(torch.bfloat16).cuda().view(B, 1, 1, Lk)
+        ref_mask = mask.float()
+    else:
+        keep = torch.ones(B, Lq, Lk)
+        keep[:, :, Lk - 7 :] = 0
+        mask = ((1 - keep) * -1e9).to(torch.bfloat16).cuda().view(B, 1, Lq, Lk)
+        ref_mask = mask.float()
+    out = torch.ops.vilbert_amd.attention(q, k, v, H, mask)
+    ref = _attn_ref(q, k, v, H, ref_mask)
+    err = (out.float() - ref).abs().max().item()
+    assert err < 4e-2, f"max err {err}"
+
+
+@pytest.mark.parametrize(
+    "B,H,Lq,Lk,D,mask_mode",
+    [
+        (3, 12, 38, 38, 64, 1),    # text self (the SWAP win shape)
+        (2, 8, 101, 101, 128, 1),  # vision self
+        (2, 8, 38, 101, 128, 1),   # co-attn t->v
+        (2, 8, 101, 38, 128, 2),   # co-attn v->t, [B,Lq,Lk] bias
+        (1, 4, 16, 32, 64, 0),     # LK_PAD=32, no mask
+        (3, 4, 17, 23, 64, 1),     # LK_PAD=32 ragged
+    ],
+)
+def test_attention_swap_vs_reference(ext, monkeypatch, B, H, Lq, Lk, D, mask_mode):
+    """The swapped-S^T register-transpose path (attention.hip SWAP=true):
+    softmax reduced across lane groups, P normalized in-lane and gathered to
+    the PV A-fragment with ds_bpermute (no P_lds)."""
+    monkeypatch.setenv("VILBERT_ATTN_SWAP", "1")
+    q = _rand_bf16(B, Lq, H * D, seed=B * 31 + Lq)
+    k = _rand_bf16(B, Lk, H * D, seed=B * 31 + Lk + 1)
+    v = _rand_bf16(B, Lk, H * D, seed=B * 31 + Lk + 2)
+    if mask_mode == 0:
+        mask = None
+        ref_mask = None
+    elif mask_mode == 1:
+        keep = torch.ones(B, Lk)
+        keep[:, Lk - 5 :] = 0
         mask = ((1 - keep) * -1e9).to(torch.bfloat16).cuda().view(B, 1, 1, Lk)
         ref_mask = mask.float()
     else:
@@ -523,6 +564,7 @@ def test_mfma_linear_no_bias(ext):
         (2, 8, 101, 101, 128),   # vision self
         (2, 8, 38, 101, 128),    # co-attn t->v (rectangular)
         (2, 8, 101, 38, 128),    # co-attn v->t
+        (2, 4, 20, 20, 64),      # LQ_PAD=LK_PAD=32 (bwd image swizzle in-row)
     ],
 )
 def test_attention_train_bwd_matches_fp32_autograd(ext, B, H, Lq, Lk, D):

@@ -48,6 +48,11 @@ DEV void attn_atomic_max_f32(float* addr, float v) {
 // with c and spreads them; it also kills the residual 2-way read conflicts
 // on K/P (16 consecutive rows now map to 16 distinct slots).
 #define SWZ(row) (((((row) & 7) ^ (((row) >> 3) & 7))) << 4)
+// row-length-masked variant for images whose row is LK_PAD*2 bytes (the V
+// transposed image and the P buffer): at LK_PAD=32 the plain SWZ (up to
+// 112) escapes the 64-byte row -> write collisions. LK_PAD >= 64 rows are
+// unaffected (mask keeps all bits).
+#define SWZR(row, rbytes) (SWZ(row) & ((rbytes) - 1))
 
 typedef __attribute__((ext_vector_type(4))) short s4_vec;
 typedef __attribute__((address_space(3))) s4_vec* lds_v4s;
@@ -66,7 +71,21 @@ DEV bf16x8 lds_b128(const char* p) {
 
 DEV void lds_store_b128(char* p, uint4 v) { *reinterpret_cast<uint4*>(p) = v; }
 
-template <int D, bool KGLOBAL, int NTMAX, bool FP8OUT = false, int THREADS = 256>
+// SWAP=true: compute S^T = K Q^T instead of S = Q K^T (same loads, swapped
+// MFMA operands). Payoffs: (1) the softmax row lives at a FIXED lane
+// (q = lane&15) so the reduction is 2 xor-shuffles (16,32) instead of the
+// 4-step group16 butterfly, and l^-1 normalization folds into P in-lane;
+// (2) P transposes to the PV A-fragment layout with 4 ds_bpermute per
+// 32-key step (cvt-pack pairs of adjacent keys, then a fixed lane remap) —
+// NO P_lds at all: the per-wave P buffer (8 KB at D=64) disappears, LDS
+// drops to K+V and occupancy rises 6 -> 10 WGs/CU for text-self.
+// Index algebra (verified): S^T C-layout puts S[q = lane&15]
+// [key = nt*16 + (lane>>4)*4 + r]; the PV A-frag wants lane (c = lane>>4,
+// q) to hold keys kk*32 + c*8 + 0..7, i.e. source lane
+// ((c&1)*2 + (j>>1))*16 + q, packed dword j&1, tile 2*kk + (c>>1).
+// Gated to the inference fast path (no probs export, no dropout, no fp8).
+template <int D, bool KGLOBAL, int NTMAX, bool FP8OUT = false, int THREADS = 256,
+          bool SWAP = false>
 __global__ __launch_bounds__(THREADS) void attn_kernel(
     const bf16* __restrict__ q, const bf16* __restrict__ kg,
     const bf16* __restrict__ vg, const bf16* __restrict__ mask,
@@ -150,7 +169,7 @@ __global__ __launch_bounds__(THREADS) void attn_kernel(
         for (int j = 0; j < 8; ++j) {
           const int d = c * 8 + j;
           *reinterpret_cast<short*>(
-              V_lds + d * (LK_PAD * 2) + ((r * 2) ^ SWZ(d))) = vv.s[j];
+              V_lds + d * (LK_PAD * 2) + ((r * 2) ^ SWZR(d, LK_PAD * 2))) = vv.s[j];
         }
       }
     }
@@ -185,8 +204,10 @@ __global__ __launch_bounds__(THREADS) void attn_kernel(
 #pragma unroll
         for (int kk = 0; kk < D / 32; ++kk) {
           const bf16x8 bk = load_bf16x8(kg + kb + kk * 32);
-          acc_s[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              (bf16x8)aq[kk], bk, acc_s[nt], 0, 0, 0);
+          acc_s[nt] = SWAP ? __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                                 bk, (bf16x8)aq[kk], acc_s[nt], 0, 0, 0)
+                           : __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                                 (bf16x8)aq[kk], bk, acc_s[nt], 0, 0, 0);
         }
       } else {
         const char* kbase = K_lds + key * (D * 2);
@@ -194,10 +215,145 @@ __global__ __launch_bounds__(THREADS) void attn_kernel(
 #pragma unroll
         for (int kk = 0; kk < D / 32; ++kk) {
           const bf16x8 bk = lds_b128(kbase + (((kk * 64) + ((lane >> 4) * 16)) ^ ksw));
-          acc_s[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              (bf16x8)aq[kk], bk, acc_s[nt], 0, 0, 0);
+          acc_s[nt] = SWAP ? __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                                 bk, (bf16x8)aq[kk], acc_s[nt], 0, 0, 0)
+                           : __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                                 (bf16x8)aq[kk], bk, acc_s[nt], 0, 0, 0);
         }
       }
+    }
+
+    if (SWAP) {
+      // S^T layout: this lane owns query q = lane&15, keys
+      // nt*16 + (lane>>4)*4 + r across (nt, r).
+      const int q16 = lane & 15;
+      const int g4 = (lane >> 4) * 4;
+      // mask + scale (key-indexed preloads, unconditional clamped)
+      if (mask_mode == 1) {
+        const long mb = (long)b * Lk;
+        float mv[NTMAX][4];
+#pragma unroll
+        for (int nt = 0; nt < NTMAX; ++nt)
+#pragma unroll
+          for (int r = 0; r < 4; ++r)
+            mv[nt][r] = bf2f(mask[mb + min(nt * 16 + g4 + r, Lk - 1)]);
+#pragma unroll
+        for (int nt = 0; nt < NTMAX; ++nt)
+#pragma unroll
+          for (int r = 0; r < 4; ++r) {
+            const int key = nt * 16 + g4 + r;
+            acc_s[nt][r] =
+                (key < Lk) ? acc_s[nt][r] * scale + mv[nt][r] : -3.0e38f;
+          }
+      } else if (mask_mode == 2) {
+        const int qrow = min(qrow0 + q16, Lq - 1);
+        const long mb = ((long)b * Lq + qrow) * Lk;
+        float mv[NTMAX][4];
+#pragma unroll
+        for (int nt = 0; nt < NTMAX; ++nt)
+#pragma unroll
+          for (int r = 0; r < 4; ++r)
+            mv[nt][r] = bf2f(mask[mb + min(nt * 16 + g4 + r, Lk - 1)]);
+#pragma unroll
+        for (int nt = 0; nt < NTMAX; ++nt)
+#pragma unroll
+          for (int r = 0; r < 4; ++r) {
+            const int key = nt * 16 + g4 + r;
+            acc_s[nt][r] =
+                (key < Lk) ? acc_s[nt][r] * scale + mv[nt][r] : -3.0e38f;
+          }
+      } else {
+#pragma unroll
+        for (int nt = 0; nt < NTMAX; ++nt)
+#pragma unroll
+          for (int r = 0; r < 4; ++r) {
+            const int key = nt * 16 + g4 + r;
+            acc_s[nt][r] = (key < Lk) ? acc_s[nt][r] * scale : -3.0e38f;
+          }
+      }
+      // softmax over keys: in-lane (nt, r) then across the 4 lane groups
+      float mx = -3.0e38f;
+#pragma unroll
+      for (int nt = 0; nt < NTMAX; ++nt) {
+        if (nt >= NT) break;
+#pragma unroll
+        for (int r = 0; r < 4; ++r) mx = fmaxf(mx, acc_s[nt][r]);
+      }
+      mx = fmaxf(mx, __shfl_xor(mx, 16));
+      mx = fmaxf(mx, __shfl_xor(mx, 32));
+      float sum = 0.f;
+#pragma unroll
+      for (int nt = 0; nt < NTMAX; ++nt) {
+        if (nt >= NT) break;
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const float p = __expf(acc_s[nt][r] - mx);
+          acc_s[nt][r] = p;
+          sum += p;
+        }
+      }
+      sum += __shfl_xor(sum, 16);
+      sum += __shfl_xor(sum, 32);
+      const float inv = 1.0f / sum;
+      // normalize in-lane (q matches this lane) and pack key pairs to bf16
+      int pk[NTMAX][2];
+#pragma unroll
+      for (int nt = 0; nt < NTMAX; ++nt) {
+        if (nt >= NT) break;
+        pk[nt][0] = (int)f2us(acc_s[nt][0] * inv) |
+                    ((int)f2us(acc_s[nt][1] * inv) << 16);
+        pk[nt][1] = (int)f2us(acc_s[nt][2] * inv) |
+                    ((int)f2us(acc_s[nt][3] * inv) << 16);
+      }
+      // PV with register-transposed P: 4 bpermute gather the A-fragment
+      const int c = lane >> 4;
+      const int idx0 = (((c & 1) * 2) * 16 + q16) * 4;       // groups 2(c&1)
+      const int idx1 = (((c & 1) * 2 + 1) * 16 + q16) * 4;   // .. and +1
+      f32x4 acc_o[D / 16];
+#pragma unroll
+      for (int nt = 0; nt < D / 16; ++nt) acc_o[nt] = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+      for (int kk = 0; kk < NTMAX / 2; ++kk) {
+        if (kk * 32 >= LK_PAD) break;
+        // the bpermute VALUE operand is evaluated in the SOURCE lane, so the
+        // tile index must be wave-uniform per instruction: gather both tiles
+        // of this 32-key step and select by the target's half (c>>1)
+        union { int d[4]; bf16x8 v; } ap;
+        const int lo0 = __builtin_amdgcn_ds_bpermute(idx0, pk[2 * kk][0]);
+        const int lo1 = __builtin_amdgcn_ds_bpermute(idx0, pk[2 * kk][1]);
+        const int lo2 = __builtin_amdgcn_ds_bpermute(idx1, pk[2 * kk][0]);
+        const int lo3 = __builtin_amdgcn_ds_bpermute(idx1, pk[2 * kk][1]);
+        const int hi0 = __builtin_amdgcn_ds_bpermute(idx0, pk[2 * kk + 1][0]);
+        const int hi1 = __builtin_amdgcn_ds_bpermute(idx0, pk[2 * kk + 1][1]);
+        const int hi2 = __builtin_amdgcn_ds_bpermute(idx1, pk[2 * kk + 1][0]);
+        const int hi3 = __builtin_amdgcn_ds_bpermute(idx1, pk[2 * kk + 1][1]);
+        const bool upper = (c >> 1) != 0;
+        ap.d[0] = upper ? hi0 : lo0;
+        ap.d[1] = upper ? hi1 : lo1;
+        ap.d[2] = upper ? hi2 : lo2;
+        ap.d[3] = upper ? hi3 : lo3;
+        const int keyoff = (kk * 64) + ((lane >> 4) * 16);
+#pragma unroll
+        for (int nt = 0; nt < D / 16; ++nt) {
+          const int d = nt * 16 + (lane & 15);
+          const bf16x8 bv =
+              lds_b128(V_lds + d * (LK_PAD * 2) + (keyoff ^ SWZR(d, LK_PAD * 2)));
+          acc_o[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ap.v, bv,
+                                                              acc_o[nt], 0, 0, 0);
+        }
+      }
+      // O store: C-layout rows are queries again; P was pre-normalized
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int row = qrow0 + (lane >> 4) * 4 + r;
+        if (row < Lq) {
+          const long obase = ((long)b * Lq + row) * HD + (long)h * D;
+#pragma unroll
+          for (int nt = 0; nt < D / 16; ++nt)
+            out[obase + nt * 16 + (lane & 15)] = f2bf(acc_o[nt][r]);
+        }
+      }
+      continue;
     }
 
     // ---- mask + softmax (rows live across the 16-lane group) -------------
@@ -274,7 +430,7 @@ __global__ __launch_bounds__(THREADS) void attn_kernel(
     for (int r = 0; r < 4; ++r) {
       const int prow = (lane >> 4) * 4 + r;
       char* prow_base = P_lds + prow * (LK_PAD * 2);
-      const int psw = SWZ(prow);
+      const int psw = SWZR(prow, LK_PAD * 2);
       float dmv[NTMAX];
       if (dropm != nullptr) {
         const int rowc = min(qrow0 + (lane >> 4) * 4 + r, Lq - 1);
@@ -322,7 +478,7 @@ __global__ __launch_bounds__(THREADS) void attn_kernel(
     for (int nt = 0; nt < D / 16; ++nt) acc_o[nt] = {0.f, 0.f, 0.f, 0.f};
     const int parow = lane & 15;
     const char* pa_base = P_lds + parow * (LK_PAD * 2);
-    const int pasw = SWZ(parow);
+    const int pasw = SWZR(parow, LK_PAD * 2);
 #pragma unroll
     for (int kk = 0; kk < NTMAX / 2; ++kk) {  // LK_PAD/32 <= NTMAX/2
       if (kk * 32 >= LK_PAD) break;
@@ -333,7 +489,7 @@ __global__ __launch_bounds__(THREADS) void attn_kernel(
       for (int nt = 0; nt < D / 16; ++nt) {
         const int d = nt * 16 + (lane & 15);
         const bf16x8 bv =
-            lds_b128(V_lds + d * (LK_PAD * 2) + (keyoff ^ SWZ(d)));
+            lds_b128(V_lds + d * (LK_PAD * 2) + (keyoff ^ SWZR(d, LK_PAD * 2)));
         acc_o[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ap, bv, acc_o[nt], 0, 0, 0);
       }
     }
@@ -534,7 +690,7 @@ __global__ __launch_bounds__(256) void attn_kernel_pipe(
     for (int r = 0; r < 4; ++r) {
       const int prow = (lane >> 4) * 4 + r;
       char* prow_base = P_lds + prow * (LK_PAD * 2);
-      const int psw = SWZ(prow);
+      const int psw = SWZR(prow, LK_PAD * 2);
 #pragma unroll
       for (int nt = 0; nt < NTMAX; ++nt) {
         if (nt >= NT) break;
@@ -552,7 +708,7 @@ __global__ __launch_bounds__(256) void attn_kernel_pipe(
     for (int nt = 0; nt < D / 16; ++nt) acc_o[nt] = {0.f, 0.f, 0.f, 0.f};
     const int parow = lane & 15;
     const char* pa_base = P_lds + parow * (LK_PAD * 2);
-    const int pasw = SWZ(parow);
+    const int pasw = SWZR(parow, LK_PAD * 2);
 #pragma unroll
     for (int kk = 0; kk < NTMAX / 2; ++kk) {
       if (kk * 32 >= LK_PAD) break;
@@ -563,7 +719,7 @@ __global__ __launch_bounds__(256) void attn_kernel_pipe(
       for (int nt = 0; nt < D / 16; ++nt) {
         const int d = nt * 16 + (lane & 15);
         const bf16x8 bv =
-            lds_b128(V_lds + d * (LK_PAD * 2) + (keyoff ^ SWZ(d)));
+            lds_b128(V_lds + d * (LK_PAD * 2) + (keyoff ^ SWZR(d, LK_PAD * 2)));
         acc_o[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ap, bv, acc_o[nt], 0, 0, 0);
       }
     }
@@ -603,7 +759,7 @@ __global__ __launch_bounds__(256) void attn_kernel_pipe(
     for (int j = 0; j < 8; ++j) {
       const int d = c * 8 + j;
       *reinterpret_cast<short*>(
-          V_lds + d * (LK_PAD * 2) + ((r * 2) ^ SWZ(d))) = vv.s[j];
+          V_lds + d * (LK_PAD * 2) + ((r * 2) ^ SWZR(d, LK_PAD * 2))) = vv.s[j];
     }
   }
   __syncthreads();  // V_lds ready for every wave
@@ -707,7 +863,7 @@ __global__ __launch_bounds__(THREADS) void attn_bhloop_kernel(
       for (int j = 0; j < 8; ++j) {
         const int d = c * 8 + j;
         *reinterpret_cast<short*>(
-            V_lds + d * (LK_PAD * 2) + ((r * 2) ^ SWZ(d))) = vv.s[j];
+            V_lds + d * (LK_PAD * 2) + ((r * 2) ^ SWZR(d, LK_PAD * 2))) = vv.s[j];
       }
     }
   };
@@ -814,7 +970,7 @@ __global__ __launch_bounds__(THREADS) void attn_bhloop_kernel(
       for (int r = 0; r < 4; ++r) {
         const int prow = (lane >> 4) * 4 + r;
         char* prow_base = P_lds + prow * (LK_PAD * 2);
-        const int psw = SWZ(prow);
+        const int psw = SWZR(prow, LK_PAD * 2);
 #pragma unroll
         for (int nt = 0; nt < NTMAX; ++nt) {
           if (nt >= NT) break;
@@ -845,7 +1001,7 @@ __global__ __launch_bounds__(THREADS) void attn_bhloop_kernel(
       for (int nt = 0; nt < D / 16; ++nt) acc_o[nt] = {0.f, 0.f, 0.f, 0.f};
       const int parow = lane & 15;
       const char* pa_base = P_lds + parow * (LK_PAD * 2);
-      const int pasw = SWZ(parow);
+      const int pasw = SWZR(parow, LK_PAD * 2);
 #pragma unroll
       for (int kk = 0; kk < NTMAX / 2; ++kk) {
         if (kk * 32 >= LK_PAD) break;
@@ -856,7 +1012,7 @@ __global__ __launch_bounds__(THREADS) void attn_bhloop_kernel(
         for (int nt = 0; nt < D / 16; ++nt) {
           const int d = nt * 16 + (lane & 15);
           const bf16x8 bv =
-              lds_b128(V_lds + d * (LK_PAD * 2) + (keyoff ^ SWZ(d)));
+              lds_b128(V_lds + d * (LK_PAD * 2) + (keyoff ^ SWZR(d, LK_PAD * 2)));
           acc_o[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ap, bv, acc_o[nt], 0, 0, 0);
         }
       }
@@ -883,6 +1039,17 @@ __global__ __launch_bounds__(THREADS) void attn_bhloop_kernel(
 // ---------------------------------------------------------------------------
 // mfma layout probe: one-wave 16x16x32 product for on-device layout checks
 // ---------------------------------------------------------------------------
+// probe: out[l] = ds_bpermute(idx[l]*4, src[l]) — verifies full-wave64
+// bpermute semantics + index addressing on this chip (SWAP-path gather).
+__global__ void bperm_probe_kernel(const int* __restrict__ idx,
+                                   const int* __restrict__ src,
+                                   int* __restrict__ outv) {
+  const int l = threadIdx.x;
+  const int v = src[l];
+  const int i = idx[l] * 4;
+  outv[l] = __builtin_amdgcn_ds_bpermute(i, v);
+}
+
 __global__ void mfma_probe_kernel(const bf16* __restrict__ a,
                                   const bf16* __restrict__ b,
                                   float* __restrict__ c) {
@@ -986,6 +1153,28 @@ void launch_attention_impl(const bf16* q, const bf16* k, const bf16* v,
 #undef LAUNCH_BHLOOP
     return;
   }
+  // swapped-S^T fast path (no P_lds; see the SWAP template doc above).
+  // VILBERT_ATTN_SWAP: 1 force-on, 0 force-off, unset = measured default.
+  const char* swap_e = getenv("VILBERT_ATTN_SWAP");
+  const int swap_env = swap_e ? atoi(swap_e) : -1;
+  const bool swap_ok = probs_out == nullptr && dropm == nullptr &&
+                       out8 == nullptr && !kglobal_env && nsplit == 1 &&
+                       !pipe_env && nwaves == 4;
+  const bool use_swap = swap_ok && (swap_env == 1);  // default off until A/B'd
+  if (use_swap) {
+    const size_t lds_swap = sizeof(bf16) * (size_t)(2 * LK_PAD * D);
+    const dim3 grids(B * H);
+#define LAUNCH_SWAP(DD, NTM)                                                  \
+    hipLaunchKernelGGL((attn_kernel<DD, false, NTM, false, 256, true>),       \
+                       grids, dim3(256), lds_swap, stream, q, k, v, mask,     \
+                       out, B, H, Lq, Lk, mask_mode, scale, qs, ks, vs, 1,    \
+                       nullptr, nullptr, nullptr, 0, nullptr, nullptr)
+    const bool smalls = LK_PAD <= 64;
+    if (D == 64) { if (smalls) LAUNCH_SWAP(64, 4); else LAUNCH_SWAP(64, 8); }
+    else         { if (smalls) LAUNCH_SWAP(128, 4); else LAUNCH_SWAP(128, 8); }
+#undef LAUNCH_SWAP
+    return;
+  }
   // (K if staged) + V + per-wave P (all bf16)
   const size_t lds = sizeof(bf16) *
       (size_t)((kglobal_env ? 1 : 2) * LK_PAD * D + nwaves * 16 * LK_PAD);
@@ -1070,6 +1259,11 @@ void launch_attention_fp8out(const bf16* q, const bf16* k, const bf16* v,
 
 void launch_mfma_probe(const bf16* a, const bf16* b, float* c, hipStream_t stream) {
   hipLaunchKernelGGL(mfma_probe_kernel, dim3(1), dim3(64), 0, stream, a, b, c);
+}
+void launch_bperm_probe(const int* idx, const int* src, int* outv,
+                        hipStream_t stream) {
+  hipLaunchKernelGGL(bperm_probe_kernel, dim3(1), dim3(64), 0, stream, idx,
+                     src, outv);
 }
 
 void launch_tr16_probe(short* out, int mode, hipStream_t stream) {

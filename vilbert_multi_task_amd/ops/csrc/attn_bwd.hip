@@ -27,6 +27,10 @@
 #include "common.h"
 
 #define SWZB(row) (((((row) & 7) ^ (((row) >> 3) & 7))) << 4)
+// row-length-masked form: the plain swizzle (up to 112 B) escapes a
+// 64-byte row at *_PAD == 32 and collides across rows (same latent bug
+// class as attention.hip's SWZR fix) — mask to the image's row length.
+#define SWZBR(row, rbytes) (SWZB(row) & ((rbytes) - 1))
 
 namespace {
 
@@ -50,7 +54,7 @@ DEV void transpose_scatter(char* img, int LPAD, int r, int c, uint4 raw) {
 #pragma unroll
   for (int j = 0; j < 8; ++j) {
     const int d = c * 8 + j;
-    *reinterpret_cast<short*>(img + d * (LPAD * 2) + ((r * 2) ^ SWZB(d))) = v.s[j];
+    *reinterpret_cast<short*>(img + d * (LPAD * 2) + ((r * 2) ^ SWZBR(d, LPAD * 2))) = v.s[j];
   }
 }
 
@@ -166,7 +170,7 @@ __global__ __launch_bounds__(256) void attn_bwd_ds_dq_kernel(
       // dS -> per-wave row image (for dQ) + global scratch (for B2)
       const int prow = (lane >> 4) * 4 + r;
       char* rowb = dS_lds + prow * (LK_PAD * 2);
-      const int psw = SWZB(prow);
+      const int psw = SWZBR(prow, LK_PAD * 2);
       const bool rowok = row < Lq;
 #pragma unroll
       for (int nt = 0; nt < NTMAX; ++nt) {
@@ -185,7 +189,7 @@ __global__ __launch_bounds__(256) void attn_bwd_ds_dq_kernel(
     for (int nt = 0; nt < D / 16; ++nt) accq[nt] = {0.f, 0.f, 0.f, 0.f};
     const int parow = lane & 15;
     const char* pa = dS_lds + parow * (LK_PAD * 2);
-    const int pasw = SWZB(parow);
+    const int pasw = SWZBR(parow, LK_PAD * 2);
 #pragma unroll
     for (int kk = 0; kk < NTMAX / 2; ++kk) {
       if (kk * 32 >= LK_PAD) break;
@@ -194,7 +198,7 @@ __global__ __launch_bounds__(256) void attn_bwd_ds_dq_kernel(
 #pragma unroll
       for (int nt = 0; nt < D / 16; ++nt) {
         const int d = nt * 16 + (lane & 15);
-        const bf16x8 bk = lds16(KT_lds + d * (LK_PAD * 2) + (keyoff ^ SWZB(d)));
+        const bf16x8 bk = lds16(KT_lds + d * (LK_PAD * 2) + (keyoff ^ SWZBR(d, LK_PAD * 2)));
         accq[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, bk, accq[nt], 0, 0, 0);
       }
     }
@@ -281,9 +285,9 @@ __global__ __launch_bounds__(256) void attn_bwd_dk_dv_kernel(
         float ptv = ok ? bf2f(probs[idx]) : 0.f;
         if (dropm != nullptr && ok) ptv *= bf2f(dropm[idx]);
         *reinterpret_cast<short*>(
-            dST_lds + key * (LQ_PAD * 2) + ((r * 2) ^ SWZB(key))) = (short)f2us(dsv);
+            dST_lds + key * (LQ_PAD * 2) + ((r * 2) ^ SWZBR(key, LQ_PAD * 2))) = (short)f2us(dsv);
         *reinterpret_cast<short*>(
-            PT_lds + key * (LQ_PAD * 2) + ((r * 2) ^ SWZB(key))) = (short)f2us(ptv);
+            PT_lds + key * (LQ_PAD * 2) + ((r * 2) ^ SWZBR(key, LQ_PAD * 2))) = (short)f2us(ptv);
       }
     }
   }
@@ -303,7 +307,7 @@ __global__ __launch_bounds__(256) void attn_bwd_dk_dv_kernel(
     const int arow = krow0 + (lane & 15);  // key row of the A fragment
     const char* dsbase = dST_lds + arow * (LQ_PAD * 2);
     const char* ptbase = PT_lds + arow * (LQ_PAD * 2);
-    const int asw = SWZB(arow);
+    const int asw = SWZBR(arow, LQ_PAD * 2);
 #pragma unroll
     for (int kk = 0; kk < NQMAX / 2; ++kk) {  // q-row slabs of 32
       if (kk * 32 >= LQ_PAD) break;
@@ -313,8 +317,8 @@ __global__ __launch_bounds__(256) void attn_bwd_dk_dv_kernel(
 #pragma unroll
       for (int nt = 0; nt < D / 16; ++nt) {
         const int d = nt * 16 + (lane & 15);
-        const bf16x8 bq = lds16(QT_lds + d * (LQ_PAD * 2) + (qoff ^ SWZB(d)));
-        const bf16x8 bo = lds16(dOT_lds + d * (LQ_PAD * 2) + (qoff ^ SWZB(d)));
+        const bf16x8 bq = lds16(QT_lds + d * (LQ_PAD * 2) + (qoff ^ SWZBR(d, LQ_PAD * 2)));
+        const bf16x8 bo = lds16(dOT_lds + d * (LQ_PAD * 2) + (qoff ^ SWZBR(d, LQ_PAD * 2)));
         acck[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_ds, bq, acck[nt], 0, 0, 0);
         accv[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_pt, bo, accv[nt], 0, 0, 0);
       }

@@ -41,6 +41,7 @@ void launch_attention_fp8out(const bf16*, const bf16*, const bf16*, const bf16*,
                              int, int, int, int, int, int, int, int, int,
                              hipStream_t);
 void launch_mfma_probe(const bf16*, const bf16*, float*, hipStream_t);
+void launch_bperm_probe(const int*, const int*, int*, hipStream_t);
 void launch_gemm256(const bf16*, const bf16*, const bf16*, const bf16*, bf16*,
                     long, long, long, bool, hipStream_t);
 void launch_adamw(bf16*, const void*, bool, float*, float*, float*, long,
@@ -414,6 +415,16 @@ at::Tensor mfma_probe(const at::Tensor& a, const at::Tensor& b) {
   return c;
 }
 
+at::Tensor bperm_probe(const at::Tensor& idx, const at::Tensor& src) {
+  TORCH_CHECK(idx.numel() == 64 && src.numel() == 64, "bperm_probe: 64 lanes");
+  auto out = at::empty({64}, idx.options());
+  launch_bperm_probe((const int*)idx.contiguous().data_ptr(),
+                     (const int*)src.contiguous().data_ptr(),
+                     (int*)out.data_ptr(),
+                     cur_stream());
+  return out;
+}
+
 at::Tensor tr16_probe(const at::Tensor& dummy, int64_t mode) {
   auto out = at::empty({64, 4}, dummy.options().dtype(at::kShort));
   launch_tr16_probe((short*)out.data_ptr(), (int)mode, cur_stream());
@@ -757,6 +768,7 @@ TORCH_LIBRARY(vilbert_amd, m) {
   m.def("attention_fp8out(Tensor q, Tensor k, Tensor v, int heads, Tensor? mask, Tensor scales, Tensor(a!) amaxes, int site) -> (Tensor, Tensor)");
   m.def("embedding_ln(Tensor ids, Tensor pos, Tensor type, Tensor word_w, Tensor pos_w, Tensor type_w, Tensor ln_w, Tensor ln_b, float eps) -> Tensor");
   m.def("mfma_probe(Tensor a, Tensor b) -> Tensor");
+  m.def("bperm_probe(Tensor idx, Tensor src) -> Tensor");
   m.def("tr16_probe(Tensor dummy, int mode) -> Tensor");
   m.def("nms_multiclass(Tensor boxes, Tensor scores, float iou_thr, float score_thr) -> Tensor");
   m.def("roi_align(Tensor input, Tensor rois, int ph, int pw, float spatial_scale, int sampling_ratio) -> Tensor");
@@ -786,6 +798,7 @@ TORCH_LIBRARY_IMPL(vilbert_amd, CUDA, m) {
   m.impl("attention_fp8out", attention_fp8out);
   m.impl("embedding_ln", embedding_ln);
   m.impl("mfma_probe", mfma_probe);
+  m.impl("bperm_probe", bperm_probe);
   m.impl("tr16_probe", tr16_probe);
   m.impl("nms_multiclass", nms_multiclass);
   m.impl("roi_align", roi_align);
