@@ -85,8 +85,8 @@ DEV void lds_store_b128(char* p, uint4 v) { *reinterpret_cast<uint4*>(p) = v; }
 // ((c&1)*2 + (j>>1))*16 + q, packed dword j&1, tile 2*kk + (c>>1).
 // Gated to the inference fast path (no probs export, no dropout, no fp8).
 template <int D, bool KGLOBAL, int NTMAX, bool FP8OUT = false, int THREADS = 256,
-          bool SWAP = false>
-__global__ __launch_bounds__(THREADS) void attn_kernel(
+          bool SWAP = false, int MINW = 1>
+__global__ __launch_bounds__(THREADS, MINW) void attn_kernel(
     const bf16* __restrict__ q, const bf16* __restrict__ kg,
     const bf16* __restrict__ vg, const bf16* __restrict__ mask,
     bf16* __restrict__ out, int B, int H, int Lq, int Lk, int mask_mode,
@@ -126,20 +126,26 @@ __global__ __launch_bounds__(THREADS) void attn_kernel(
 
   // ---- stage K (swizzled row-major) and V (tr-readable subtiles) ---------
   {
-    const int rows_per_pass = blockDim.x / KCH;  // 16 (D=128) or 32 (D=64)
+    constexpr int RPP = THREADS / KCH;  // rows per pass: 16 (D=128), 32 (D=64)
+    // max passes is a TEMPLATE constant (LK_PAD <= NTMAX*16): sizing the
+    // staging arrays [8] regardless made the allocator hold dead uint4 slots
+    // live across the batch-issue window — 48 extra VGPRs for the D=64/NT=4
+    // text-self template, capping occupancy at 4 waves/SIMD where LDS allows 6
+    constexpr int NPMAX = (NTMAX * 16 + RPP - 1) / RPP;  // <= 8
+    const int rows_per_pass = RPP;
     const int r0 = tid / KCH;
     const int c = tid % KCH;
     const long kbase0 = (long)b * Lk * ks + (long)h * D;
     const long vbase0 = (long)b * Lk * vs + (long)h * D;
-    // issue EVERY staging load first (up to 8 passes x K+V), then the LDS
+    // issue EVERY staging load first (up to NPMAX passes x K+V), then the LDS
     // store pass: one batch of HBM round trips instead of one per pass
-    const int npass = (LK_PAD + rows_per_pass - 1) / rows_per_pass;  // <= 8
-    uint4 kr[8], vr[8];
+    const int npass = (LK_PAD + rows_per_pass - 1) / rows_per_pass;
+    uint4 kr[NPMAX], vr[NPMAX];
     // unconditional clamped loads (no per-element r<Lk branch — §5 trap
     // (c)); pad rows re-read row Lk-1: K pads are masked to -3e38 in the
     // softmax and V pads multiply by P == 0
 #pragma unroll
-    for (int pi = 0; pi < 8; ++pi) {
+    for (int pi = 0; pi < NPMAX; ++pi) {
       if (pi >= npass) break;
       const long r = min(r0 + pi * rows_per_pass, Lk - 1);
       if (!KGLOBAL)
@@ -147,7 +153,7 @@ __global__ __launch_bounds__(THREADS) void attn_kernel(
       vr[pi] = *reinterpret_cast<const uint4*>(vg + vbase0 + r * vs + c * 8);
     }
 #pragma unroll
-    for (int pi = 0; pi < 8; ++pi) {
+    for (int pi = 0; pi < NPMAX; ++pi) {
       if (pi >= npass) break;
       const int r = r0 + pi * rows_per_pass;
       if (r >= LK_PAD) break;
@@ -1151,6 +1157,28 @@ void launch_attention_impl(const bf16* q, const bf16* k, const bf16* v,
     if (D == 64) { if (small2) LAUNCH_BHLOOP(64, 4); else LAUNCH_BHLOOP(64, 8); }
     else         { if (small2) LAUNCH_BHLOOP(128, 4); else LAUNCH_BHLOOP(128, 8); }
 #undef LAUNCH_BHLOOP
+    return;
+  }
+  // VILBERT_ATTN_OCC=5|6: register-capped D=64/NT=4 instantiation (min
+  // waves per EU via launch_bounds) — probes whether text-self's 45%-of-
+  // roofline gap is workgroup-concurrency-bound (104+16 regs cap it at
+  // 4 waves/SIMD where its 24 KB LDS would allow 6).
+  const char* occ_e = getenv("VILBERT_ATTN_OCC");
+  const int occ_env = occ_e ? atoi(occ_e) : 0;
+  if (occ_env >= 5 && D == 64 && LK_PAD <= 64 && !kglobal_env && nsplit == 1 &&
+      !pipe_env && out8 == nullptr && probs_out == nullptr && dropm == nullptr) {
+    const size_t lds_occ = sizeof(bf16) * (size_t)(2 * LK_PAD * 64 + 4 * 16 * LK_PAD);
+    const dim3 grido(B * H);
+    if (occ_env >= 6)
+      hipLaunchKernelGGL((attn_kernel<64, false, 4, false, 256, false, 6>),
+                         grido, dim3(256), lds_occ, stream, q, k, v, mask, out,
+                         B, H, Lq, Lk, mask_mode, scale, qs, ks, vs, 1,
+                         nullptr, nullptr, nullptr, 0, nullptr, nullptr);
+    else
+      hipLaunchKernelGGL((attn_kernel<64, false, 4, false, 256, false, 5>),
+                         grido, dim3(256), lds_occ, stream, q, k, v, mask, out,
+                         B, H, Lq, Lk, mask_mode, scale, qs, ks, vs, 1,
+                         nullptr, nullptr, nullptr, 0, nullptr, nullptr);
     return;
   }
   // swapped-S^T fast path (no P_lds; see the SWAP template doc above).
