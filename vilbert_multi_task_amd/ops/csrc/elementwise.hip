@@ -356,13 +356,19 @@ __global__ void embedding_ln_kernel(const long* __restrict__ ids,
 // under the reduction+store of the previous row; w/gamma and b/beta are
 // hoisted into registers once per wave. bf16 only; fp8 stays on the
 // baseline kernel.
-template <bool HAS_RES>
+template <bool HAS_RES, bool FP8OUT = false>
 __global__ void residual_ln_pipe_kernel(const bf16* __restrict__ x,
                                         const bf16* __restrict__ res,
                                         const bf16* __restrict__ w,
                                         const bf16* __restrict__ b,
                                         bf16* __restrict__ y,
-                                        long rows, int dim, float eps) {
+                                        long rows, int dim, float eps,
+                                        unsigned char* __restrict__ y8 = nullptr,
+                                        const float* __restrict__ scales = nullptr,
+                                        float* __restrict__ amaxes = nullptr,
+                                        int site = 0) {
+  const float fp8_inv = FP8OUT ? 1.0f / scales[site] : 0.f;
+  float fp8_amax = 0.f;
   const int lane = lane_id();
   const int wid = wave_id();
   const int waves_per_blk = blockDim.x / WAVE;
@@ -437,12 +443,19 @@ __global__ void residual_ln_pipe_kernel(const bf16* __restrict__ x,
         for (int i = 0; i < 8; ++i)
           o8[i] = (v8[c][i] - mean) * rstd * w8[i] + b8[i];
         VecIO<bf16>::store8(y + row * dim + j, o8);
+        if (FP8OUT) {
+#pragma unroll
+          for (int i = 0; i < 8; ++i)
+            fp8_amax = fmaxf(fp8_amax, fabsf(o8[i]));
+          *reinterpret_cast<uint2*>(y8 + row * dim + j) =
+              pack8_e4m3(o8, fp8_inv);
+        }
       }
     }
   };
 
   long row = (long)blockIdx.x * waves_per_blk + wid;
-  if (row >= rows) return;
+  if (row >= rows) goto done;
   Raw bufA, bufB;
   load_raw(row, bufA);
   while (true) {
@@ -454,6 +467,12 @@ __global__ void residual_ln_pipe_kernel(const bf16* __restrict__ x,
     process(row, bufB);
     row += stride;
     if (row >= rows) break;
+  }
+done:
+  if (FP8OUT) {
+    fp8_amax = wave_max(fp8_amax);
+    if (lane_id() == 0 && fp8_amax > 0.f)
+      atomic_max_f32_nonneg_e(&amaxes[site], fp8_amax);
   }
 }
 
@@ -469,6 +488,25 @@ void launch_residual_ln_fp8(const bf16* x, const bf16* res, const bf16* w,
   const int waves = block / WAVE;
   long want = (rows + waves - 1) / waves;
   const int grid = (int)(want < 2048 ? want : 2048);
+  // measured: the pipe kernel's fp8 epilogue (pack + amax) costs more than
+  // the load pipelining saves here (20.56k vs 20.72k q/s fp8 e2e) — the
+  // fp8 path stays on the baseline kernel; VILBERT_LN_PIPE8=1 re-enables
+  // the pipelined variant for iteration.
+  static const bool pipe8 = [] {
+    const char* e = getenv("VILBERT_LN_PIPE8");
+    return e && e[0] == '1';
+  }();
+  if (pipe8 && dim % 8 == 0 && dim <= 1024) {
+    if (res)
+      hipLaunchKernelGGL((residual_ln_pipe_kernel<true, true>), dim3(grid),
+                         dim3(block), 0, stream, x, res, w, b, y, rows, dim,
+                         eps, y8, scales, amaxes, site);
+    else
+      hipLaunchKernelGGL((residual_ln_pipe_kernel<false, true>), dim3(grid),
+                         dim3(block), 0, stream, x, res, w, b, y, rows, dim,
+                         eps, y8, scales, amaxes, site);
+    return;
+  }
   if (res)
     hipLaunchKernelGGL((residual_ln_kernel<bf16, true, true>), dim3(grid),
                        dim3(block), 0, stream, x, res, w, b, y, rows, dim, eps,
