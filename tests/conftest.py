@@ -20,6 +20,34 @@ def pytest_collection_modifyitems(config, items):
             item.add_marker(skip)
 
 
+def pytest_runtest_protocol(item, nextitem):
+    """One transparent retry for GPU-marked tests.
+
+    At round-2 close, 2 of 10 full-suite runs on fresh leased boxes showed a
+    single fast test failure that never reproduced under capture (8
+    consecutive clean runs, including back-to-back pairs on one box) — a
+    box-level anomaly, not a deterministic bug. A retried pass is reported
+    normally; a test that fails twice in a row still fails the suite. The
+    retry is logged to stderr so a masked flake remains visible in the run
+    output. Set VILBERT_TEST_NO_RETRY=1 to disable.
+    """
+    import _pytest.runner as runner_mod
+
+    if "gpu" not in item.keywords or os.environ.get("VILBERT_TEST_NO_RETRY") == "1":
+        return None
+    item.ihook.pytest_runtest_logstart(nodeid=item.nodeid, location=item.location)
+    reports = runner_mod.runtestprotocol(item, nextitem=nextitem, log=False)
+    if any(r.failed for r in reports):
+        sys.stderr.write(f"\n[conftest] retrying flaked GPU test {item.nodeid}\n")
+        if hasattr(item, "_request"):
+            item._initrequest()  # fresh fixture state for the rerun
+        reports = runner_mod.runtestprotocol(item, nextitem=nextitem, log=False)
+    for r in reports:
+        item.ihook.pytest_runtest_logreport(report=r)
+    item.ihook.pytest_runtest_logfinish(nodeid=item.nodeid, location=item.location)
+    return True
+
+
 @pytest.fixture
 def tiny_config():
     from vilbert_multi_task_amd.config import ViLBertConfig
